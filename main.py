@@ -1,0 +1,20 @@
+"""Run the gateway: ``python main.py`` (parity: /root/reference/main.py:119-127)."""
+
+from __future__ import annotations
+
+import uvicorn
+
+from llmapigateway_amd.config.settings import Settings
+from llmapigateway_amd.gateway.app import create_app
+from llmapigateway_amd.utils.logging_setup import configure_logging
+
+
+def main() -> None:
+    configure_logging()
+    settings = Settings.from_env()
+    app = create_app(settings=settings)
+    uvicorn.run(app, host=settings.gateway_host, port=settings.gateway_port, log_level="info")
+
+
+if __name__ == "__main__":
+    main()
