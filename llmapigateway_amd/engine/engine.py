@@ -1,0 +1,407 @@
+"""LLMEngine: continuous-batching inference engine for one MI355X (or CPU).
+
+The engine replaces the reference's remote upstream as the thing a
+"provider" resolves to. One engine owns one GPU-resident
+model + paged KV cache; the gateway's fallback loop treats engine errors
+(including injected faults) exactly like upstream HTTP errors — raised
+before the first streamed byte so fallback can engage.
+
+Scheduling policy (vLLM-v0-style): each step() is either one prefill
+forward over newly admitted prompts (their first token samples from that
+same forward) or one decode forward over all running sequences. Decode
+out-of-block conditions preempt the youngest running sequence back to the
+waiting queue (its KV is freed; re-admission re-prefills prompt+generated).
+"""
+
+from __future__ import annotations
+
+import itertools
+import logging
+import threading
+import time
+from collections import deque
+from dataclasses import dataclass, field
+from typing import Callable, Dict, List, Optional
+
+import torch
+
+from .. import ops
+from ..models.configs import ModelConfig, get_model_config
+from ..models.llama import ForwardBatch, LlamaModel
+from .kvcache import PagedKVCache
+
+logger = logging.getLogger(__name__)
+
+_req_counter = itertools.count()
+
+
+@dataclass
+class SamplingParams:
+    temperature: float = 0.0
+    top_p: float = 1.0
+    top_k: int = 0
+    max_tokens: int = 128
+    seed: Optional[int] = None
+    stop: List[str] = field(default_factory=list)
+    ignore_eos: bool = False
+
+    @classmethod
+    def from_payload(cls, payload: dict, default_max_tokens: int = 256) -> "SamplingParams":
+        stop = payload.get("stop") or []
+        if isinstance(stop, str):
+            stop = [stop]
+        return cls(
+            temperature=float(payload.get("temperature", 0.0) or 0.0),
+            top_p=float(payload.get("top_p", 1.0) or 1.0),
+            top_k=int(payload.get("top_k", 0) or 0),
+            max_tokens=int(
+                payload.get("max_completion_tokens")
+                or payload.get("max_tokens")
+                or default_max_tokens
+            ),
+            seed=payload.get("seed"),
+            stop=[s for s in stop if isinstance(s, str)],
+        )
+
+
+class EngineRequest:
+    def __init__(
+        self,
+        prompt_ids: List[int],
+        params: SamplingParams,
+        on_token: Optional[Callable[["EngineRequest", int], None]] = None,
+        on_finish: Optional[Callable[["EngineRequest"], None]] = None,
+    ):
+        self.id = f"req-{next(_req_counter)}"
+        self.prompt_ids = list(prompt_ids)
+        self.params = params
+        self.out_ids: List[int] = []
+        self.block_table: List[int] = []
+        self.state = "waiting"  # waiting | running | finished | failed
+        self.finish_reason: Optional[str] = None
+        self.error: Optional[str] = None
+        self.on_token = on_token
+        self.on_finish = on_finish
+        self.created = time.monotonic()
+        self.first_token_time: Optional[float] = None
+
+    @property
+    def num_tokens(self) -> int:
+        return len(self.prompt_ids) + len(self.out_ids)
+
+    def __repr__(self) -> str:
+        return f"<EngineRequest {self.id} {self.state} {len(self.prompt_ids)}+{len(self.out_ids)}>"
+
+
+class LLMEngine:
+    def __init__(
+        self,
+        model: str | ModelConfig = "llama-3-8b",
+        device: torch.device | str = "cpu",
+        dtype: torch.dtype = torch.bfloat16,
+        block_size: int = 64,
+        max_batch_size: int = 64,
+        num_blocks: Optional[int] = None,
+        hbm_fraction: float = 0.90,
+        max_model_len: Optional[int] = None,
+        seed: int = 0,
+        tp_group: Optional[object] = None,
+        tp_size: int = 1,
+    ):
+        full_config = get_model_config(model) if isinstance(model, str) else model
+        self.full_config = full_config
+        config = full_config.scaled_for_tp(tp_size) if tp_size > 1 else full_config
+        self.config = config
+        self.device = torch.device(device)
+        self.dtype = dtype
+        self.max_model_len = min(
+            max_model_len or full_config.max_positions, full_config.max_positions
+        )
+        self.max_batch_size = max_batch_size
+
+        if self.device.type == "cuda":
+            torch.cuda.set_device(self.device)
+        self.model = LlamaModel(
+            config, device=self.device, dtype=dtype, seed=seed,
+            tp_group=tp_group, full_config=full_config,
+        )
+        if num_blocks is None:
+            num_blocks = PagedKVCache.fit_num_blocks(
+                config, block_size, self.device, dtype, hbm_fraction
+            )
+        self.kv = PagedKVCache(config, num_blocks, block_size, self.device, dtype)
+
+        self._gen = torch.Generator(device=self.device).manual_seed(seed ^ 0x5EED)
+        self._lock = threading.Lock()
+        self._work = threading.Condition(self._lock)
+        self.waiting: deque[EngineRequest] = deque()
+        self.running: List[EngineRequest] = []
+        self.stats: Dict[str, float] = {
+            "requests": 0,
+            "finished": 0,
+            "failed": 0,
+            "prefill_tokens": 0,
+            "decode_tokens": 0,
+            "steps": 0,
+        }
+
+    # ---- request intake ----
+    def add_request(self, req: EngineRequest) -> EngineRequest:
+        if len(req.prompt_ids) >= self.max_model_len:
+            raise ValueError(
+                f"Prompt of {len(req.prompt_ids)} tokens exceeds max_model_len={self.max_model_len}"
+            )
+        with self._work:
+            self.waiting.append(req)
+            self.stats["requests"] += 1
+            self._work.notify_all()
+        return req
+
+    def abort_request(self, req: EngineRequest) -> None:
+        with self._work:
+            if req in self.waiting:
+                self.waiting.remove(req)
+            self._finish(req, "aborted")
+
+    def has_work(self) -> bool:
+        with self._lock:
+            return bool(self.waiting or self.running)
+
+    def wait_for_work(self, timeout: float = 0.2) -> bool:
+        with self._work:
+            if self.waiting or self.running:
+                return True
+            self._work.wait(timeout)
+            return bool(self.waiting or self.running)
+
+    # ---- scheduling ----
+    def _admit(self) -> List[EngineRequest]:
+        admitted: List[EngineRequest] = []
+        budget_tokens = 8192  # prefill-batch token budget per step
+        while self.waiting and len(self.running) + len(admitted) < self.max_batch_size:
+            req = self.waiting[0]
+            need = len(req.prompt_ids)
+            if admitted and sum(len(r.prompt_ids) for r in admitted) + need > budget_tokens:
+                break
+            if not self.kv.manager.can_allocate(need + 1):
+                break
+            self.waiting.popleft()
+            req.block_table = self.kv.manager.allocate(need)
+            req.state = "running"
+            admitted.append(req)
+        return admitted
+
+    def _preempt_youngest(self) -> bool:
+        if not self.running:
+            return False
+        victim = self.running.pop()
+        self.kv.manager.free(victim.block_table)
+        victim.block_table = []
+        victim.state = "waiting"
+        # re-admission re-prefills prompt + generated so far
+        victim.prompt_ids = victim.prompt_ids + victim.out_ids
+        victim.out_ids = []
+        self.waiting.appendleft(victim)
+        logger.warning("Preempted %s (KV blocks exhausted)", victim.id)
+        return True
+
+    # ---- the step ----
+    def step(self) -> int:
+        """Run one engine iteration. Returns number of tokens produced."""
+        with self._lock:
+            admitted = self._admit()
+        try:
+            if admitted:
+                produced = self._prefill_step(admitted)
+            else:
+                with self._lock:
+                    if not self.running:
+                        return 0
+                produced = self._decode_step()
+        except Exception as e:
+            logger.exception("engine step failed")
+            with self._lock:
+                for req in list(self.running) + admitted:
+                    req.error = f"engine error: {e}"
+                    self._finish(req, "error")
+                self.running.clear()
+            raise
+        self.stats["steps"] += 1
+        return produced
+
+    def _prefill_step(self, reqs: List[EngineRequest]) -> int:
+        device = self.device
+        bs = self.kv.block_size
+        token_ids, positions, slots, cu, logits_idx = [], [], [], [0], []
+        max_len = 0
+        for req in reqs:
+            L = len(req.prompt_ids)
+            token_ids.extend(req.prompt_ids)
+            positions.extend(range(L))
+            slots.extend(
+                req.block_table[p // bs] * bs + p % bs for p in range(L)
+            )
+            cu.append(cu[-1] + L)
+            logits_idx.append(cu[-1] - 1)
+            max_len = max(max_len, L)
+
+        batch = ForwardBatch(
+            kind="prefill",
+            token_ids=torch.tensor(token_ids, dtype=torch.long, device=device),
+            positions=torch.tensor(positions, dtype=torch.long, device=device),
+            slot_mapping=torch.tensor(slots, dtype=torch.long, device=device),
+            cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=device),
+            max_seqlen=max_len,
+            logits_indices=torch.tensor(logits_idx, dtype=torch.long, device=device),
+        )
+        logits = self.model.forward(batch, self.kv.k_caches, self.kv.v_caches)
+        tokens = self._sample(logits, reqs)
+        self.stats["prefill_tokens"] += len(token_ids)
+        with self._lock:
+            self.running.extend(reqs)
+            self._deliver(reqs, tokens)
+        return len(reqs)
+
+    def _decode_step(self) -> int:
+        device = self.device
+        bs = self.kv.block_size
+        with self._lock:
+            # ensure every running seq has a block for the incoming token
+            i = 0
+            while i < len(self.running):
+                req = self.running[i]
+                try:
+                    self.kv.manager.extend(req.block_table, req.num_tokens, req.num_tokens + 1)
+                    i += 1
+                except RuntimeError:
+                    if not self._preempt_youngest():
+                        raise
+                    # if we preempted the request we were extending, skip it
+            reqs = list(self.running)
+        if not reqs:
+            return 0
+
+        last_tokens = [
+            (req.out_ids[-1] if req.out_ids else req.prompt_ids[-1]) for req in reqs
+        ]
+        pos = [req.num_tokens - 1 for req in reqs]
+        slots = [req.block_table[p // bs] * bs + p % bs for req, p in zip(reqs, pos)]
+        max_blocks = max(len(req.block_table) for req in reqs)
+        tables = torch.zeros(len(reqs), max_blocks, dtype=torch.int32, device=device)
+        for i, req in enumerate(reqs):
+            tables[i, : len(req.block_table)] = torch.tensor(
+                req.block_table, dtype=torch.int32
+            )
+
+        batch = ForwardBatch(
+            kind="decode",
+            token_ids=torch.tensor(last_tokens, dtype=torch.long, device=device),
+            positions=torch.tensor(pos, dtype=torch.long, device=device),
+            slot_mapping=torch.tensor(slots, dtype=torch.long, device=device),
+            block_tables=tables,
+            context_lens=torch.tensor(
+                [p + 1 for p in pos], dtype=torch.int32, device=device
+            ),
+            logits_indices=None,
+        )
+        logits = self.model.forward(batch, self.kv.k_caches, self.kv.v_caches)
+        tokens = self._sample(logits, reqs)
+        self.stats["decode_tokens"] += len(reqs)
+        with self._lock:
+            self._deliver(reqs, tokens)
+        return len(reqs)
+
+    # ---- sampling ----
+    def _sample(self, logits: torch.Tensor, reqs: List[EngineRequest]) -> List[int]:
+        temps = torch.tensor(
+            [r.params.temperature for r in reqs], dtype=torch.float32, device=logits.device
+        )
+        need_noise = bool((temps > 0).any())
+        filtered = logits
+        if need_noise:
+            filtered = self._apply_topk_topp(logits.clone(), reqs)
+            noise = torch.rand(
+                logits.shape, generator=self._gen, device=logits.device, dtype=torch.float32
+            )
+        else:
+            noise = None
+        out = ops.sample(filtered, temps, noise)
+        return out.tolist()
+
+    def _apply_topk_topp(self, logits: torch.Tensor, reqs: List[EngineRequest]) -> torch.Tensor:
+        needs = [
+            i
+            for i, r in enumerate(reqs)
+            if r.params.temperature > 0 and (r.params.top_k > 0 or r.params.top_p < 1.0)
+        ]
+        if not needs:
+            return logits
+        sub = logits[needs].float()
+        sorted_logits, sorted_idx = torch.sort(sub, descending=True, dim=-1)
+        mask = torch.zeros_like(sub, dtype=torch.bool)
+        for row, i in enumerate(needs):
+            p = reqs[i].params
+            if p.top_k > 0:
+                mask[row].scatter_(0, sorted_idx[row, p.top_k :], True)
+            if p.top_p < 1.0:
+                probs = torch.softmax(sorted_logits[row], dim=-1)
+                cdf = probs.cumsum(dim=-1)
+                cut = torch.searchsorted(cdf, torch.tensor(p.top_p, device=cdf.device)) + 1
+                mask[row].scatter_(0, sorted_idx[row, cut:], True)
+        sub.masked_fill_(mask, float("-inf"))
+        logits[needs] = sub.to(logits.dtype)
+        return logits
+
+    # ---- delivery / lifecycle (call with lock held) ----
+    def _deliver(self, reqs: List[EngineRequest], tokens: List[int]) -> None:
+        now = time.monotonic()
+        for req, tok in zip(reqs, tokens):
+            if req.state != "running":
+                continue
+            req.out_ids.append(int(tok))
+            if req.first_token_time is None:
+                req.first_token_time = now
+            if req.on_token is not None:
+                try:
+                    req.on_token(req, int(tok))
+                except Exception:
+                    logger.exception("on_token callback failed for %s", req.id)
+            reason = self._finish_reason(req, int(tok))
+            if reason:
+                self.running.remove(req)
+                self._finish(req, reason)
+
+    def _finish_reason(self, req: EngineRequest, tok: int) -> Optional[str]:
+        from .tokenizer import ByteTokenizer
+
+        if not req.params.ignore_eos and tok == ByteTokenizer.EOS:
+            return "stop"
+        if len(req.out_ids) >= req.params.max_tokens:
+            return "length"
+        if req.num_tokens >= self.max_model_len:
+            return "length"
+        return None
+
+    def _finish(self, req: EngineRequest, reason: str) -> None:
+        if req.state in ("finished", "failed"):
+            return
+        req.state = "failed" if reason == "error" else "finished"
+        req.finish_reason = reason
+        self.stats["finished" if req.state == "finished" else "failed"] += 1
+        if req.block_table:
+            self.kv.manager.free(req.block_table)
+            req.block_table = []
+        if req.on_finish is not None:
+            try:
+                req.on_finish(req)
+            except Exception:
+                logger.exception("on_finish callback failed for %s", req.id)
+
+    # ---- convenience driver (tests / bench) ----
+    def generate(self, prompt_ids: List[int], params: SamplingParams) -> EngineRequest:
+        req = EngineRequest(prompt_ids, params)
+        self.add_request(req)
+        while req.state in ("waiting", "running"):
+            if self.step() == 0 and req.state == "waiting":
+                raise RuntimeError("engine made no progress")
+        return req

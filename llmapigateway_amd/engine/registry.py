@@ -1,0 +1,339 @@
+"""EngineRegistry: local providers -> GPU-resident engines.
+
+The registry is the local-backend half of the dispatch service (SURVEY.md
+§2b): it keeps one LLMEngine per (model, device, tp) key, each driven by its
+own daemon thread, and turns OpenAI chat payloads into engine requests with
+the reference's make_llm_request semantics
+(/root/reference/llm_gateway_core/services/request_handler.py:8): returns
+(response, None) on success or (None, error_detail) on any failure that
+happens *before the first streamed byte* — injected faults, engine build
+errors, OOM — so the chat fallback loop can move to the next model.
+
+Failure injection (BASELINE configs[2]) is first-class: per-provider
+fail_rate / fail_requests from the EngineSpec, overridable at runtime via
+set_failure() (exposed by the admin API route).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+import logging
+import threading
+import time
+import uuid
+from typing import Any, Dict, List, Optional, Tuple
+
+import torch
+from fastapi.responses import StreamingResponse
+
+from ..config.loader import EngineSpec
+from ..config.settings import Settings
+from .engine import EngineRequest, LLMEngine, SamplingParams
+from .tokenizer import ByteTokenizer
+
+logger = logging.getLogger(__name__)
+
+STREAM_HEADERS = {"X-Accel-Buffering": "no"}
+
+
+class _EngineHandle:
+    def __init__(self, engine: LLMEngine, name: str):
+        self.engine = engine
+        self.name = name
+        self.stop_event = threading.Event()
+        self.thread = threading.Thread(target=self._loop, name=f"engine-{name}", daemon=True)
+        self.thread.start()
+
+    def _loop(self) -> None:
+        logger.info("engine loop started: %s", self.name)
+        while not self.stop_event.is_set():
+            try:
+                if self.engine.wait_for_work(timeout=0.05):
+                    self.engine.step()
+            except Exception:
+                # step() already failed the affected requests; keep serving
+                logger.exception("engine %s step error", self.name)
+                time.sleep(0.01)
+
+    def stop(self) -> None:
+        self.stop_event.set()
+        self.thread.join(timeout=5.0)
+
+
+class _FailureState:
+    def __init__(self, fail_rate: float = 0.0, fail_requests: Optional[int] = None):
+        self.fail_rate = fail_rate
+        self.fail_requests = fail_requests
+        self.counter = 0
+
+    def should_fail(self) -> bool:
+        self.counter += 1
+        if self.fail_requests is not None and self.counter <= self.fail_requests:
+            return True
+        if self.fail_rate > 0:
+            # deterministic pseudo-random by counter (reproducible tests)
+            x = (self.counter * 2654435761) & 0xFFFFFFFF
+            return (x / 0xFFFFFFFF) < self.fail_rate
+        return False
+
+
+class EngineRegistry:
+    def __init__(self, settings: Optional[Settings] = None):
+        self.settings = settings or Settings()
+        self._engines: Dict[str, _EngineHandle] = {}
+        self._failures: Dict[str, _FailureState] = {}
+        self._create_lock = threading.Lock()
+        self.tokenizer = ByteTokenizer()
+
+    # ---- engine lifecycle ----
+    def _engine_key(self, spec: EngineSpec) -> str:
+        return f"{spec.model}@{spec.device}x{spec.tp}"
+
+    def _resolve_device(self, spec: EngineSpec) -> str:
+        if torch.cuda.is_available():
+            n = torch.cuda.device_count()
+            if spec.device >= n:
+                raise RuntimeError(
+                    f"Engine spec asks for GPU {spec.device} but only {n} are visible"
+                )
+            return f"cuda:{spec.device}"
+        return "cpu"
+
+    def get_engine(self, spec: EngineSpec) -> _EngineHandle:
+        key = self._engine_key(spec)
+        handle = self._engines.get(key)
+        if handle is not None:
+            return handle
+        with self._create_lock:
+            handle = self._engines.get(key)
+            if handle is not None:
+                return handle
+            device = self._resolve_device(spec)
+            dtype = torch.bfloat16 if device.startswith("cuda") else torch.float32
+            if spec.dtype in ("float16", "fp16"):
+                dtype = torch.float16
+            engine = LLMEngine(
+                model=spec.model,
+                device=device,
+                dtype=dtype,
+                block_size=spec.kv_block_size or self.settings.engine_kv_block_size,
+                max_batch_size=spec.max_batch_size or self.settings.engine_max_batch_size,
+                hbm_fraction=self.settings.engine_hbm_fraction,
+                num_blocks=None if device.startswith("cuda") else 256,
+            )
+            handle = _EngineHandle(engine, key)
+            self._engines[key] = handle
+            logger.info("created engine %s on %s", key, device)
+            return handle
+
+    def _failure_state(self, provider: str, spec: EngineSpec) -> _FailureState:
+        st = self._failures.get(provider)
+        if st is None:
+            st = _FailureState(spec.fail_rate, spec.fail_requests)
+            self._failures[provider] = st
+        return st
+
+    def set_failure(
+        self, provider: str, fail_rate: float = 0.0, fail_requests: Optional[int] = None
+    ) -> None:
+        st = _FailureState(fail_rate, fail_requests)
+        self._failures[provider] = st
+
+    # ---- the request path ----
+    async def make_request(
+        self,
+        provider_name: str,
+        spec: EngineSpec,
+        payload: Dict[str, Any],
+        is_streaming: bool,
+    ) -> Tuple[Optional[Any], Optional[str]]:
+        fail = self._failure_state(provider_name, spec)
+        if fail.should_fail():
+            detail = f"Injected failure on local provider '{provider_name}' (request #{fail.counter})"
+            logger.warning(detail)
+            return None, detail
+
+        try:
+            handle = await asyncio.to_thread(self.get_engine, spec)
+        except Exception as e:
+            return None, f"Engine for provider '{provider_name}' failed to start: {e}"
+        engine = handle.engine
+
+        messages = payload.get("messages")
+        if messages:
+            prompt_text = self.tokenizer.render_chat(messages)
+        else:
+            prompt_text = str(payload.get("prompt", ""))
+        prompt_ids = self.tokenizer.encode(prompt_text)
+        if len(prompt_ids) >= engine.max_model_len:
+            return None, (
+                f"Prompt of {len(prompt_ids)} tokens exceeds model context "
+                f"({engine.max_model_len}) on '{provider_name}'"
+            )
+        params = SamplingParams.from_payload(payload)
+        params.max_tokens = min(params.max_tokens, engine.max_model_len - len(prompt_ids) - 1)
+        # random-init weights rarely emit EOS; respect explicit ignore_eos
+        if "ignore_eos" in payload:
+            params.ignore_eos = bool(payload["ignore_eos"])
+
+        loop = asyncio.get_running_loop()
+        queue: asyncio.Queue = asyncio.Queue()
+
+        def on_token(req: EngineRequest, tok: int) -> None:
+            loop.call_soon_threadsafe(queue.put_nowait, ("token", tok))
+
+        def on_finish(req: EngineRequest) -> None:
+            loop.call_soon_threadsafe(queue.put_nowait, ("finish", req))
+
+        req = EngineRequest(prompt_ids, params, on_token=on_token, on_finish=on_finish)
+        try:
+            engine.add_request(req)
+        except Exception as e:
+            return None, f"Engine '{provider_name}' rejected request: {e}"
+
+        model_name = payload.get("model", spec.model)
+        completion_id = f"chatcmpl-{uuid.uuid4().hex[:24]}"
+        created = int(time.time())
+
+        # prime: wait for the first token (or failure) BEFORE returning, so a
+        # failing request yields (None, error) with zero bytes sent —
+        # first-chunk semantics parity (request_handler.py:67-100)
+        try:
+            kind, value = await asyncio.wait_for(queue.get(), timeout=300.0)
+        except asyncio.TimeoutError:
+            engine.abort_request(req)
+            return None, f"Engine '{provider_name}' timed out before first token"
+        if kind == "finish" and value.state == "failed":
+            return None, value.error or "engine failure"
+
+        if not is_streaming:
+            return await self._collect_nonstream(
+                req, queue, kind, value, completion_id, created, model_name, provider_name
+            )
+        return (
+            self._stream_response(
+                req, engine, queue, kind, value, completion_id, created, model_name
+            ),
+            None,
+        )
+
+    async def _collect_nonstream(
+        self, req, queue, kind, value, completion_id, created, model_name, provider_name
+    ):
+        tokens: List[int] = []
+        finished: Optional[EngineRequest] = None
+        while True:
+            if kind == "token":
+                tokens.append(value)
+            else:
+                finished = value
+                break
+            kind, value = await queue.get()
+        if finished is not None and finished.state == "failed":
+            return None, finished.error or "engine failure"
+        text = self.tokenizer.decode(tokens)
+        usage = {
+            "prompt_tokens": len(req.prompt_ids),
+            "completion_tokens": len(req.out_ids),
+            "total_tokens": len(req.prompt_ids) + len(req.out_ids),
+        }
+        return (
+            {
+                "id": completion_id,
+                "object": "chat.completion",
+                "created": created,
+                "model": model_name,
+                "choices": [
+                    {
+                        "index": 0,
+                        "message": {"role": "assistant", "content": text},
+                        "finish_reason": req.finish_reason or "stop",
+                    }
+                ],
+                "usage": usage,
+            },
+            None,
+        )
+
+    def _stream_response(
+        self, req, engine, queue, first_kind, first_value, completion_id, created, model_name
+    ) -> StreamingResponse:
+        tokenizer = self.tokenizer
+
+        def chunk(delta: Dict[str, Any], finish: Optional[str] = None, usage=None) -> bytes:
+            obj: Dict[str, Any] = {
+                "id": completion_id,
+                "object": "chat.completion.chunk",
+                "created": created,
+                "model": model_name,
+                "choices": [{"index": 0, "delta": delta, "finish_reason": finish}],
+            }
+            if usage is not None:
+                obj["usage"] = usage
+            return b"data: " + json.dumps(obj, separators=(",", ":")).encode() + b"\n\n"
+
+        async def gen():
+            try:
+                yield chunk({"role": "assistant", "content": ""})
+                kind, value = first_kind, first_value
+                while True:
+                    if kind == "token":
+                        yield chunk({"content": tokenizer.decode([value])})
+                    else:
+                        fin: EngineRequest = value
+                        usage = {
+                            "prompt_tokens": len(fin.prompt_ids),
+                            "completion_tokens": len(fin.out_ids),
+                            "total_tokens": len(fin.prompt_ids) + len(fin.out_ids),
+                        }
+                        yield chunk({}, finish=fin.finish_reason or "stop", usage=usage)
+                        yield b"data: [DONE]\n\n"
+                        return
+                    kind, value = await queue.get()
+            finally:
+                if req.state in ("waiting", "running"):
+                    engine.abort_request(req)
+
+        return StreamingResponse(
+            gen(), media_type="text/event-stream", headers=dict(STREAM_HEADERS)
+        )
+
+    # ---- introspection ----
+    def list_models(self, spec: EngineSpec) -> List[Dict[str, Any]]:
+        from ..models.configs import MODEL_PRESETS
+
+        out = []
+        for name, cfg in sorted(MODEL_PRESETS.items()):
+            out.append(
+                {
+                    "id": name,
+                    "object": "model",
+                    "owned_by": "llmapigateway-amd-engine",
+                    "context_length": cfg.max_positions,
+                }
+            )
+        return out
+
+    def stats(self) -> List[Dict[str, Any]]:
+        out = []
+        for key, handle in self._engines.items():
+            eng = handle.engine
+            out.append(
+                {
+                    "engine": key,
+                    "device": str(eng.device),
+                    "model": eng.full_config.name,
+                    "waiting": len(eng.waiting),
+                    "running": len(eng.running),
+                    "kv_blocks_free": eng.kv.manager.num_free_blocks,
+                    "kv_blocks_total": eng.kv.num_blocks,
+                    **{k: int(v) for k, v in eng.stats.items()},
+                }
+            )
+        return out
+
+    async def aclose(self) -> None:
+        for handle in self._engines.values():
+            handle.stop()
+        self._engines.clear()

@@ -1,5 +1,6 @@
 from fastapi import APIRouter
 
+from .admin import router as admin_router
 from .chat import router as chat_router
 from .models import router as models_router
 from .rules_editor import router as rules_editor_router
@@ -12,5 +13,6 @@ v1_router.include_router(chat_router, prefix="/chat", tags=["Chat"])
 v1_router.include_router(models_router, prefix="/models", tags=["Models"])
 v1_router.include_router(rules_editor_router, tags=["Config Editor"])
 v1_router.include_router(stats_router, tags=["Usage Stats"])
+v1_router.include_router(admin_router, tags=["Admin"])
 
 __all__ = ["v1_router"]

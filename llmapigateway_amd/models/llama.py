@@ -1,0 +1,161 @@
+"""Llama-family decoder (Llama-3, Mistral) built directly on the MI355X ops.
+
+Inference-only, bf16 weights, random init (no network for checkpoints —
+BASELINE.json: synthetic data / random-init weights). The forward pass is a
+flat varlen batch (ForwardBatch) in one of two modes:
+
+- prefill: fresh prompts; ops.attention_prefill over the batch's own K/V
+  (and K/V written to the paged cache for later decode);
+- decode: one token per running sequence; ops.attention_decode over the
+  paged cache via block tables.
+
+Projections run on hipBLASLt/rocBLAS through F.linear (library GEMMs); the
+fused hot ops (rmsnorm+residual, rope, swiglu, attention, sampling) are the
+hand-written CDNA4 kernels behind llmapigateway_amd.ops.
+
+Tensor parallelism: construct with ``tp_group`` + a config pre-sharded via
+ModelConfig.scaled_for_tp; qkv/gate_up are column-sharded, o/down
+row-sharded with an RCCL all-reduce over xGMI after each (2 per layer).
+"""
+
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+from typing import Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn.functional as F
+
+from .. import ops
+from .configs import ModelConfig
+
+
+@dataclass
+class ForwardBatch:
+    kind: str  # "prefill" | "decode"
+    token_ids: torch.Tensor          # [T] int64
+    positions: torch.Tensor          # [T] int64
+    slot_mapping: torch.Tensor       # [T] int64 (global KV slots; -1 = don't write)
+    # prefill:
+    cu_seqlens: Optional[torch.Tensor] = None   # [B+1] int32
+    max_seqlen: int = 0
+    # decode:
+    block_tables: Optional[torch.Tensor] = None  # [B, max_blocks] int32
+    context_lens: Optional[torch.Tensor] = None  # [B] int32
+    # rows of the flat batch at which logits are needed (last token per seq)
+    logits_indices: Optional[torch.Tensor] = None  # [B] int64
+
+
+class LlamaModel:
+    def __init__(
+        self,
+        config: ModelConfig,
+        device: torch.device | str = "cpu",
+        dtype: torch.dtype = torch.bfloat16,
+        seed: int = 0,
+        tp_group: Optional[object] = None,
+        full_config: Optional[ModelConfig] = None,
+    ):
+        self.config = config
+        self.full_config = full_config or config
+        self.device = torch.device(device)
+        self.dtype = dtype
+        self.tp_group = tp_group
+        self.scale = config.head_dim ** -0.5
+        self.layers: List[Dict[str, torch.Tensor]] = []
+        self._init_weights(seed)
+        self.cos_sin = ops.build_rope_cache(
+            config.max_positions, config.head_dim, config.rope_theta, device=self.device
+        )
+
+    # ---- weights ----
+    def _init_weights(self, seed: int) -> None:
+        c = self.config
+        gen = torch.Generator(device="cpu").manual_seed(seed)
+        hidden = c.hidden_size
+
+        def mk(rows: int, cols: int, std: float) -> torch.Tensor:
+            w = torch.empty(rows, cols, dtype=torch.float32)
+            w.normal_(0.0, std, generator=gen)
+            return w.to(self.dtype).to(self.device)
+
+        std = 0.02
+        out_std = 0.02 / math.sqrt(2 * c.num_layers)
+        self.embed = mk(self.full_config.vocab_size, hidden, std)
+        for _ in range(c.num_layers):
+            self.layers.append(
+                {
+                    "input_norm": torch.ones(hidden, dtype=self.dtype, device=self.device),
+                    "qkv": mk(c.q_size + 2 * c.kv_size, hidden, std),
+                    "o": mk(hidden, c.q_size, out_std),
+                    "post_norm": torch.ones(hidden, dtype=self.dtype, device=self.device),
+                    "gate_up": mk(2 * c.intermediate_size, hidden, std),
+                    "down": mk(hidden, c.intermediate_size, out_std),
+                }
+            )
+        self.final_norm = torch.ones(hidden, dtype=self.dtype, device=self.device)
+        self.lm_head = self.embed if c.tie_embeddings else mk(self.full_config.vocab_size, hidden, std)
+
+    def param_bytes(self) -> int:
+        total = self.embed.numel() + self.final_norm.numel()
+        if self.lm_head is not self.embed:
+            total += self.lm_head.numel()
+        for layer in self.layers:
+            total += sum(t.numel() for t in layer.values())
+        return total * self.embed.element_size()
+
+    # ---- forward ----
+    def _maybe_all_reduce(self, x: torch.Tensor) -> torch.Tensor:
+        if self.tp_group is not None:
+            dist.all_reduce(x, group=self.tp_group)
+        return x
+
+    @torch.inference_mode()
+    def forward(
+        self,
+        batch: ForwardBatch,
+        k_caches: List[torch.Tensor],
+        v_caches: List[torch.Tensor],
+    ) -> torch.Tensor:
+        """Returns logits [B, vocab] at batch.logits_indices."""
+        c = self.config
+        h = F.embedding(batch.token_ids, self.embed)
+        residual = torch.zeros_like(h)
+        T = h.shape[0]
+
+        for i, layer in enumerate(self.layers):
+            if i == 0:
+                x, residual = ops.rmsnorm(h, layer["input_norm"], c.rms_eps), h
+            else:
+                x, residual = ops.rmsnorm_residual(h, residual, layer["input_norm"], c.rms_eps)
+
+            qkv = F.linear(x, layer["qkv"])
+            q, k, v = qkv.split([c.q_size, c.kv_size, c.kv_size], dim=-1)
+            q = q.view(T, c.num_heads, c.head_dim)
+            k = k.view(T, c.num_kv_heads, c.head_dim)
+            v = v.view(T, c.num_kv_heads, c.head_dim)
+            ops.rope_inplace(q, k, batch.positions, self.cos_sin)
+            ops.kv_cache_write(k, v, k_caches[i], v_caches[i], batch.slot_mapping)
+
+            if batch.kind == "prefill":
+                attn = ops.attention_prefill(
+                    q, k, v, batch.cu_seqlens, batch.max_seqlen, self.scale
+                )
+            else:
+                attn = ops.attention_decode(
+                    q, k_caches[i], v_caches[i], batch.block_tables, batch.context_lens, self.scale
+                )
+
+            h = self._maybe_all_reduce(F.linear(attn.reshape(T, c.q_size), layer["o"]))
+
+            x, residual = ops.rmsnorm_residual(h, residual, layer["post_norm"], c.rms_eps)
+            h = self._maybe_all_reduce(
+                F.linear(ops.swiglu(F.linear(x, layer["gate_up"])), layer["down"])
+            )
+
+        x, _ = ops.rmsnorm_residual(h, residual, self.final_norm, c.rms_eps)
+        if batch.logits_indices is not None:
+            x = x[batch.logits_indices]
+        return F.linear(x, self.lm_head).float()

@@ -1,0 +1,156 @@
+"""Op dispatch: gfx950 HIP kernels on GPU, PyTorch reference on CPU.
+
+The HIP extension (llmapigateway_amd/ops/_C*.so, built in-tree by
+``python setup.py build_ext --inplace`` / __graft_entry__.build) is the ONLY
+compute path on GPU tensors — if a CUDA tensor reaches an op and the
+extension is missing, we raise instead of silently falling back to eager
+PyTorch. CPU tensors use ops.reference (fp32 semantics contract).
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from . import reference
+from .reference import build_rope_cache  # re-export (host-side table builder)
+
+_C = None
+_C_ERR: Optional[str] = None
+try:
+    from . import _C as _C_mod  # type: ignore
+
+    _C = _C_mod
+except ImportError as e:  # extension not built (CPU-only envs are fine)
+    _C_ERR = str(e)
+
+
+def have_native() -> bool:
+    return _C is not None
+
+
+def _native():
+    if _C is None:
+        raise RuntimeError(
+            "llmapigateway_amd.ops._C (gfx950 HIP extension) is not built but a GPU tensor "
+            "reached the ops layer. Build it with `python setup.py build_ext --inplace` "
+            f"(import error: {_C_ERR})"
+        )
+    return _C
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    if x.is_cuda:
+        out = torch.empty_like(x)
+        _native().rmsnorm(out, x, weight, eps)
+        return out
+    return reference.rmsnorm(x, weight, eps)
+
+
+def rmsnorm_residual(
+    x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    if x.is_cuda:
+        # fused: residual += x; x_out = rmsnorm(residual) — one HBM pass
+        out = torch.empty_like(x)
+        _native().rmsnorm_residual(out, x, residual, weight, eps)
+        return out, residual
+    return reference.rmsnorm_residual(x, residual, weight, eps)
+
+
+def rope_inplace(
+    q: torch.Tensor, k: torch.Tensor, positions: torch.Tensor, cos_sin: torch.Tensor
+) -> None:
+    if q.is_cuda:
+        _native().rope_inplace(q, k, positions, cos_sin)
+        return
+    reference.rope_inplace(q, k, positions, cos_sin)
+
+
+def swiglu(x: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda:
+        inter = x.shape[-1] // 2
+        out = torch.empty(
+            (*x.shape[:-1], inter), dtype=x.dtype, device=x.device
+        )
+        _native().swiglu(out, x)
+        return out
+    return reference.swiglu(x)
+
+
+def kv_cache_write(
+    k: torch.Tensor,
+    v: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    slot_mapping: torch.Tensor,
+) -> None:
+    if k.is_cuda:
+        _native().kv_cache_write(k, v, k_cache, v_cache, slot_mapping)
+        return
+    reference.kv_cache_write(k, v, k_cache, v_cache, slot_mapping)
+
+
+def attention_prefill(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    cu_seqlens: torch.Tensor,
+    max_seqlen: int,
+    scale: Optional[float] = None,
+    causal: bool = True,
+) -> torch.Tensor:
+    if scale is None:
+        scale = float(q.shape[-1]) ** -0.5
+    if q.is_cuda:
+        out = torch.empty_like(q)
+        _native().attention_prefill(out, q, k, v, cu_seqlens, int(max_seqlen), float(scale), causal)
+        return out
+    return reference.attention_prefill(q, k, v, cu_seqlens, scale, causal)
+
+
+def attention_decode(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,
+    context_lens: torch.Tensor,
+    scale: Optional[float] = None,
+) -> torch.Tensor:
+    if scale is None:
+        scale = float(q.shape[-1]) ** -0.5
+    if q.is_cuda:
+        out = torch.empty_like(q)
+        _native().attention_decode(
+            out, q, k_cache, v_cache, block_tables, context_lens, float(scale)
+        )
+        return out
+    return reference.attention_decode(q, k_cache, v_cache, block_tables, context_lens, scale)
+
+
+def sample(
+    logits: torch.Tensor,
+    temperature: torch.Tensor,
+    noise: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    if logits.is_cuda:
+        out = torch.empty(logits.shape[0], dtype=torch.long, device=logits.device)
+        _native().sample(out, logits, temperature, noise)
+        return out
+    return reference.sample(logits, temperature, noise)
+
+
+__all__ = [
+    "have_native",
+    "build_rope_cache",
+    "rmsnorm",
+    "rmsnorm_residual",
+    "rope_inplace",
+    "swiglu",
+    "kv_cache_write",
+    "attention_prefill",
+    "attention_decode",
+    "sample",
+    "reference",
+]

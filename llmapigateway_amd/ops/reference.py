@@ -1,0 +1,175 @@
+"""Pure-PyTorch fp32 reference implementations of every engine op.
+
+These are the *semantic contract* for the HIP kernels in csrc/: GPU numerics
+tests compare each gfx950 kernel against these (run in fp32) per-shape, and
+the CPU engine path runs on them directly so the whole gateway + engine
+stack is testable without a GPU.
+
+Shapes (T = total tokens in a varlen batch, B = sequences, D = head dim):
+- rmsnorm:            x [T, H]          -> y [T, H]
+- rmsnorm_residual:   x, residual       -> (y, x+residual)
+- rope_inplace:       q [T, Hq, D], k [T, Hkv, D], positions [T]
+- swiglu:             x [T, 2I]         -> silu(x[:, :I]) * x[:, I:]
+- kv_cache_write:     k/v [T, Hkv, D] -> cache [nblocks, Hkv, block, D]
+- attention_prefill:  varlen causal flash (q [T,Hq,D], k/v [T,Hkv,D], cu_seqlens)
+- attention_decode:   q [B, Hq, D] vs paged cache via block_tables
+- sample:             logits [B, V] (+ per-seq temperature, optional noise)
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import torch
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    dtype = x.dtype
+    xf = x.float()
+    var = xf.pow(2).mean(dim=-1, keepdim=True)
+    y = xf * torch.rsqrt(var + eps) * weight.float()
+    return y.to(dtype)
+
+
+def rmsnorm_residual(
+    x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    r = (x.float() + residual.float()).to(x.dtype)
+    return rmsnorm(r, weight, eps), r
+
+
+def build_rope_cache(
+    max_positions: int, head_dim: int, theta: float = 500000.0, device="cpu"
+) -> torch.Tensor:
+    """Return [max_positions, head_dim] fp32 cache: [cos(half) | sin(half)]."""
+    half = head_dim // 2
+    inv_freq = 1.0 / (theta ** (torch.arange(half, dtype=torch.float32, device=device) / half))
+    pos = torch.arange(max_positions, dtype=torch.float32, device=device)
+    ang = torch.outer(pos, inv_freq)  # [P, half]
+    return torch.cat([ang.cos(), ang.sin()], dim=-1).contiguous()
+
+
+def rope_inplace(
+    q: torch.Tensor, k: torch.Tensor, positions: torch.Tensor, cos_sin: torch.Tensor
+) -> None:
+    """Llama-style rotate-half RoPE applied in place to q and k."""
+    half = q.shape[-1] // 2
+    cs = cos_sin[positions]  # [T, D]
+    cos = cs[:, :half].unsqueeze(1)  # [T, 1, half]
+    sin = cs[:, half:].unsqueeze(1)
+    for t in (q, k):
+        tf = t.float()
+        x1, x2 = tf[..., :half], tf[..., half:]
+        t[..., :half] = (x1 * cos - x2 * sin).to(t.dtype)
+        t[..., half:] = (x2 * cos + x1 * sin).to(t.dtype)
+
+
+def swiglu(x: torch.Tensor) -> torch.Tensor:
+    inter = x.shape[-1] // 2
+    g = x[..., :inter].float()
+    u = x[..., inter:].float()
+    return (g * torch.sigmoid(g) * u).to(x.dtype)
+
+
+def kv_cache_write(
+    k: torch.Tensor,
+    v: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    slot_mapping: torch.Tensor,
+) -> None:
+    """Scatter new K/V rows into the paged cache.
+
+    cache layout: [num_blocks, num_kv_heads, block_size, head_dim];
+    slot_mapping[t] = block_id * block_size + block_offset.
+    """
+    block_size = k_cache.shape[2]
+    blocks = torch.div(slot_mapping, block_size, rounding_mode="floor")
+    offs = slot_mapping % block_size
+    k_cache[blocks, :, offs, :] = k.to(k_cache.dtype)
+    v_cache[blocks, :, offs, :] = v.to(v_cache.dtype)
+
+
+def attention_prefill(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    cu_seqlens: torch.Tensor,
+    scale: Optional[float] = None,
+    causal: bool = True,
+) -> torch.Tensor:
+    """Varlen causal attention with GQA over fresh (non-cached) K/V."""
+    T, Hq, D = q.shape
+    Hkv = k.shape[1]
+    group = Hq // Hkv
+    if scale is None:
+        scale = 1.0 / math.sqrt(D)
+    out = torch.empty_like(q)
+    cu = cu_seqlens.tolist()
+    for i in range(len(cu) - 1):
+        s, e = cu[i], cu[i + 1]
+        qi = q[s:e].float()  # [L, Hq, D]
+        ki = k[s:e].float().repeat_interleave(group, dim=1)
+        vi = v[s:e].float().repeat_interleave(group, dim=1)
+        scores = torch.einsum("qhd,khd->hqk", qi, ki) * scale
+        if causal:
+            L = e - s
+            mask = torch.triu(torch.ones(L, L, dtype=torch.bool, device=q.device), diagonal=1)
+            scores.masked_fill_(mask, float("-inf"))
+        p = torch.softmax(scores, dim=-1)
+        out[s:e] = torch.einsum("hqk,khd->qhd", p, vi).to(q.dtype)
+    return out
+
+
+def attention_decode(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,
+    context_lens: torch.Tensor,
+    scale: Optional[float] = None,
+) -> torch.Tensor:
+    """Single-token decode attention over the paged KV cache."""
+    B, Hq, D = q.shape
+    Hkv = k_cache.shape[1]
+    block_size = k_cache.shape[2]
+    group = Hq // Hkv
+    if scale is None:
+        scale = 1.0 / math.sqrt(D)
+    out = torch.empty_like(q)
+    for b in range(B):
+        L = int(context_lens[b])
+        nblocks = (L + block_size - 1) // block_size
+        blocks = block_tables[b, :nblocks].long()
+        # gather [L, Hkv, D]
+        kk = k_cache[blocks].permute(0, 2, 1, 3).reshape(-1, Hkv, D)[:L].float()
+        vv = v_cache[blocks].permute(0, 2, 1, 3).reshape(-1, Hkv, D)[:L].float()
+        kk = kk.repeat_interleave(group, dim=1)  # [L, Hq, D]
+        vv = vv.repeat_interleave(group, dim=1)
+        qb = q[b].float()  # [Hq, D]
+        scores = torch.einsum("hd,khd->hk", qb, kk) * scale
+        p = torch.softmax(scores, dim=-1)
+        out[b] = torch.einsum("hk,khd->hd", p, vv).to(q.dtype)
+    return out
+
+
+def sample(
+    logits: torch.Tensor,
+    temperature: torch.Tensor,
+    noise: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """Greedy when temperature<=0, else Gumbel-max sampling.
+
+    temperature: [B] fp32; noise: [B, V] uniform(0,1) — supplied by the
+    caller so GPU and CPU paths can be compared with identical randomness.
+    """
+    B, V = logits.shape
+    lf = logits.float()
+    temps = temperature.view(B, 1).float()
+    greedy = temps <= 0
+    if noise is None or bool(greedy.all()):
+        return lf.argmax(dim=-1)
+    gumbel = -torch.log(-torch.log(noise.float().clamp_min(1e-20)).clamp_min(1e-20))
+    scored = torch.where(greedy, lf, lf / temps.clamp_min(1e-6) + gumbel)
+    return scored.argmax(dim=-1)

@@ -1,0 +1,187 @@
+"""End-to-end gateway tests with LOCAL engine providers on CPU (tiny model).
+
+This exercises the full BASELINE configs[1..3] plumbing without a GPU:
+local provider serving, 2-provider fallback with injected failures,
+rotation across engine replicas, SSE streaming with first-chunk-error
+semantics, and usage accounting of engine-served requests.
+"""
+
+import json
+
+import pytest
+from fastapi.testclient import TestClient
+
+from llmapigateway_amd.config.settings import Settings
+from llmapigateway_amd.gateway.app import create_app
+
+PROVIDERS = """
+[
+    { "engine0": { "baseUrl": "local://tiny-llama?device=0", "apikey": "" } },
+    { "engine1": { "baseUrl": "local://tiny-llama?device=1", "apikey": "" } },
+    { "flaky-engine": {
+        "baseUrl": "local://tiny-llama?device=0",
+        "apikey": "",
+        "engine": { "model": "tiny-llama", "fail_requests": 2 }
+    } }
+]
+"""
+
+RULES = """
+[
+    { "gateway_model_name": "local/simple",
+      "fallback_models": [ { "provider": "engine0", "model": "tiny-llama" } ] },
+    { "gateway_model_name": "local/fallback",
+      "fallback_models": [
+          { "provider": "flaky-engine", "model": "tiny-llama" },
+          { "provider": "engine1", "model": "tiny-llama" } ] },
+    { "gateway_model_name": "local/retry",
+      "fallback_models": [
+          { "provider": "flaky-engine", "model": "tiny-llama",
+            "retry_count": 3, "retry_delay": 1 } ] },
+    { "gateway_model_name": "local/rotate", "rotate_models": true,
+      "fallback_models": [
+          { "provider": "engine0", "model": "tiny-llama" },
+          { "provider": "engine1", "model": "tiny-llama" } ] }
+]
+"""
+
+
+@pytest.fixture
+def client(tmp_path):
+    (tmp_path / "providers.json").write_text(PROVIDERS)
+    (tmp_path / "models_fallback_rules.json").write_text(RULES)
+    settings = Settings(fallback_provider="engine0")
+    app = create_app(
+        settings=settings,
+        providers_path=str(tmp_path / "providers.json"),
+        fallback_rules_path=str(tmp_path / "models_fallback_rules.json"),
+        db_dir=str(tmp_path / "db"),
+        log_dir=str(tmp_path / "logs"),
+    )
+    with TestClient(app) as c:
+        yield c
+
+
+def chat(client, model, stream=False, max_tokens=6, **kw):
+    return client.post(
+        "/v1/chat/completions",
+        json={
+            "model": model,
+            "messages": [{"role": "user", "content": "write a story"}],
+            "stream": stream,
+            "max_tokens": max_tokens,
+            "ignore_eos": True,
+            **kw,
+        },
+    )
+
+
+def test_local_nonstreaming(client):
+    r = chat(client, "local/simple")
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert body["object"] == "chat.completion"
+    assert isinstance(body["choices"][0]["message"]["content"], str)
+    assert body["usage"]["completion_tokens"] == 6
+    assert body["usage"]["prompt_tokens"] > 0
+
+
+def test_local_streaming(client):
+    r = chat(client, "local/simple", stream=True)
+    assert r.status_code == 200, r.text
+    assert r.headers["content-type"].startswith("text/event-stream")
+    frames = [f for f in r.content.decode().split("\n\n") if f.startswith("data: ")]
+    assert frames[-1] == "data: [DONE]"
+    objs = [json.loads(f[6:]) for f in frames if f.startswith("data: {")]
+    content = "".join(
+        c["delta"].get("content", "") for o in objs for c in o["choices"]
+    )
+    assert len(content) > 0
+    # final chunk carries usage
+    assert objs[-1]["usage"]["completion_tokens"] == 6
+
+
+def test_local_fallback_on_injected_failure(client):
+    # flaky-engine fails its first 2 requests -> engine1 serves them
+    r1 = chat(client, "local/fallback")
+    assert r1.status_code == 200
+    r2 = chat(client, "local/fallback", stream=True)
+    assert r2.status_code == 200
+    # third request: flaky-engine works now
+    r3 = chat(client, "local/fallback")
+    assert r3.status_code == 200
+
+
+def test_local_retry_until_success(client):
+    # same flaky provider, no fallback, retries instead
+    client.post(
+        "/v1/admin/engines/flaky-engine/failures",
+        json={"fail_requests": 2},
+    )
+    r = chat(client, "local/retry")
+    assert r.status_code == 200
+
+
+def test_runtime_failure_injection_all_fail(client):
+    assert (
+        client.post(
+            "/v1/admin/engines/engine0/failures", json={"fail_rate": 1.0}
+        ).status_code
+        == 200
+    )
+    r = chat(client, "local/simple")
+    assert r.status_code == 503
+    assert "Injected failure" in r.json()["detail"]
+    # reset
+    client.post("/v1/admin/engines/engine0/failures", json={"fail_rate": 0.0})
+    assert chat(client, "local/simple").status_code == 200
+
+
+def test_streaming_failure_before_first_byte(client):
+    client.post("/v1/admin/engines/engine0/failures", json={"fail_requests": 1})
+    r = chat(client, "local/simple", stream=True)
+    # the only provider fails -> clean 503, no partial SSE bytes
+    assert r.status_code == 503
+    r = chat(client, "local/simple", stream=True)
+    assert r.status_code == 200
+
+
+def test_rotation_across_engines(client):
+    for _ in range(4):
+        assert chat(client, "local/rotate").status_code == 200
+    stats = client.get("/v1/api/engine-stats").json()["engines"]
+    assert len(stats) >= 1
+    total_reqs = sum(s["requests"] for s in stats)
+    assert total_reqs >= 4
+
+
+def test_usage_accounting_local_streaming(client):
+    chat(client, "local/simple", stream=True)
+    db = client.app.state.usage_db
+    assert db.get_total_records_count() >= 1
+    rec = db.get_latest_usage_records(1)[0]
+    assert rec["provider"] == "engine0"
+    assert rec["completion_tokens"] == 6
+
+
+def test_engine_stats_endpoint(client):
+    chat(client, "local/simple")
+    stats = client.get("/v1/api/engine-stats").json()["engines"]
+    assert stats and stats[0]["model"] == "tiny-llama"
+    assert stats[0]["kv_blocks_total"] > 0
+
+
+def test_deterministic_greedy_same_prompt(client):
+    r1 = chat(client, "local/simple")
+    r2 = chat(client, "local/simple")
+    assert (
+        r1.json()["choices"][0]["message"]["content"]
+        == r2.json()["choices"][0]["message"]["content"]
+    )
+
+
+def test_local_models_listed(client):
+    data = client.get("/v1/models").json()["data"]
+    ids = [m["id"] for m in data]
+    assert "local/simple" in ids
+    assert "tiny-llama" in ids  # engine presets via local fallback provider
