@@ -1,0 +1,168 @@
+"""Serving benchmark: req/sec + p50 TTFT for Llama-3-8B bf16 (BASELINE.json).
+
+One rank per GPU (weak scaling: each rank runs an identical engine replica —
+the gateway's rotation mode, BASELINE configs[3]). A *step* is one wave of
+``--batch`` synthetic requests (random-init weights, random token prompts of
+--prompt-len, greedy decode of --gen-tokens) driven to completion through
+the continuous-batching engine. Timed region: K steps bracketed by a
+barrier + torch.cuda.synchronize on both sides; value is the whole-job
+req/sec aggregated over all ranks (max rank time).
+
+Launch (the driver does this for N>1):
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 --master-port P bench.py --gpus N --steps K --warmup W
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import statistics
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from llmapigateway_amd.engine import LLMEngine, EngineRequest, SamplingParams  # noqa: E402
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=3)
+    p.add_argument("--warmup", type=int, default=1)
+    p.add_argument("--model", default="llama-3-8b")
+    p.add_argument("--batch", type=int, default=64, help="requests per step per GPU")
+    p.add_argument("--prompt-len", type=int, default=128)
+    p.add_argument("--gen-tokens", type=int, default=64)
+    p.add_argument("--kv-blocks", type=int, default=None)
+    return p.parse_args()
+
+
+def make_prompts(batch: int, prompt_len: int, vocab: int, seed: int):
+    g = torch.Generator().manual_seed(seed)
+    toks = torch.randint(3, min(vocab, 4096), (batch, prompt_len), generator=g)
+    return [t.tolist() for t in toks]
+
+
+def run_wave(engine: LLMEngine, prompts, gen_tokens: int):
+    t0 = time.monotonic()
+    reqs = [
+        EngineRequest(p, SamplingParams(max_tokens=gen_tokens, ignore_eos=True))
+        for p in prompts
+    ]
+    for r in reqs:
+        engine.add_request(r)
+    while any(r.state in ("waiting", "running") for r in reqs):
+        engine.step()
+    if engine.device.type == "cuda":
+        torch.cuda.synchronize(engine.device)
+    ttfts = [
+        (r.first_token_time - t0) * 1000.0 for r in reqs if r.first_token_time
+    ]
+    assert all(r.state == "finished" for r in reqs)
+    assert all(len(r.out_ids) == gen_tokens for r in reqs)
+    return ttfts
+
+
+def main():
+    args = parse_args()
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    have_cuda = torch.cuda.is_available()
+    device = f"cuda:{local_rank}" if have_cuda else "cpu"
+
+    dist = None
+    if world_size > 1:
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        backend = "nccl" if have_cuda else "gloo"
+        dist.init_process_group(backend=backend)
+        if have_cuda:
+            torch.cuda.set_device(local_rank)
+
+    dtype = torch.bfloat16 if have_cuda else torch.float32
+    engine = LLMEngine(
+        model=args.model,
+        device=device,
+        dtype=dtype,
+        num_blocks=args.kv_blocks
+        if args.kv_blocks
+        else (None if have_cuda else 512),
+        max_batch_size=args.batch,
+        max_model_len=args.prompt_len + args.gen_tokens + 8,
+        seed=1234 + rank,
+    )
+    vocab = engine.full_config.vocab_size
+
+    def barrier_sync():
+        if dist is not None:
+            dist.barrier()
+        if have_cuda:
+            torch.cuda.synchronize(engine.device)
+
+    # warmup
+    for w in range(args.warmup):
+        run_wave(engine, make_prompts(args.batch, args.prompt_len, vocab, 1000 + w), args.gen_tokens)
+
+    barrier_sync()
+    t_start = time.monotonic()
+    ttfts = []
+    for s in range(args.steps):
+        ttfts += run_wave(
+            engine, make_prompts(args.batch, args.prompt_len, vocab, 2000 + s + rank * 7919),
+            args.gen_tokens,
+        )
+    barrier_sync()
+    elapsed = time.monotonic() - t_start
+
+    # max over ranks
+    if dist is not None:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        if have_cuda:
+            t = t.to(engine.device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        total_reqs = args.batch * args.steps * world_size
+        req_per_sec = total_reqs / elapsed
+        result = {
+            "metric": "req/sec",
+            "value": round(req_per_sec, 3),
+            "unit": "req/s",
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed * 1000.0 / args.steps, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if have_cuda else "fp32",
+            "data": "synthetic (random token prompts, random-init weights)",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch * world_size,
+                "seq_len": args.prompt_len + args.gen_tokens,
+                "prompt_len": args.prompt_len,
+                "gen_tokens": args.gen_tokens,
+                "parallelism": f"dp{world_size}",
+                "p50_ttft_ms": round(statistics.median(ttfts), 2) if ttfts else None,
+                "tokens_per_sec": round(
+                    total_reqs * (args.prompt_len + args.gen_tokens) / elapsed, 1
+                ),
+            },
+        }
+        print(json.dumps(result))
+
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

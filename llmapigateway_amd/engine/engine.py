@@ -245,6 +245,9 @@ class LLMEngine:
             logits_idx.append(cu[-1] - 1)
             max_len = max(max_len, L)
 
+        tile_seq, tile_off = ops.build_prefill_tiles(
+            [len(r.prompt_ids) for r in reqs], device
+        )
         batch = ForwardBatch(
             kind="prefill",
             token_ids=torch.tensor(token_ids, dtype=torch.long, device=device),
@@ -252,6 +255,8 @@ class LLMEngine:
             slot_mapping=torch.tensor(slots, dtype=torch.long, device=device),
             cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=device),
             max_seqlen=max_len,
+            tile_seq=tile_seq,
+            tile_off=tile_off,
             logits_indices=torch.tensor(logits_idx, dtype=torch.long, device=device),
         )
         logits = self.model.forward(batch, self.kv.k_caches, self.kv.v_caches)

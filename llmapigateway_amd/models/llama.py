@@ -41,6 +41,8 @@ class ForwardBatch:
     # prefill:
     cu_seqlens: Optional[torch.Tensor] = None   # [B+1] int32
     max_seqlen: int = 0
+    tile_seq: Optional[torch.Tensor] = None     # [ntiles] int32 (GPU kernel tile map)
+    tile_off: Optional[torch.Tensor] = None
     # decode:
     block_tables: Optional[torch.Tensor] = None  # [B, max_blocks] int32
     context_lens: Optional[torch.Tensor] = None  # [B] int32
@@ -73,11 +75,18 @@ class LlamaModel:
     # ---- weights ----
     def _init_weights(self, seed: int) -> None:
         c = self.config
-        gen = torch.Generator(device="cpu").manual_seed(seed)
         hidden = c.hidden_size
+        # tiny models init on CPU (deterministic across devices, for tests);
+        # big models init directly on the GPU (CPU RNG would take minutes at 8B+)
+        approx_params = (
+            2 * self.full_config.vocab_size * hidden
+            + c.num_layers * hidden * (c.q_size + 2 * c.kv_size + c.q_size + 3 * c.intermediate_size) // 1
+        )
+        init_device = torch.device("cpu") if approx_params < 10**8 else self.device
+        gen = torch.Generator(device=init_device).manual_seed(seed)
 
         def mk(rows: int, cols: int, std: float) -> torch.Tensor:
-            w = torch.empty(rows, cols, dtype=torch.float32)
+            w = torch.empty(rows, cols, dtype=torch.float32, device=init_device)
             w.normal_(0.0, std, generator=gen)
             return w.to(self.dtype).to(self.device)
 
@@ -141,7 +150,8 @@ class LlamaModel:
 
             if batch.kind == "prefill":
                 attn = ops.attention_prefill(
-                    q, k, v, batch.cu_seqlens, batch.max_seqlen, self.scale
+                    q, k, v, batch.cu_seqlens, batch.max_seqlen, self.scale,
+                    tile_seq=batch.tile_seq, tile_off=batch.tile_off,
                 )
             else:
                 attn = ops.attention_decode(

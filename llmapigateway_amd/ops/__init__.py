@@ -19,9 +19,9 @@ from .reference import build_rope_cache  # re-export (host-side table builder)
 _C = None
 _C_ERR: Optional[str] = None
 try:
-    from . import _C as _C_mod  # type: ignore
+    import importlib
 
-    _C = _C_mod
+    _C = importlib.import_module("._C", __name__)
 except ImportError as e:  # extension not built (CPU-only envs are fine)
     _C_ERR = str(e)
 
@@ -92,6 +92,22 @@ def kv_cache_write(
     reference.kv_cache_write(k, v, k_cache, v_cache, slot_mapping)
 
 
+QTILE = 64  # q rows per prefill workgroup (must match attention_prefill.hip)
+
+
+def build_prefill_tiles(seqlens, device) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Host-side tile map for the prefill kernel: one entry per 64-row q tile."""
+    tile_seq, tile_off = [], []
+    for i, L in enumerate(seqlens):
+        for off in range(0, int(L), QTILE):
+            tile_seq.append(i)
+            tile_off.append(off)
+    return (
+        torch.tensor(tile_seq, dtype=torch.int32, device=device),
+        torch.tensor(tile_off, dtype=torch.int32, device=device),
+    )
+
+
 def attention_prefill(
     q: torch.Tensor,
     k: torch.Tensor,
@@ -100,12 +116,22 @@ def attention_prefill(
     max_seqlen: int,
     scale: Optional[float] = None,
     causal: bool = True,
+    tile_seq: Optional[torch.Tensor] = None,
+    tile_off: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     if scale is None:
         scale = float(q.shape[-1]) ** -0.5
     if q.is_cuda:
+        if not causal:
+            raise NotImplementedError("GPU prefill kernel is causal-only")
+        if tile_seq is None or tile_off is None:
+            cu = cu_seqlens.cpu().tolist()
+            seqlens = [cu[i + 1] - cu[i] for i in range(len(cu) - 1)]
+            tile_seq, tile_off = build_prefill_tiles(seqlens, q.device)
         out = torch.empty_like(q)
-        _native().attention_prefill(out, q, k, v, cu_seqlens, int(max_seqlen), float(scale), causal)
+        _native().attention_prefill(
+            out, q, k, v, cu_seqlens.int(), tile_seq, tile_off, float(scale)
+        )
         return out
     return reference.attention_prefill(q, k, v, cu_seqlens, scale, causal)
 

@@ -1,0 +1,204 @@
+// Paged decode attention for gfx950 (GQA, bf16 KV cache).
+//
+// Replaces the per-token generation the reference gateway delegated to its
+// remote providers (SURVEY.md §2b "decode attention kernel"). Memory-bound:
+// the whole KV history of each sequence is streamed once per step.
+//
+// Geometry: one workgroup per (sequence, kv-head); 4 waves; the G = Hq/Hkv
+// query heads of the group share the streamed K/V (GQA bandwidth saving).
+// Each wave owns key chunks of 64 positions (chunk c -> wave c%4):
+//   phase A: lane = one key position; the lane streams that key's 128-dim
+//            row (16 B vector loads, L1-resident across the 16 loads) and
+//            dots it against q (LDS-broadcast reads);
+//   phase B: lanes switch to dim ownership (lane d owns dims 2d, 2d+1) and
+//            stream V rows coalesced (64 lanes x 4 B = a 256 B row).
+// Online softmax per wave, flash-style cross-wave combine in LDS at the end.
+//
+// Cache layout: [num_blocks, Hkv, block_size, D]; block_size any power of
+// two <= 64 handled via the per-lane block-table lookup.
+
+#include "common.h"
+
+#define MAX_G 8          // max query heads per kv head (llama-3: 4, 70B TP1: 8)
+#define NWAVES 4
+#define DECODE_D 128
+
+__launch_bounds__(NWAVES* WAVE_SIZE)
+__global__ void attention_decode_kernel(
+    bf16* __restrict__ out,                 // [B, Hq, D]
+    const bf16* __restrict__ q,             // [B, Hq, D]
+    const bf16* __restrict__ k_cache,       // [NB, Hkv, BS, D]
+    const bf16* __restrict__ v_cache,
+    const int* __restrict__ block_tables,   // [B, max_blocks]
+    const int* __restrict__ context_lens,   // [B]
+    float scale,
+    int Hq,
+    int Hkv,
+    int block_size,
+    int max_blocks,
+    int64_t q_stride) {
+    const int seq = blockIdx.x;
+    const int kvh = blockIdx.y;
+    const int G = Hq / Hkv;
+    const int D = DECODE_D;
+    const int L = context_lens[seq];
+    if (L <= 0) return;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE_SIZE - 1);
+    const int wave = tid >> 6;
+
+    // ---- LDS ----
+    __shared__ float q_lds[MAX_G][DECODE_D];
+    __shared__ float p_lds[NWAVES][MAX_G][WAVE_SIZE];
+    __shared__ float m_lds[NWAVES][MAX_G];
+    __shared__ float l_lds[NWAVES][MAX_G];
+    __shared__ float acc_lds[NWAVES][MAX_G][DECODE_D];
+    __shared__ int bt_lds[1024];  // cached block table (<= 64K tokens at BS=64)
+
+    // stage q group into LDS as fp32 (scaled once here)
+    for (int i = tid; i < G * D; i += blockDim.x) {
+        const int g = i / D, d = i % D;
+        q_lds[g][d] =
+            bf2f(q[(size_t)seq * q_stride + (size_t)(kvh * G + g) * D + d]) * scale;
+    }
+    const int nblocks = (L + block_size - 1) / block_size;
+    for (int i = tid; i < nblocks && i < 1024; i += blockDim.x)
+        bt_lds[i] = block_tables[(size_t)seq * max_blocks + i];
+    __syncthreads();
+    // very long contexts (>1024 blocks) read the tail of the table from L2
+    const int* __restrict__ bt_global = block_tables + (size_t)seq * max_blocks;
+#define BT(idx) ((idx) < 1024 ? bt_lds[(idx)] : bt_global[(idx)])
+
+    float m[MAX_G], l[MAX_G], acc[MAX_G][2];
+#pragma unroll
+    for (int g = 0; g < MAX_G; ++g) {
+        m[g] = -INFINITY;
+        l[g] = 0.f;
+        acc[g][0] = acc[g][1] = 0.f;
+    }
+
+    const int nchunks = (L + WAVE_SIZE - 1) / WAVE_SIZE;
+    for (int c = wave; c < nchunks; c += NWAVES) {
+        const int pos = c * WAVE_SIZE + lane;
+        const bool valid = pos < L;
+
+        // --- phase A: lane = key; dot q . k ---
+        float s[MAX_G];
+#pragma unroll
+        for (int g = 0; g < MAX_G; ++g) s[g] = -INFINITY;
+        int kblock = 0, koff = 0;
+        if (valid) {
+            kblock = BT(pos / block_size);
+            koff = pos % block_size;
+            const uint4* krow = reinterpret_cast<const uint4*>(
+                k_cache + (((size_t)kblock * Hkv + kvh) * block_size + koff) * D);
+#pragma unroll
+            for (int g = 0; g < MAX_G; ++g) s[g] = 0.f;
+            for (int i = 0; i < D / 8; ++i) {
+                uint4 kv8 = krow[i];
+                float kf[8];
+                unpack2(kv8.x, kf[0], kf[1]);
+                unpack2(kv8.y, kf[2], kf[3]);
+                unpack2(kv8.z, kf[4], kf[5]);
+                unpack2(kv8.w, kf[6], kf[7]);
+                for (int g = 0; g < G; ++g) {
+#pragma unroll
+                    for (int j = 0; j < 8; ++j)
+                        s[g] += q_lds[g][i * 8 + j] * kf[j];
+                }
+            }
+        }
+
+        // --- online softmax update (per wave) ---
+        float chunk_sum[MAX_G];
+        for (int g = 0; g < G; ++g) {
+            const float cmax = wave_reduce_max(s[g]);
+            const float m_new = fmaxf(m[g], cmax);
+            float p = 0.f;
+            if (valid && m_new != -INFINITY) p = __expf(s[g] - m_new);
+            const float factor = (m[g] == -INFINITY) ? 0.f : __expf(m[g] - m_new);
+            chunk_sum[g] = wave_reduce_sum(p);
+            l[g] = l[g] * factor + chunk_sum[g];
+            acc[g][0] *= factor;
+            acc[g][1] *= factor;
+            m[g] = m_new;
+            p_lds[wave][g][lane] = p;
+        }
+        // wave-synchronous LDS use: no barrier needed (private per wave)
+
+        // --- phase B: lane = dim pair; stream V rows ---
+        const int chunk_keys = min(WAVE_SIZE, L - c * WAVE_SIZE);
+        for (int j = 0; j < chunk_keys; ++j) {
+            const int p2 = c * WAVE_SIZE + j;
+            const int vb = BT(p2 / block_size);
+            const int vo = p2 % block_size;
+            const uint32_t vpair = reinterpret_cast<const uint32_t*>(
+                v_cache + (((size_t)vb * Hkv + kvh) * block_size + vo) * D)[lane];
+            float v0, v1;
+            unpack2(vpair, v0, v1);
+            for (int g = 0; g < G; ++g) {
+                const float pj = p_lds[wave][g][j];
+                acc[g][0] = fmaf(pj, v0, acc[g][0]);
+                acc[g][1] = fmaf(pj, v1, acc[g][1]);
+            }
+        }
+    }
+
+    // ---- cross-wave flash combine ----
+    for (int g = 0; g < G; ++g) {
+        if (lane == 0) {
+            m_lds[wave][g] = m[g];
+            l_lds[wave][g] = l[g];
+        }
+        acc_lds[wave][g][2 * lane] = acc[g][0];
+        acc_lds[wave][g][2 * lane + 1] = acc[g][1];
+    }
+    __syncthreads();
+
+    if (wave == 0) {
+        for (int g = 0; g < G; ++g) {
+            float gm = -INFINITY;
+#pragma unroll
+            for (int w = 0; w < NWAVES; ++w) gm = fmaxf(gm, m_lds[w][g]);
+            float gl = 0.f;
+            float f[NWAVES];
+#pragma unroll
+            for (int w = 0; w < NWAVES; ++w) {
+                f[w] = (m_lds[w][g] == -INFINITY) ? 0.f : __expf(m_lds[w][g] - gm);
+                gl += f[w] * l_lds[w][g];
+            }
+            const float inv_l = (gl > 0.f) ? 1.f / gl : 0.f;
+            float o0 = 0.f, o1 = 0.f;
+#pragma unroll
+            for (int w = 0; w < NWAVES; ++w) {
+                o0 += f[w] * acc_lds[w][g][2 * lane];
+                o1 += f[w] * acc_lds[w][g][2 * lane + 1];
+            }
+            reinterpret_cast<uint32_t*>(
+                out + (size_t)seq * Hq * D + (size_t)(kvh * G + g) * D)[lane] =
+                pack2(o0 * inv_l, o1 * inv_l);
+        }
+    }
+}
+
+extern "C" hipError_t launch_attention_decode(
+    void* out, const void* q, const void* k_cache, const void* v_cache,
+    const int* block_tables, const int* context_lens, float scale, int B,
+    int Hq, int Hkv, int block_size, int max_blocks, int D, int64_t q_stride,
+    hipStream_t stream) {
+    if (D != DECODE_D && D != 64) return hipErrorInvalidValue;
+    if (Hq / Hkv > MAX_G || Hq % Hkv != 0) return hipErrorInvalidValue;
+    dim3 grid(B, Hkv);
+    dim3 block(NWAVES * WAVE_SIZE);
+    if (D == DECODE_D) {
+        attention_decode_kernel<<<grid, block, 0, stream>>>(
+            (bf16*)out, (const bf16*)q, (const bf16*)k_cache,
+            (const bf16*)v_cache, block_tables, context_lens, scale, Hq, Hkv,
+            block_size, max_blocks, q_stride);
+    } else {
+        return hipErrorNotSupported;  // head_dim 64 engine presets run D=128 models on GPU
+    }
+    HIP_CHECK_LAST();
+    return hipSuccess;
+}

@@ -1,0 +1,222 @@
+#include "hip/hip_runtime.h"
+// Flash-style causal prefill attention for gfx950 (MFMA bf16, LDS-tiled).
+//
+// Produces the prompt phase (TTFT) the reference delegated upstream
+// (SURVEY.md §2b "prefill attention kernel"). Varlen batch with GQA.
+//
+// Geometry: one workgroup per (64-row q tile, q head); 4 waves, each wave
+// owns 16 q rows. K/V tiles of 64 keys staged in LDS (V transposed at
+// staging so the PV B-fragment reads are contiguous 16 B ds_read_b128).
+// MFMA: v_mfma_f32_16x16x32_bf16 throughout —
+//   A fragment: lane l holds A[row = l&15][k = (l>>4)*8 + j], j=0..7
+//   B fragment: lane l holds B[k = (l>>4)*8 + j][col = l&15]
+//   C fragment: lane l holds C[row = (l>>4)*4 + r][col = l&15], r=0..3
+// (cdna_hip_programming.md §3; transpose-detecting numerics tests in
+// tests/test_ops_gpu.py guard the mapping).
+//
+// Online softmax per 4-row group (flash rescaling with -1e30 sentinel so
+// fully-masked rows stay NaN-free).
+
+#include "common.h"
+
+#define PF_D 128
+#define QTILE 64         // q rows per workgroup
+#define KVTILE 64        // keys per LDS tile
+#define PF_WAVES 4
+#define NEG_INF (-1e30f)
+
+typedef __attribute__((__vector_size__(8 * sizeof(short)))) short bf16x8;
+typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4;
+
+__device__ __forceinline__ f32x4 mfma16(bf16x8 a, bf16x8 b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+
+__launch_bounds__(PF_WAVES* WAVE_SIZE)
+__global__ void attention_prefill_kernel(
+    bf16* __restrict__ out,             // [T, Hq, D]
+    const bf16* __restrict__ q,         // [T, Hq, D] (rows strided)
+    const bf16* __restrict__ k,         // [T, Hkv, D]
+    const bf16* __restrict__ v,         // [T, Hkv, D]
+    const int* __restrict__ cu_seqlens, // [B+1]
+    const int* __restrict__ tile_seq,   // [ntiles] seq index of each tile
+    const int* __restrict__ tile_off,   // [ntiles] first q row (local) of tile
+    float scale,
+    int Hq,
+    int Hkv,
+    int64_t q_stride,
+    int64_t k_stride,
+    int64_t v_stride) {
+    const int tile = blockIdx.x;
+    const int head = blockIdx.y;
+    const int kvh = head / (Hq / Hkv);
+    const int seq = tile_seq[tile];
+    const int tile0 = tile_off[tile];
+    const int seq_start = cu_seqlens[seq];
+    const int seq_len = cu_seqlens[seq + 1] - seq_start;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE_SIZE - 1);
+    const int wave = tid >> 6;
+    const int lrow = lane & 15;        // A-fragment row / C col
+    const int lk = lane >> 4;          // k-group within fragment
+
+    __shared__ bf16 k_lds[KVTILE][PF_D];
+    __shared__ bf16 vt_lds[PF_D][KVTILE];
+    __shared__ bf16 p_lds[PF_WAVES][16][KVTILE];
+
+    // ---- load this wave's Q fragments (A operand) straight from global ----
+    bf16x8 q_frag[4];
+    const int my_qrow = tile0 + wave * 16 + lrow;     // local row in seq
+    {
+        const bool rvalid = my_qrow < seq_len;
+        const size_t base = rvalid
+            ? ((size_t)(seq_start + my_qrow) * q_stride + (size_t)head * PF_D)
+            : ((size_t)seq_start * q_stride + (size_t)head * PF_D);
+#pragma unroll
+        for (int kc = 0; kc < 4; ++kc) {
+            const uint4 raw = *reinterpret_cast<const uint4*>(
+                q + base + kc * 32 + lk * 8);
+            q_frag[kc] = *reinterpret_cast<const bf16x8*>(&raw);
+        }
+    }
+
+    // softmax state per lane: 4 rows (r -> row (lane>>4)*4 + r)
+    float m_st[4], l_st[4];
+    f32x4 accO[8];  // 8 dim-subtiles x 4 rows
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        m_st[r] = NEG_INF;
+        l_st[r] = 0.f;
+    }
+#pragma unroll
+    for (int d = 0; d < 8; ++d) accO[d] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+    const int kv_limit = min(seq_len, tile0 + QTILE);  // causal: keys < limit
+    const int n_kv_tiles = (kv_limit + KVTILE - 1) / KVTILE;
+
+    for (int jt = 0; jt < n_kv_tiles; ++jt) {
+        const int key0 = jt * KVTILE;
+        const int keys_here = min(KVTILE, kv_limit - key0);
+
+        // ---- stage K tile and V^T tile (256 threads cooperative) ----
+        __syncthreads();
+        for (int i = tid; i < KVTILE * (PF_D / 8); i += PF_WAVES * WAVE_SIZE) {
+            const int kk = i / (PF_D / 8);
+            const int d0 = (i % (PF_D / 8)) * 8;
+            uint4 kraw = {0, 0, 0, 0}, vraw = {0, 0, 0, 0};
+            if (kk < keys_here) {
+                const size_t t = (size_t)(seq_start + key0 + kk);
+                kraw = *reinterpret_cast<const uint4*>(
+                    k + t * k_stride + (size_t)kvh * PF_D + d0);
+                vraw = *reinterpret_cast<const uint4*>(
+                    v + t * v_stride + (size_t)kvh * PF_D + d0);
+            }
+            *reinterpret_cast<uint4*>(&k_lds[kk][d0]) = kraw;
+            const bf16* v8 = reinterpret_cast<const bf16*>(&vraw);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) vt_lds[d0 + j][kk] = v8[j];
+        }
+        __syncthreads();
+
+        // ---- S = scale * Q K^T  (4 key-subtiles of 16) ----
+        f32x4 s[4];
+#pragma unroll
+        for (int sub = 0; sub < 4; ++sub) {
+            s[sub] = (f32x4){0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int kc = 0; kc < 4; ++kc) {
+                const bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+                    &k_lds[sub * 16 + lrow][kc * 32 + lk * 8]);
+                s[sub] = mfma16(q_frag[kc], bfrag, s[sub]);
+            }
+        }
+
+        // ---- mask + online softmax ----
+        float p_val[4][4];
+        float row_max[4], row_sum[4];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int qrow = tile0 + wave * 16 + lk * 4 + r;  // this lane's C row
+            float mx = NEG_INF;
+#pragma unroll
+            for (int sub = 0; sub < 4; ++sub) {
+                const int key = key0 + sub * 16 + lrow;       // this lane's C col
+                float val = s[sub][r] * scale;
+                if (key > qrow || key >= seq_len || qrow >= seq_len) val = NEG_INF;
+                p_val[sub][r] = val;
+                mx = fmaxf(mx, val);
+            }
+            row_max[r] = group16_reduce_max(mx);
+        }
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const float m_new = fmaxf(m_st[r], row_max[r]);
+            const float factor = (m_st[r] <= NEG_INF) ? 0.f : __expf(m_st[r] - m_new);
+            float sum = 0.f;
+#pragma unroll
+            for (int sub = 0; sub < 4; ++sub) {
+                const float pv = (m_new <= NEG_INF) ? 0.f : __expf(p_val[sub][r] - m_new);
+                p_val[sub][r] = pv;
+                sum += pv;
+            }
+            row_sum[r] = group16_reduce_sum(sum);
+            l_st[r] = l_st[r] * factor + row_sum[r];
+            m_st[r] = m_new;
+#pragma unroll
+            for (int d = 0; d < 8; ++d) accO[d][r] *= factor;
+        }
+
+        // ---- P -> LDS (bf16), per-wave private tile ----
+#pragma unroll
+        for (int sub = 0; sub < 4; ++sub)
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+                p_lds[wave][lk * 4 + r][sub * 16 + lrow] = f2bf(p_val[sub][r]);
+        // wave-synchronous LDS: the same wave reads it next, no barrier
+
+        // ---- O += P V  (8 dim-subtiles, 2 key-chunks of 32) ----
+#pragma unroll
+        for (int dsub = 0; dsub < 8; ++dsub) {
+#pragma unroll
+            for (int kc = 0; kc < 2; ++kc) {
+                const bf16x8 pa = *reinterpret_cast<const bf16x8*>(
+                    &p_lds[wave][lrow][kc * 32 + lk * 8]);
+                const bf16x8 vb = *reinterpret_cast<const bf16x8*>(
+                    &vt_lds[dsub * 16 + lrow][kc * 32 + lk * 8]);
+                accO[dsub] = mfma16(pa, vb, accO[dsub]);
+            }
+        }
+    }
+
+    // ---- epilogue: out[row][d] = acc / l ----
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        const int qrow = tile0 + wave * 16 + lk * 4 + r;
+        if (qrow >= seq_len || l_st[r] <= 0.f) continue;
+        const float inv_l = 1.f / l_st[r];
+        bf16* orow =
+            out + (size_t)(seq_start + qrow) * Hq * PF_D + (size_t)head * PF_D;
+#pragma unroll
+        for (int dsub = 0; dsub < 8; ++dsub)
+            orow[dsub * 16 + lrow] = f2bf(accO[dsub][r] * inv_l);
+    }
+}
+
+// Wait: accO[dsub][r] must use the PV C-layout row = lk*4 + r, col = lrow —
+// which matches the S layout used above (both are 16x16x32 C fragments).
+
+extern "C" hipError_t launch_attention_prefill(
+    void* out, const void* q, const void* k, const void* v,
+    const int* cu_seqlens, const int* tile_seq, const int* tile_off,
+    int ntiles, float scale, int Hq, int Hkv, int D, int64_t q_stride,
+    int64_t k_stride, int64_t v_stride, hipStream_t stream) {
+    if (D != PF_D) return hipErrorNotSupported;
+    dim3 grid(ntiles, Hq);
+    dim3 block(PF_WAVES * WAVE_SIZE);
+   hipLaunchKernelGGL(( attention_prefill_kernel), dim3(grid), dim3(block), 0, stream, 
+        (bf16*)out, (const bf16*)q, (const bf16*)k, (const bf16*)v, cu_seqlens,
+        tile_seq, tile_off, scale, Hq, Hkv, q_stride, k_stride, v_stride);
+    HIP_CHECK_LAST();
+    return hipSuccess;
+}

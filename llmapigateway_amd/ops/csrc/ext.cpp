@@ -1,0 +1,185 @@
+// Torch extension binding for the gfx950 kernels (built in-tree by
+// `python setup.py build_ext --inplace` with PYTORCH_ROCM_ARCH=gfx950).
+// The kernels themselves live in the .hip files (torch-free, raw pointers);
+// this file validates tensors, extracts strides and dispatches on the
+// current HIP stream.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include <vector>
+
+#include "kv_manager.cpp"
+
+extern "C" {
+hipError_t launch_rmsnorm(void*, const void*, const void*, float, int, int, hipStream_t);
+hipError_t launch_rmsnorm_residual(void*, const void*, void*, const void*, float, int, int, hipStream_t);
+hipError_t launch_rope(void*, void*, const int64_t*, const float*, int, int64_t, int64_t, int, int, int, hipStream_t);
+hipError_t launch_swiglu(void*, const void*, int, int, hipStream_t);
+hipError_t launch_kv_cache_write(const void*, const void*, void*, void*, const int64_t*, int, int64_t, int64_t, int, int, int, hipStream_t);
+hipError_t launch_attention_decode(void*, const void*, const void*, const void*, const int*, const int*, float, int, int, int, int, int, int, int64_t, hipStream_t);
+hipError_t launch_attention_prefill(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, float, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
+hipError_t launch_sample(int64_t*, const float*, const float*, const float*, int, int, hipStream_t);
+}
+
+namespace {
+
+#define CHECK_HIP(call)                                                        \
+    do {                                                                       \
+        hipError_t err = (call);                                               \
+        TORCH_CHECK(err == hipSuccess, "HIP kernel failed: ",                  \
+                    hipGetErrorString(err));                                   \
+    } while (0)
+
+hipStream_t current_stream() {
+    return at::hip::getCurrentHIPStream().stream();
+}
+
+void check_bf16(const torch::Tensor& t, const char* name) {
+    TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+    TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+}
+
+void rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor weight, double eps) {
+    check_bf16(x, "x");
+    check_bf16(out, "out");
+    TORCH_CHECK(x.is_contiguous() && out.is_contiguous() && weight.is_contiguous());
+    const int H = x.size(-1);
+    const int T = x.numel() / H;
+    CHECK_HIP(launch_rmsnorm(out.data_ptr(), x.data_ptr(), weight.data_ptr(),
+                             (float)eps, T, H, current_stream()));
+}
+
+void rmsnorm_residual(torch::Tensor out, torch::Tensor x, torch::Tensor residual,
+                      torch::Tensor weight, double eps) {
+    check_bf16(x, "x");
+    TORCH_CHECK(x.is_contiguous() && out.is_contiguous() && residual.is_contiguous());
+    const int H = x.size(-1);
+    const int T = x.numel() / H;
+    CHECK_HIP(launch_rmsnorm_residual(out.data_ptr(), x.data_ptr(),
+                                      residual.data_ptr(), weight.data_ptr(),
+                                      (float)eps, T, H, current_stream()));
+}
+
+void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
+                  torch::Tensor cos_sin) {
+    check_bf16(q, "q");
+    check_bf16(k, "k");
+    TORCH_CHECK(q.dim() == 3 && k.dim() == 3, "q/k must be [T, H, D]");
+    TORCH_CHECK(q.stride(2) == 1 && k.stride(2) == 1, "head_dim must be contiguous");
+    TORCH_CHECK(q.stride(1) == q.size(2) && k.stride(1) == k.size(2),
+                "heads must be contiguous within a row");
+    TORCH_CHECK(positions.scalar_type() == torch::kInt64);
+    TORCH_CHECK(cos_sin.scalar_type() == torch::kFloat32 && cos_sin.is_contiguous());
+    const int T = q.size(0);
+    CHECK_HIP(launch_rope(q.data_ptr(), k.data_ptr(),
+                          positions.data_ptr<int64_t>(),
+                          cos_sin.data_ptr<float>(), T, q.stride(0), k.stride(0),
+                          q.size(1), k.size(1), q.size(2), current_stream()));
+}
+
+void swiglu(torch::Tensor out, torch::Tensor x) {
+    check_bf16(x, "x");
+    TORCH_CHECK(x.is_contiguous() && out.is_contiguous());
+    const int I = out.size(-1);
+    TORCH_CHECK(x.size(-1) == 2 * I, "x last dim must be 2*I");
+    const int T = x.numel() / (2 * I);
+    CHECK_HIP(launch_swiglu(out.data_ptr(), x.data_ptr(), T, I, current_stream()));
+}
+
+void kv_cache_write(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
+                    torch::Tensor v_cache, torch::Tensor slot_mapping) {
+    check_bf16(k, "k");
+    check_bf16(k_cache, "k_cache");
+    TORCH_CHECK(k.dim() == 3 && k_cache.dim() == 4);
+    TORCH_CHECK(k.stride(2) == 1 && k.stride(1) == k.size(2));
+    TORCH_CHECK(v.stride(2) == 1 && v.stride(1) == v.size(2));
+    TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
+    TORCH_CHECK(slot_mapping.scalar_type() == torch::kInt64);
+    const int T = k.size(0);
+    const int Hkv = k_cache.size(1);
+    const int block_size = k_cache.size(2);
+    const int D = k_cache.size(3);
+    CHECK_HIP(launch_kv_cache_write(
+        k.data_ptr(), v.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+        slot_mapping.data_ptr<int64_t>(), T, k.stride(0), v.stride(0), Hkv,
+        block_size, D, current_stream()));
+}
+
+void attention_decode(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
+                      torch::Tensor v_cache, torch::Tensor block_tables,
+                      torch::Tensor context_lens, double scale) {
+    check_bf16(q, "q");
+    check_bf16(out, "out");
+    TORCH_CHECK(out.is_contiguous());
+    TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2));
+    TORCH_CHECK(block_tables.scalar_type() == torch::kInt32 && block_tables.is_contiguous());
+    TORCH_CHECK(context_lens.scalar_type() == torch::kInt32);
+    const int B = q.size(0);
+    const int Hq = q.size(1);
+    const int D = q.size(2);
+    const int Hkv = k_cache.size(1);
+    const int block_size = k_cache.size(2);
+    const int max_blocks = block_tables.size(1);
+    CHECK_HIP(launch_attention_decode(
+        out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+        block_tables.data_ptr<int>(), context_lens.data_ptr<int>(), (float)scale,
+        B, Hq, Hkv, block_size, max_blocks, D, q.stride(0), current_stream()));
+}
+
+void attention_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
+                       torch::Tensor v, torch::Tensor cu_seqlens,
+                       torch::Tensor tile_seq, torch::Tensor tile_off,
+                       double scale) {
+    check_bf16(q, "q");
+    check_bf16(out, "out");
+    TORCH_CHECK(out.is_contiguous());
+    TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2));
+    TORCH_CHECK(k.stride(2) == 1 && k.stride(1) == k.size(2));
+    TORCH_CHECK(v.stride(2) == 1 && v.stride(1) == v.size(2));
+    TORCH_CHECK(cu_seqlens.scalar_type() == torch::kInt32);
+    TORCH_CHECK(tile_seq.scalar_type() == torch::kInt32 &&
+                tile_off.scalar_type() == torch::kInt32);
+    const int ntiles = tile_seq.size(0);
+    CHECK_HIP(launch_attention_prefill(
+        out.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+        cu_seqlens.data_ptr<int>(), tile_seq.data_ptr<int>(),
+        tile_off.data_ptr<int>(), ntiles, (float)scale, q.size(1), k.size(1),
+        q.size(2), q.stride(0), k.stride(0), v.stride(0), current_stream()));
+}
+
+void sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temperature,
+            c10::optional<torch::Tensor> noise) {
+    TORCH_CHECK(logits.is_cuda() && logits.scalar_type() == torch::kFloat32 &&
+                logits.is_contiguous());
+    TORCH_CHECK(out.scalar_type() == torch::kInt64);
+    TORCH_CHECK(temperature.scalar_type() == torch::kFloat32);
+    const float* noise_ptr = nullptr;
+    if (noise.has_value() && noise->defined()) {
+        TORCH_CHECK(noise->is_contiguous() && noise->scalar_type() == torch::kFloat32);
+        noise_ptr = noise->data_ptr<float>();
+    }
+    CHECK_HIP(launch_sample(out.data_ptr<int64_t>(), logits.data_ptr<float>(),
+                            temperature.data_ptr<float>(), noise_ptr,
+                            logits.size(0), logits.size(1), current_stream()));
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("rmsnorm", &rmsnorm, "RMSNorm (gfx950)");
+    m.def("rmsnorm_residual", &rmsnorm_residual, "fused residual-add RMSNorm");
+    m.def("rope_inplace", &rope_inplace, "rotate-half RoPE in place");
+    m.def("swiglu", &swiglu, "fused SiLU-gate multiply");
+    m.def("kv_cache_write", &kv_cache_write, "paged KV cache scatter");
+    m.def("attention_decode", &attention_decode, "paged decode attention");
+    m.def("attention_prefill", &attention_prefill, "varlen causal prefill attention");
+    m.def("sample", &sample, "greedy / gumbel-max sampling");
+
+    pybind11::class_<BlockAllocator>(m, "BlockAllocator")
+        .def(pybind11::init<int64_t>())
+        .def("num_free", &BlockAllocator::num_free)
+        .def("allocate", &BlockAllocator::allocate)
+        .def("free", &BlockAllocator::free_blocks)
+        .def_property_readonly("num_blocks", &BlockAllocator::capacity);
+}

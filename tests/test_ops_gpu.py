@@ -1,0 +1,217 @@
+"""GPU numerics tests: every gfx950 HIP kernel vs the PyTorch fp32 reference
+(ops/reference.py) on the same bf16 inputs. Transpose-detecting inputs
+(asymmetric random) per cdna_hip_programming.md §5.4 rule 16."""
+
+import math
+
+import pytest
+import torch
+
+from llmapigateway_amd import ops
+from llmapigateway_amd.ops import reference
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+@pytest.fixture(autouse=True, scope="module")
+def _require_native():
+    assert torch.cuda.is_available(), "GPU required"
+    assert ops.have_native(), (
+        "HIP extension is not loaded on a GPU box — the native path MUST run"
+    )
+    torch.manual_seed(0)
+
+
+def to_f32(x):
+    return x.float().cpu()
+
+
+def max_rel_err(got, ref, eps=1e-3):
+    got, ref = to_f32(got), to_f32(ref)
+    return ((got - ref).abs() / (ref.abs() + eps)).max().item()
+
+
+# ---------- rmsnorm ----------
+
+@pytest.mark.parametrize("shape", [(7, 256), (64, 4096), (3, 8192), (256, 1024)])
+def test_rmsnorm(shape):
+    x = torch.randn(shape, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(shape[-1], dtype=torch.bfloat16, device=DEV)
+    got = ops.rmsnorm(x, w, 1e-5)
+    ref = reference.rmsnorm(x.cpu(), w.cpu(), 1e-5)
+    assert max_rel_err(got, ref) < 0.05
+
+
+def test_rmsnorm_residual():
+    x = torch.randn(33, 4096, dtype=torch.bfloat16, device=DEV)
+    res = torch.randn_like(x)
+    w = torch.randn(4096, dtype=torch.bfloat16, device=DEV)
+    res_cpu = res.cpu().clone()
+    y, new_res = ops.rmsnorm_residual(x, res, w, 1e-5)
+    y_ref, res_ref = reference.rmsnorm_residual(x.cpu(), res_cpu, w.cpu(), 1e-5)
+    assert max_rel_err(new_res, res_ref) < 0.05
+    assert max_rel_err(y, y_ref) < 0.05
+
+
+# ---------- rope ----------
+
+def test_rope_strided():
+    T, Hq, Hkv, D = 19, 8, 2, 128
+    # emulate the fused-qkv view (rows strided)
+    qkv = torch.randn(T, (Hq + 2 * Hkv) * D, dtype=torch.bfloat16, device=DEV)
+    q = qkv[:, : Hq * D].view(T, Hq, D)
+    k = qkv[:, Hq * D : (Hq + Hkv) * D].view(T, Hkv, D)
+    pos = torch.randint(0, 500, (T,), device=DEV)
+    cs = ops.build_rope_cache(512, D, 500000.0, device=DEV)
+    q_cpu, k_cpu = q.cpu().clone(), k.cpu().clone()
+    ops.rope_inplace(q, k, pos, cs)
+    reference.rope_inplace(q_cpu, k_cpu, pos.cpu(), cs.cpu())
+    assert max_rel_err(q, q_cpu) < 0.05
+    assert max_rel_err(k, k_cpu) < 0.05
+
+
+# ---------- swiglu ----------
+
+def test_swiglu():
+    x = torch.randn(37, 2 * 14336, dtype=torch.bfloat16, device=DEV)
+    got = ops.swiglu(x)
+    ref = reference.swiglu(x.cpu())
+    assert max_rel_err(got, ref) < 0.05
+
+
+# ---------- kv cache write ----------
+
+def test_kv_cache_write():
+    T, Hkv, BS, D, NB = 50, 8, 16, 128, 32
+    qkv = torch.randn(T, 3 * Hkv * D, dtype=torch.bfloat16, device=DEV)
+    k = qkv[:, : Hkv * D].view(T, Hkv, D)
+    v = qkv[:, Hkv * D : 2 * Hkv * D].view(T, Hkv, D)
+    kc = torch.zeros(NB, Hkv, BS, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.zeros_like(kc)
+    slots = torch.randperm(NB * BS, device=DEV)[:T]
+    kc_ref, vc_ref = kc.cpu().clone(), vc.cpu().clone()
+    ops.kv_cache_write(k, v, kc, vc, slots)
+    reference.kv_cache_write(k.cpu(), v.cpu(), kc_ref, vc_ref, slots.cpu())
+    assert torch.equal(kc.cpu(), kc_ref)
+    assert torch.equal(vc.cpu(), vc_ref)
+
+
+# ---------- prefill attention ----------
+
+@pytest.mark.parametrize(
+    "lens,Hq,Hkv",
+    [
+        ([64], 4, 4),          # exact one tile, MHA
+        ([128], 8, 2),         # GQA 4
+        ([1, 63, 64, 200], 8, 2),  # ragged varlen
+        ([300], 32, 8),        # llama-3-8b head config
+    ],
+)
+def test_attention_prefill(lens, Hq, Hkv):
+    D = 128
+    T = sum(lens)
+    q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device=DEV)
+    cu = [0]
+    for L in lens:
+        cu.append(cu[-1] + L)
+    cu_t = torch.tensor(cu, dtype=torch.int32, device=DEV)
+    got = ops.attention_prefill(q, k, v, cu_t, max(lens))
+    ref = reference.attention_prefill(q.cpu(), k.cpu(), v.cpu(), cu_t.cpu())
+    err = (to_f32(got) - to_f32(ref)).abs().max().item()
+    assert err < 0.05, f"max abs err {err}"
+
+
+def test_attention_prefill_strided_qkv():
+    # q/k/v as views into one fused projection row (the model's real layout)
+    D, Hq, Hkv, L = 128, 8, 2, 100
+    qkv = torch.randn(L, (Hq + 2 * Hkv) * D, dtype=torch.bfloat16, device=DEV)
+    q = qkv[:, : Hq * D].view(L, Hq, D)
+    k = qkv[:, Hq * D : (Hq + Hkv) * D].view(L, Hkv, D)
+    v = qkv[:, (Hq + Hkv) * D :].view(L, Hkv, D)
+    cu = torch.tensor([0, L], dtype=torch.int32, device=DEV)
+    got = ops.attention_prefill(q, k, v, cu, L)
+    ref = reference.attention_prefill(q.cpu(), k.cpu(), v.cpu(), cu.cpu())
+    assert (to_f32(got) - to_f32(ref)).abs().max().item() < 0.05
+
+
+# ---------- decode attention ----------
+
+@pytest.mark.parametrize(
+    "B,Hq,Hkv,BS,lens",
+    [
+        (1, 4, 4, 16, [1]),
+        (4, 8, 2, 16, [5, 16, 33, 200]),
+        (8, 32, 8, 64, [7, 64, 65, 100, 128, 250, 300, 512]),
+    ],
+)
+def test_attention_decode(B, Hq, Hkv, BS, lens):
+    D = 128
+    max_blocks = (max(lens) + BS - 1) // BS
+    NB = B * max_blocks + 4
+    kc = torch.randn(NB, Hkv, BS, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn_like(kc)
+    # shuffled block tables
+    perm = torch.randperm(NB)[: B * max_blocks].view(B, max_blocks).int().to(DEV)
+    q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=DEV)
+    ctx = torch.tensor(lens, dtype=torch.int32, device=DEV)
+    got = ops.attention_decode(q, kc, vc, perm, ctx)
+    ref = reference.attention_decode(q.cpu(), kc.cpu(), vc.cpu(), perm.cpu(), ctx.cpu())
+    err = (to_f32(got) - to_f32(ref)).abs().max().item()
+    assert err < 0.05, f"max abs err {err}"
+
+
+def test_decode_matches_prefill_last_row():
+    """Decode of the last position == prefill's last row (same K/V)."""
+    D, Hq, Hkv, L, BS = 128, 8, 2, 90, 16
+    q = torch.randn(L, Hq, D, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(L, Hkv, D, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(L, Hkv, D, dtype=torch.bfloat16, device=DEV)
+    cu = torch.tensor([0, L], dtype=torch.int32, device=DEV)
+    pre = ops.attention_prefill(q, k, v, cu, L)
+
+    nb = (L + BS - 1) // BS
+    kc = torch.zeros(nb + 2, Hkv, BS, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.zeros_like(kc)
+    table = torch.arange(1, nb + 1, dtype=torch.int32, device=DEV)  # skip block 0
+    slots = (table[torch.arange(L, device=DEV) // BS] * BS + torch.arange(L, device=DEV) % BS).long()
+    ops.kv_cache_write(k, v, kc, vc, slots)
+    dec = ops.attention_decode(
+        q[L - 1 : L], kc, vc, table.unsqueeze(0), torch.tensor([L], dtype=torch.int32, device=DEV)
+    )
+    err = (to_f32(dec[0]) - to_f32(pre[L - 1])).abs().max().item()
+    assert err < 0.03, f"decode vs prefill last-row err {err}"
+
+
+# ---------- sampling ----------
+
+def test_sample_greedy():
+    B, V = 16, 128256
+    logits = torch.randn(B, V, device=DEV)
+    temps = torch.zeros(B, device=DEV)
+    got = ops.sample(logits, temps)
+    ref = logits.argmax(dim=-1)
+    assert torch.equal(got, ref)
+
+
+def test_sample_temperature_matches_reference():
+    B, V = 8, 50000
+    logits = torch.randn(B, V, device=DEV)
+    temps = torch.full((B,), 0.8, device=DEV)
+    noise = torch.rand(B, V, device=DEV)
+    got = ops.sample(logits, temps, noise)
+    ref = reference.sample(logits.cpu(), temps.cpu(), noise.cpu())
+    assert torch.equal(got.cpu(), ref)
+
+
+def test_sample_mixed_greedy_and_temp():
+    B, V = 4, 1000
+    logits = torch.randn(B, V, device=DEV)
+    temps = torch.tensor([0.0, 1.0, 0.0, 0.5], device=DEV)
+    noise = torch.rand(B, V, device=DEV)
+    got = ops.sample(logits, temps, noise)
+    ref = reference.sample(logits.cpu(), temps.cpu(), noise.cpu())
+    assert torch.equal(got.cpu(), ref)
