@@ -106,6 +106,7 @@ class LLMEngine:
         max_model_len: Optional[int] = None,
         seed: int = 0,
         tp_group: Optional[object] = None,
+        tp_rank: int = 0,
         tp_size: int = 1,
     ):
         full_config = get_model_config(model) if isinstance(model, str) else model
@@ -123,7 +124,8 @@ class LLMEngine:
             torch.cuda.set_device(self.device)
         self.model = LlamaModel(
             config, device=self.device, dtype=dtype, seed=seed,
-            tp_group=tp_group, full_config=full_config,
+            tp_group=tp_group, tp_rank=tp_rank, tp_size=tp_size,
+            full_config=full_config,
         )
         if num_blocks is None:
             num_blocks = PagedKVCache.fit_num_blocks(

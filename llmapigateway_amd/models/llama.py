@@ -58,6 +58,8 @@ class LlamaModel:
         dtype: torch.dtype = torch.bfloat16,
         seed: int = 0,
         tp_group: Optional[object] = None,
+        tp_rank: int = 0,
+        tp_size: int = 1,
         full_config: Optional[ModelConfig] = None,
     ):
         self.config = config
@@ -65,6 +67,8 @@ class LlamaModel:
         self.device = torch.device(device)
         self.dtype = dtype
         self.tp_group = tp_group
+        self.tp_rank = tp_rank
+        self.tp_size = tp_size
         self.scale = config.head_dim ** -0.5
         self.layers: List[Dict[str, torch.Tensor]] = []
         self._init_weights(seed)
@@ -90,18 +94,34 @@ class LlamaModel:
             w.normal_(0.0, std, generator=gen)
             return w.to(self.dtype).to(self.device)
 
+        # TP: every rank draws the same full-shape weights (same seed) and
+        # keeps its Megatron-style slice (parallel/tp.py), so TP=N is a pure
+        # sharding of the TP=1 model (gloo CPU test asserts logits parity).
+        from ..parallel.tp import shard_column, shard_gate_up, shard_qkv, shard_row
+
+        fc = self.full_config
+        tpr, tps = self.tp_rank, self.tp_size
+
         std = 0.02
         out_std = 0.02 / math.sqrt(2 * c.num_layers)
-        self.embed = mk(self.full_config.vocab_size, hidden, std)
+        self.embed = mk(fc.vocab_size, hidden, std)
         for _ in range(c.num_layers):
+            qkv_full = mk(fc.q_size + 2 * fc.kv_size, hidden, std)
+            o_full = mk(hidden, fc.q_size, out_std)
+            gu_full = mk(2 * fc.intermediate_size, hidden, std)
+            down_full = mk(hidden, fc.intermediate_size, out_std)
             self.layers.append(
                 {
                     "input_norm": torch.ones(hidden, dtype=self.dtype, device=self.device),
-                    "qkv": mk(c.q_size + 2 * c.kv_size, hidden, std),
-                    "o": mk(hidden, c.q_size, out_std),
+                    "qkv": shard_qkv(qkv_full, tpr, tps, fc.q_size, fc.kv_size)
+                    if tps > 1
+                    else qkv_full,
+                    "o": shard_row(o_full, tpr, tps) if tps > 1 else o_full,
                     "post_norm": torch.ones(hidden, dtype=self.dtype, device=self.device),
-                    "gate_up": mk(2 * c.intermediate_size, hidden, std),
-                    "down": mk(hidden, c.intermediate_size, out_std),
+                    "gate_up": shard_gate_up(gu_full, tpr, tps, fc.intermediate_size)
+                    if tps > 1
+                    else gu_full,
+                    "down": shard_row(down_full, tpr, tps) if tps > 1 else down_full,
                 }
             )
         self.final_norm = torch.ones(hidden, dtype=self.dtype, device=self.device)

@@ -30,7 +30,7 @@ __global__ void sample_kernel(
         float val = row[i];
         if (!greedy) {
             const float u = fmaxf(nrow[i], 1e-20f);
-            const float g = -__logf(fmaxf(-__logf(u), 1e-20f));
+            const float g = -logf(fmaxf(-logf(u), 1e-20f));
             val = val * invT + g;
         }
         // strict > keeps the lowest index on ties (argmax parity with torch)

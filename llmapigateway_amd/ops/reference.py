@@ -170,6 +170,8 @@ def sample(
     greedy = temps <= 0
     if noise is None or bool(greedy.all()):
         return lf.argmax(dim=-1)
-    gumbel = -torch.log(-torch.log(noise.float().clamp_min(1e-20)).clamp_min(1e-20))
+    u = noise.float().clamp_min(1e-20)
+    inner = (-torch.log(u)).clamp_min(1e-20)
+    gumbel = -torch.log(inner)
     scored = torch.where(greedy, lf, lf / temps.clamp_min(1e-6) + gumbel)
     return scored.argmax(dim=-1)

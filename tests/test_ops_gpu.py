@@ -197,14 +197,28 @@ def test_sample_greedy():
     assert torch.equal(got, ref)
 
 
+def _assert_sample_equiv(got, logits, temps, noise):
+    """GPU argmax must pick an entry whose gumbel-perturbed score ties the
+    CPU reference winner (logf rounding can flip exact index on near-ties)."""
+    ref = reference.sample(logits.cpu(), temps.cpu(), noise.cpu())
+    lf = logits.float().cpu()
+    u = noise.float().cpu().clamp_min(1e-20)
+    g = -torch.log((-torch.log(u)).clamp_min(1e-20))
+    for b in range(logits.shape[0]):
+        if temps[b].item() <= 0:
+            assert got[b].item() == ref[b].item()
+            continue
+        scored = lf[b] / temps[b].item() + g[b]
+        assert scored[got[b].item()] >= scored[ref[b].item()] - 1e-3
+
+
 def test_sample_temperature_matches_reference():
     B, V = 8, 50000
     logits = torch.randn(B, V, device=DEV)
     temps = torch.full((B,), 0.8, device=DEV)
     noise = torch.rand(B, V, device=DEV)
-    got = ops.sample(logits, temps, noise)
-    ref = reference.sample(logits.cpu(), temps.cpu(), noise.cpu())
-    assert torch.equal(got.cpu(), ref)
+    got = ops.sample(logits, temps, noise).cpu()
+    _assert_sample_equiv(got, logits, temps, noise)
 
 
 def test_sample_mixed_greedy_and_temp():
@@ -212,6 +226,5 @@ def test_sample_mixed_greedy_and_temp():
     logits = torch.randn(B, V, device=DEV)
     temps = torch.tensor([0.0, 1.0, 0.0, 0.5], device=DEV)
     noise = torch.rand(B, V, device=DEV)
-    got = ops.sample(logits, temps, noise)
-    ref = reference.sample(logits.cpu(), temps.cpu(), noise.cpu())
-    assert torch.equal(got.cpu(), ref)
+    got = ops.sample(logits, temps, noise).cpu()
+    _assert_sample_equiv(got, logits, temps, noise)
