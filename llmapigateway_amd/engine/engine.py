@@ -105,6 +105,7 @@ class LLMEngine:
         hbm_fraction: float = 0.90,
         max_model_len: Optional[int] = None,
         seed: int = 0,
+        use_hipgraph: Optional[bool] = None,
         tp_group: Optional[object] = None,
         tp_rank: int = 0,
         tp_size: int = 1,
@@ -132,6 +133,22 @@ class LLMEngine:
                 config, block_size, self.device, dtype, hbm_fraction
             )
         self.kv = PagedKVCache(config, num_blocks, block_size, self.device, dtype)
+
+        # hipGraph-captured decode (GPU only; TP group ops are capturable
+        # with RCCL but kept off by default under TP until validated)
+        if use_hipgraph is None:
+            use_hipgraph = self.device.type == "cuda" and tp_size == 1
+        self.graph_runner = None
+        if use_hipgraph and self.device.type == "cuda":
+            from .graph_runner import DecodeGraphRunner
+
+            self.graph_runner = DecodeGraphRunner(
+                self.model,
+                self.kv.k_caches,
+                self.kv.v_caches,
+                max_batch=max_batch_size,
+                max_blocks=(self.max_model_len + block_size - 1) // block_size,
+            )
 
         self._gen = torch.Generator(device=self.device).manual_seed(seed ^ 0x5EED)
         self._lock = threading.Lock()
@@ -293,25 +310,34 @@ class LLMEngine:
         ]
         pos = [req.num_tokens - 1 for req in reqs]
         slots = [req.block_table[p // bs] * bs + p % bs for req, p in zip(reqs, pos)]
-        max_blocks = max(len(req.block_table) for req in reqs)
-        tables = torch.zeros(len(reqs), max_blocks, dtype=torch.int32, device=device)
-        for i, req in enumerate(reqs):
-            tables[i, : len(req.block_table)] = torch.tensor(
-                req.block_table, dtype=torch.int32
-            )
+        ctx = [p + 1 for p in pos]
 
-        batch = ForwardBatch(
-            kind="decode",
-            token_ids=torch.tensor(last_tokens, dtype=torch.long, device=device),
-            positions=torch.tensor(pos, dtype=torch.long, device=device),
-            slot_mapping=torch.tensor(slots, dtype=torch.long, device=device),
-            block_tables=tables,
-            context_lens=torch.tensor(
-                [p + 1 for p in pos], dtype=torch.int32, device=device
-            ),
-            logits_indices=None,
-        )
-        logits = self.model.forward(batch, self.kv.k_caches, self.kv.v_caches)
+        logits = None
+        if self.graph_runner is not None:
+            try:
+                logits = self.graph_runner.run(
+                    last_tokens, pos, slots, [r.block_table for r in reqs], ctx
+                )
+            except Exception:
+                logger.exception("hipGraph decode failed; falling back to eager")
+                self.graph_runner = None
+        if logits is None:
+            max_blocks = max(len(req.block_table) for req in reqs)
+            tables = torch.zeros(len(reqs), max_blocks, dtype=torch.int32, device=device)
+            for i, req in enumerate(reqs):
+                tables[i, : len(req.block_table)] = torch.tensor(
+                    req.block_table, dtype=torch.int32
+                )
+            batch = ForwardBatch(
+                kind="decode",
+                token_ids=torch.tensor(last_tokens, dtype=torch.long, device=device),
+                positions=torch.tensor(pos, dtype=torch.long, device=device),
+                slot_mapping=torch.tensor(slots, dtype=torch.long, device=device),
+                block_tables=tables,
+                context_lens=torch.tensor(ctx, dtype=torch.int32, device=device),
+                logits_indices=None,
+            )
+            logits = self.model.forward(batch, self.kv.k_caches, self.kv.v_caches)
         tokens = self._sample(logits, reqs)
         self.stats["decode_tokens"] += len(reqs)
         with self._lock:

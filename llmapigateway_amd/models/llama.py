@@ -151,7 +151,7 @@ class LlamaModel:
         """Returns logits [B, vocab] at batch.logits_indices."""
         c = self.config
         h = F.embedding(batch.token_ids, self.embed)
-        residual = torch.zeros_like(h)
+        residual = h  # placeholder; layer 0 sets the real residual stream
         T = h.shape[0]
 
         for i, layer in enumerate(self.layers):

@@ -7,23 +7,25 @@
 //
 // Geometry: one workgroup per (sequence, kv-head); 4 waves; the G = Hq/Hkv
 // query heads of the group share the streamed K/V (GQA bandwidth saving).
+// G is a TEMPLATE parameter: with runtime G the per-lane s/m/l/acc arrays
+// are runtime-indexed and spill to scratch (cdna_hip_programming.md §5.4
+// rule 20) — the first profiled build ran 12x off roofline exactly there.
+//
 // Each wave owns key chunks of 64 positions (chunk c -> wave c%4):
 //   phase A: lane = one key position; the lane streams that key's 128-dim
-//            row (16 B vector loads, L1-resident across the 16 loads) and
-//            dots it against q (LDS-broadcast reads);
+//            row (16 B vector loads) and dots it against q staged in LDS
+//            (vectorized float4 broadcast reads);
 //   phase B: lanes switch to dim ownership (lane d owns dims 2d, 2d+1) and
 //            stream V rows coalesced (64 lanes x 4 B = a 256 B row).
 // Online softmax per wave, flash-style cross-wave combine in LDS at the end.
-//
-// Cache layout: [num_blocks, Hkv, block_size, D]; block_size any power of
-// two <= 64 handled via the per-lane block-table lookup.
 
 #include "common.h"
 
-#define MAX_G 8          // max query heads per kv head (llama-3: 4, 70B TP1: 8)
+#define MAX_G 8
 #define NWAVES 4
 #define DECODE_D 128
 
+template <int G>
 __launch_bounds__(NWAVES* WAVE_SIZE)
 __global__ void attention_decode_kernel(
     bf16* __restrict__ out,                 // [B, Hq, D]
@@ -40,7 +42,6 @@ __global__ void attention_decode_kernel(
     int64_t q_stride) {
     const int seq = blockIdx.x;
     const int kvh = blockIdx.y;
-    const int G = Hq / Hkv;
     const int D = DECODE_D;
     const int L = context_lens[seq];
     if (L <= 0) return;
@@ -49,15 +50,13 @@ __global__ void attention_decode_kernel(
     const int lane = tid & (WAVE_SIZE - 1);
     const int wave = tid >> 6;
 
-    // ---- LDS ----
-    __shared__ float q_lds[MAX_G][DECODE_D];
-    __shared__ float p_lds[NWAVES][MAX_G][WAVE_SIZE];
-    __shared__ float m_lds[NWAVES][MAX_G];
-    __shared__ float l_lds[NWAVES][MAX_G];
-    __shared__ float acc_lds[NWAVES][MAX_G][DECODE_D];
-    __shared__ int bt_lds[1024];  // cached block table (<= 64K tokens at BS=64)
+    __shared__ float q_lds[G][DECODE_D];
+    __shared__ float p_lds[NWAVES][G][WAVE_SIZE];
+    __shared__ float m_lds[NWAVES][G];
+    __shared__ float l_lds[NWAVES][G];
+    __shared__ float acc_lds[NWAVES][G][DECODE_D];
+    __shared__ int bt_lds[1024];
 
-    // stage q group into LDS as fp32 (scaled once here)
     for (int i = tid; i < G * D; i += blockDim.x) {
         const int g = i / D, d = i % D;
         q_lds[g][d] =
@@ -67,16 +66,15 @@ __global__ void attention_decode_kernel(
     for (int i = tid; i < nblocks && i < 1024; i += blockDim.x)
         bt_lds[i] = block_tables[(size_t)seq * max_blocks + i];
     __syncthreads();
-    // very long contexts (>1024 blocks) read the tail of the table from L2
     const int* __restrict__ bt_global = block_tables + (size_t)seq * max_blocks;
 #define BT(idx) ((idx) < 1024 ? bt_lds[(idx)] : bt_global[(idx)])
 
-    float m[MAX_G], l[MAX_G], acc[MAX_G][2];
+    float m[G], l[G], acc0[G], acc1[G];
 #pragma unroll
-    for (int g = 0; g < MAX_G; ++g) {
+    for (int g = 0; g < G; ++g) {
         m[g] = -INFINITY;
         l[g] = 0.f;
-        acc[g][0] = acc[g][1] = 0.f;
+        acc0[g] = acc1[g] = 0.f;
     }
 
     const int nchunks = (L + WAVE_SIZE - 1) / WAVE_SIZE;
@@ -84,51 +82,57 @@ __global__ void attention_decode_kernel(
         const int pos = c * WAVE_SIZE + lane;
         const bool valid = pos < L;
 
-        // --- phase A: lane = key; dot q . k ---
-        float s[MAX_G];
+        // --- phase A: lane = key ---
+        float s[G];
 #pragma unroll
-        for (int g = 0; g < MAX_G; ++g) s[g] = -INFINITY;
-        int kblock = 0, koff = 0;
+        for (int g = 0; g < G; ++g) s[g] = valid ? 0.f : -INFINITY;
         if (valid) {
-            kblock = BT(pos / block_size);
-            koff = pos % block_size;
+            const int kblock = BT(pos / block_size);
+            const int koff = pos % block_size;
             const uint4* krow = reinterpret_cast<const uint4*>(
                 k_cache + (((size_t)kblock * Hkv + kvh) * block_size + koff) * D);
-#pragma unroll
-            for (int g = 0; g < MAX_G; ++g) s[g] = 0.f;
+#pragma unroll 4
             for (int i = 0; i < D / 8; ++i) {
-                uint4 kv8 = krow[i];
+                const uint4 kv8 = krow[i];
                 float kf[8];
                 unpack2(kv8.x, kf[0], kf[1]);
                 unpack2(kv8.y, kf[2], kf[3]);
                 unpack2(kv8.z, kf[4], kf[5]);
                 unpack2(kv8.w, kf[6], kf[7]);
-                for (int g = 0; g < G; ++g) {
 #pragma unroll
-                    for (int j = 0; j < 8; ++j)
-                        s[g] += q_lds[g][i * 8 + j] * kf[j];
+                for (int g = 0; g < G; ++g) {
+                    const float4 qa = *reinterpret_cast<const float4*>(&q_lds[g][i * 8]);
+                    const float4 qb = *reinterpret_cast<const float4*>(&q_lds[g][i * 8 + 4]);
+                    s[g] = fmaf(qa.x, kf[0], s[g]);
+                    s[g] = fmaf(qa.y, kf[1], s[g]);
+                    s[g] = fmaf(qa.z, kf[2], s[g]);
+                    s[g] = fmaf(qa.w, kf[3], s[g]);
+                    s[g] = fmaf(qb.x, kf[4], s[g]);
+                    s[g] = fmaf(qb.y, kf[5], s[g]);
+                    s[g] = fmaf(qb.z, kf[6], s[g]);
+                    s[g] = fmaf(qb.w, kf[7], s[g]);
                 }
             }
         }
 
         // --- online softmax update (per wave) ---
-        float chunk_sum[MAX_G];
+#pragma unroll
         for (int g = 0; g < G; ++g) {
             const float cmax = wave_reduce_max(s[g]);
             const float m_new = fmaxf(m[g], cmax);
             float p = 0.f;
             if (valid && m_new != -INFINITY) p = __expf(s[g] - m_new);
             const float factor = (m[g] == -INFINITY) ? 0.f : __expf(m[g] - m_new);
-            chunk_sum[g] = wave_reduce_sum(p);
-            l[g] = l[g] * factor + chunk_sum[g];
-            acc[g][0] *= factor;
-            acc[g][1] *= factor;
+            const float csum = wave_reduce_sum(p);
+            l[g] = l[g] * factor + csum;
+            acc0[g] *= factor;
+            acc1[g] *= factor;
             m[g] = m_new;
             p_lds[wave][g][lane] = p;
         }
-        // wave-synchronous LDS use: no barrier needed (private per wave)
+        // wave-synchronous LDS use: private per wave, no barrier needed
 
-        // --- phase B: lane = dim pair; stream V rows ---
+        // --- phase B: lane = dim pair; stream V rows coalesced ---
         const int chunk_keys = min(WAVE_SIZE, L - c * WAVE_SIZE);
         for (int j = 0; j < chunk_keys; ++j) {
             const int p2 = c * WAVE_SIZE + j;
@@ -138,26 +142,29 @@ __global__ void attention_decode_kernel(
                 v_cache + (((size_t)vb * Hkv + kvh) * block_size + vo) * D)[lane];
             float v0, v1;
             unpack2(vpair, v0, v1);
+#pragma unroll
             for (int g = 0; g < G; ++g) {
                 const float pj = p_lds[wave][g][j];
-                acc[g][0] = fmaf(pj, v0, acc[g][0]);
-                acc[g][1] = fmaf(pj, v1, acc[g][1]);
+                acc0[g] = fmaf(pj, v0, acc0[g]);
+                acc1[g] = fmaf(pj, v1, acc1[g]);
             }
         }
     }
 
     // ---- cross-wave flash combine ----
+#pragma unroll
     for (int g = 0; g < G; ++g) {
         if (lane == 0) {
             m_lds[wave][g] = m[g];
             l_lds[wave][g] = l[g];
         }
-        acc_lds[wave][g][2 * lane] = acc[g][0];
-        acc_lds[wave][g][2 * lane + 1] = acc[g][1];
+        acc_lds[wave][g][2 * lane] = acc0[g];
+        acc_lds[wave][g][2 * lane + 1] = acc1[g];
     }
     __syncthreads();
 
     if (wave == 0) {
+#pragma unroll
         for (int g = 0; g < G; ++g) {
             float gm = -INFINITY;
 #pragma unroll
@@ -177,7 +184,7 @@ __global__ void attention_decode_kernel(
                 o1 += f[w] * acc_lds[w][g][2 * lane + 1];
             }
             reinterpret_cast<uint32_t*>(
-                out + (size_t)seq * Hq * D + (size_t)(kvh * G + g) * D)[lane] =
+                out + (size_t)seq * Hq * DECODE_D + (size_t)(kvh * G + g) * DECODE_D)[lane] =
                 pack2(o0 * inv_l, o1 * inv_l);
         }
     }
@@ -188,18 +195,24 @@ extern "C" hipError_t launch_attention_decode(
     const int* block_tables, const int* context_lens, float scale, int B,
     int Hq, int Hkv, int block_size, int max_blocks, int D, int64_t q_stride,
     hipStream_t stream) {
-    if (D != DECODE_D && D != 64) return hipErrorInvalidValue;
-    if (Hq / Hkv > MAX_G || Hq % Hkv != 0) return hipErrorInvalidValue;
+    if (D != DECODE_D) return hipErrorNotSupported;
+    if (Hq % Hkv != 0) return hipErrorInvalidValue;
+    const int G = Hq / Hkv;
     dim3 grid(B, Hkv);
     dim3 block(NWAVES * WAVE_SIZE);
-    if (D == DECODE_D) {
-       hipLaunchKernelGGL(( attention_decode_kernel), dim3(grid), dim3(block), 0, stream, 
-            (bf16*)out, (const bf16*)q, (const bf16*)k_cache,
-            (const bf16*)v_cache, block_tables, context_lens, scale, Hq, Hkv,
-            block_size, max_blocks, q_stride);
-    } else {
-        return hipErrorNotSupported;  // head_dim 64 engine presets run D=128 models on GPU
+#define LAUNCH_G(GV)                                                           \
+   hipLaunchKernelGGL(( attention_decode_kernel<GV>), dim3(grid), dim3(block), 0, stream,                    \
+        (bf16*)out, (const bf16*)q, (const bf16*)k_cache,                      \
+        (const bf16*)v_cache, block_tables, context_lens, scale, Hq, Hkv,      \
+        block_size, max_blocks, q_stride)
+    switch (G) {
+        case 1: LAUNCH_G(1); break;
+        case 2: LAUNCH_G(2); break;
+        case 4: LAUNCH_G(4); break;
+        case 8: LAUNCH_G(8); break;
+        default: return hipErrorInvalidValue;
     }
+#undef LAUNCH_G
     HIP_CHECK_LAST();
     return hipSuccess;
 }

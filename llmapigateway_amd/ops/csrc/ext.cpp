@@ -19,7 +19,7 @@ hipError_t launch_swiglu(void*, const void*, int, int, hipStream_t);
 hipError_t launch_kv_cache_write(const void*, const void*, void*, void*, const int64_t*, int, int64_t, int64_t, int, int, int, hipStream_t);
 hipError_t launch_attention_decode(void*, const void*, const void*, const void*, const int*, const int*, float, int, int, int, int, int, int, int64_t, hipStream_t);
 hipError_t launch_attention_prefill(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, float, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
-hipError_t launch_sample(int64_t*, const float*, const float*, const float*, int, int, hipStream_t);
+hipError_t launch_sample(int64_t*, const float*, const float*, const float*, float*, int*, int, int, hipStream_t);
 }
 
 namespace {
@@ -159,9 +159,15 @@ void sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temperature,
         TORCH_CHECK(noise->is_contiguous() && noise->scalar_type() == torch::kFloat32);
         noise_ptr = noise->data_ptr<float>();
     }
+    const int B = logits.size(0);
+    auto part_val = torch::empty(
+        {B, 16}, torch::TensorOptions().dtype(torch::kFloat32).device(logits.device()));
+    auto part_idx = torch::empty(
+        {B, 16}, torch::TensorOptions().dtype(torch::kInt32).device(logits.device()));
     CHECK_HIP(launch_sample(out.data_ptr<int64_t>(), logits.data_ptr<float>(),
                             temperature.data_ptr<float>(), noise_ptr,
-                            logits.size(0), logits.size(1), current_stream()));
+                            part_val.data_ptr<float>(), part_idx.data_ptr<int>(),
+                            B, logits.size(1), current_stream()));
 }
 
 }  // namespace
