@@ -76,6 +76,7 @@ class EngineRequest:
         self.prompt_ids = list(prompt_ids)
         self.params = params
         self.out_ids: List[int] = []
+        self.text = ""  # incrementally decoded output (stop-string detection)
         self.block_table: List[int] = []
         self.state = "waiting"  # waiting | running | finished | failed
         self.finish_reason: Optional[str] = None
@@ -388,10 +389,14 @@ class LLMEngine:
     # ---- delivery / lifecycle (call with lock held) ----
     def _deliver(self, reqs: List[EngineRequest], tokens: List[int]) -> None:
         now = time.monotonic()
+        from .tokenizer import ByteTokenizer
+
         for req, tok in zip(reqs, tokens):
             if req.state != "running":
                 continue
             req.out_ids.append(int(tok))
+            if req.params.stop:
+                req.text += ByteTokenizer().decode([int(tok)])
             if req.first_token_time is None:
                 req.first_token_time = now
             if req.on_token is not None:
@@ -408,6 +413,8 @@ class LLMEngine:
         from .tokenizer import ByteTokenizer
 
         if not req.params.ignore_eos and tok == ByteTokenizer.EOS:
+            return "stop"
+        if req.params.stop and any(s in req.text for s in req.params.stop):
             return "stop"
         if len(req.out_ids) >= req.params.max_tokens:
             return "length"

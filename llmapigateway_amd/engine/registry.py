@@ -233,6 +233,10 @@ class EngineRegistry:
         if finished is not None and finished.state == "failed":
             return None, finished.error or "engine failure"
         text = self.tokenizer.decode(tokens)
+        for s in req.params.stop:  # OpenAI semantics: stop string excluded
+            idx = text.find(s)
+            if idx >= 0:
+                text = text[:idx]
         usage = {
             "prompt_tokens": len(req.prompt_ids),
             "completion_tokens": len(req.out_ids),
@@ -273,15 +277,42 @@ class EngineRegistry:
                 obj["usage"] = usage
             return b"data: " + json.dumps(obj, separators=(",", ":")).encode() + b"\n\n"
 
+        stops = req.params.stop
+        holdback = max((len(s) for s in stops), default=1) - 1 if stops else 0
+
         async def gen():
+            pending = ""
+            stopped = False
             try:
                 yield chunk({"role": "assistant", "content": ""})
                 kind, value = first_kind, first_value
                 while True:
                     if kind == "token":
-                        yield chunk({"content": tokenizer.decode([value])})
+                        if stops and not stopped:
+                            # hold back enough text to cleanly cut a stop
+                            # string before it reaches the client
+                            pending += tokenizer.decode([value])
+                            cut = -1
+                            for s in stops:
+                                i = pending.find(s)
+                                if i >= 0 and (cut < 0 or i < cut):
+                                    cut = i
+                            if cut >= 0:
+                                if pending[:cut]:
+                                    yield chunk({"content": pending[:cut]})
+                                pending = ""
+                                stopped = True
+                            elif len(pending) > holdback:
+                                emit = pending[: len(pending) - holdback]
+                                pending = pending[len(pending) - holdback :]
+                                yield chunk({"content": emit})
+                        elif not stopped:
+                            yield chunk({"content": tokenizer.decode([value])})
                     else:
                         fin: EngineRequest = value
+                        if pending and not stopped:
+                            yield chunk({"content": pending})
+                            pending = ""
                         usage = {
                             "prompt_tokens": len(fin.prompt_ids),
                             "completion_tokens": len(fin.out_ids),
