@@ -9,7 +9,18 @@ PyTorch. CPU tensors use ops.reference (fp32 semantics contract).
 
 from __future__ import annotations
 
+import os
 from typing import Optional, Tuple
+
+# Enable the pre-tuned hipBLASLt/rocBLAS GEMM selections for gfx950 (3.2x on
+# the decode lm_head GEMM). torch appends the device ordinal before ".csv",
+# so the shipped table is tunableop_gfx950<N>.csv and FILENAME names the
+# base. Must be set before the first GEMM executes; honor user overrides.
+_TUNE_BASE = os.path.join(os.path.dirname(os.path.abspath(__file__)), "tunableop_gfx950.csv")
+if os.path.exists(_TUNE_BASE.replace(".csv", "0.csv")):
+    os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME", _TUNE_BASE)
 
 import torch
 

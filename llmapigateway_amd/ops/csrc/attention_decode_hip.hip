@@ -132,9 +132,34 @@ __global__ void attention_decode_kernel(
         }
         // wave-synchronous LDS use: private per wave, no barrier needed
 
-        // --- phase B: lane = dim pair; stream V rows coalesced ---
+        // --- phase B: lane = dim pair; stream V rows coalesced. 4 rows per
+        // iteration: 4 independent loads in flight (the serial 1-row loop
+        // was latency-bound on the per-row L2 round trip) ---
         const int chunk_keys = min(WAVE_SIZE, L - c * WAVE_SIZE);
-        for (int j = 0; j < chunk_keys; ++j) {
+        int j = 0;
+        for (; j + 4 <= chunk_keys; j += 4) {
+            uint32_t vp[4];
+#pragma unroll
+            for (int u = 0; u < 4; ++u) {
+                const int p2 = c * WAVE_SIZE + j + u;
+                const int vb = BT(p2 / block_size);
+                const int vo = p2 % block_size;
+                vp[u] = reinterpret_cast<const uint32_t*>(
+                    v_cache + (((size_t)vb * Hkv + kvh) * block_size + vo) * D)[lane];
+            }
+#pragma unroll
+            for (int u = 0; u < 4; ++u) {
+                float v0, v1;
+                unpack2(vp[u], v0, v1);
+#pragma unroll
+                for (int g = 0; g < G; ++g) {
+                    const float pj = p_lds[wave][g][j + u];
+                    acc0[g] = fmaf(pj, v0, acc0[g]);
+                    acc1[g] = fmaf(pj, v1, acc1[g]);
+                }
+            }
+        }
+        for (; j < chunk_keys; ++j) {
             const int p2 = c * WAVE_SIZE + j;
             const int vb = BT(p2 / block_size);
             const int vo = p2 % block_size;
