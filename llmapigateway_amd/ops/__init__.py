@@ -24,6 +24,19 @@ if os.path.exists(_TUNE_BASE.replace(".csv", "0.csv")):
 
 import torch
 
+# Belt-and-braces: activate TunableOp through the python API too (the env
+# vars alone were not picked up when torch was imported first elsewhere).
+if torch.version.hip and os.path.exists(_TUNE_BASE.replace(".csv", "0.csv")):
+    try:
+        import torch.cuda.tunable as _tunable
+
+        _tunable.enable(True)
+        _tunable.tuning_enable(False)
+        _tunable.set_filename(_TUNE_BASE, insert_device_ordinal=True)
+        _tunable.read_file(_TUNE_BASE.replace(".csv", "0.csv"))
+    except Exception:  # pragma: no cover - best effort
+        pass
+
 from . import reference
 from .reference import build_rope_cache  # re-export (host-side table builder)
 
