@@ -61,6 +61,18 @@ class _EngineHandle:
         self.thread.join(timeout=5.0)
 
 
+class _TPGroupHandle:
+    """Handle for a multi-process TP engine group (no local step loop —
+    the lockstep loop runs inside the worker processes)."""
+
+    def __init__(self, client, name: str):
+        self.engine = client
+        self.name = name
+
+    def stop(self) -> None:
+        self.engine.stop()
+
+
 class _FailureState:
     def __init__(self, fail_rate: float = 0.0, fail_requests: Optional[int] = None):
         self.fail_rate = fail_rate
@@ -108,6 +120,25 @@ class EngineRegistry:
         with self._create_lock:
             handle = self._engines.get(key)
             if handle is not None:
+                return handle
+            if spec.tp > 1:
+                from .tp_group import TPEngineClient
+
+                if torch.cuda.is_available() and torch.cuda.device_count() < spec.tp:
+                    raise RuntimeError(
+                        f"Engine spec asks for TP={spec.tp} but only "
+                        f"{torch.cuda.device_count()} GPUs are visible"
+                    )
+                client = TPEngineClient(
+                    model=spec.model,
+                    tp=spec.tp,
+                    dtype="float16" if spec.dtype in ("float16", "fp16") else "bfloat16",
+                    max_batch_size=spec.max_batch_size or self.settings.engine_max_batch_size,
+                    kv_block_size=spec.kv_block_size or self.settings.engine_kv_block_size,
+                )
+                handle = _TPGroupHandle(client, key)
+                self._engines[key] = handle
+                logger.info("created TP=%d engine group %s", spec.tp, key)
                 return handle
             device = self._resolve_device(spec)
             dtype = torch.bfloat16 if device.startswith("cuda") else torch.float32
@@ -350,18 +381,22 @@ class EngineRegistry:
         out = []
         for key, handle in self._engines.items():
             eng = handle.engine
-            out.append(
-                {
-                    "engine": key,
-                    "device": str(eng.device),
-                    "model": eng.full_config.name,
-                    "waiting": len(eng.waiting),
-                    "running": len(eng.running),
-                    "kv_blocks_free": eng.kv.manager.num_free_blocks,
-                    "kv_blocks_total": eng.kv.num_blocks,
-                    **{k: int(v) for k, v in eng.stats.items()},
-                }
-            )
+            row = {
+                "engine": key,
+                "model": eng.full_config.name,
+                **{k: int(v) for k, v in eng.stats.items()},
+            }
+            if hasattr(eng, "kv"):  # single-process engine
+                row.update(
+                    device=str(eng.device),
+                    waiting=len(eng.waiting),
+                    running=len(eng.running),
+                    kv_blocks_free=eng.kv.manager.num_free_blocks,
+                    kv_blocks_total=eng.kv.num_blocks,
+                )
+            else:  # TP worker group (state lives in the worker processes)
+                row.update(device=f"tp{eng.tp}", in_flight=len(eng._reqs))
+            out.append(row)
         return out
 
     async def aclose(self) -> None:

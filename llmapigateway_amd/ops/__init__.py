@@ -31,9 +31,12 @@ if torch.version.hip and os.path.exists(_TUNE_BASE.replace(".csv", "0.csv")):
         import torch.cuda.tunable as _tunable
 
         _tunable.enable(True)
-        _tunable.tuning_enable(False)
-        _tunable.set_filename(_TUNE_BASE, insert_device_ordinal=True)
-        _tunable.read_file(_TUNE_BASE.replace(".csv", "0.csv"))
+        # honor an explicit tuning request (tools/tune_gemms.py sets "1");
+        # otherwise lock to lookup-only so serving never pays tuning cost
+        if os.environ.get("PYTORCH_TUNABLEOP_TUNING") != "1":
+            _tunable.tuning_enable(False)
+            _tunable.set_filename(_TUNE_BASE, insert_device_ordinal=True)
+            _tunable.read_file(_TUNE_BASE.replace(".csv", "0.csv"))
     except Exception:  # pragma: no cover - best effort
         pass
 

@@ -1,0 +1,102 @@
+"""TP worker-group serving path (engine/tp_group.py) on CPU over gloo.
+
+Covers BASELINE configs[4] plumbing without GPUs: a TPEngineClient with
+tp=2 must serve requests through the lockstep multi-process loop and emit
+exactly the tokens a single-process TP=1 engine produces (greedy), because
+TP=N is a pure sharding of the TP=1 model (parallel/tp.py docstring).
+"""
+
+import sys
+
+import pytest
+import torch
+
+from llmapigateway_amd.engine import EngineRequest, LLMEngine, SamplingParams
+from llmapigateway_amd.engine.tp_group import TPEngineClient
+
+pytestmark = pytest.mark.skipif(
+    sys.platform != "linux", reason="multiprocessing spawn test is linux-only"
+)
+
+PROMPT = list(range(5, 37))
+
+
+def _reference_tokens(n=8):
+    engine = LLMEngine(
+        model="tiny-llama", device="cpu", dtype=torch.float32, num_blocks=64, seed=0
+    )
+    req = engine.generate(PROMPT, SamplingParams(max_tokens=n, ignore_eos=True))
+    assert req.state == "finished"
+    return req.out_ids
+
+
+@pytest.fixture(scope="module")
+def tp_client():
+    client = TPEngineClient(
+        model="tiny-llama", tp=2, max_batch_size=8, kv_block_size=16, num_blocks=64,
+        start_timeout=180.0,
+    )
+    yield client
+    client.stop()
+
+
+def test_tp2_greedy_matches_tp1(tp_client):
+    import threading
+
+    done = threading.Event()
+    req = EngineRequest(
+        PROMPT,
+        SamplingParams(max_tokens=8, ignore_eos=True),
+        on_finish=lambda r: done.set(),
+    )
+    tp_client.add_request(req)
+    assert done.wait(timeout=120.0), "TP group did not finish the request"
+    assert req.state == "finished"
+    assert req.out_ids == _reference_tokens(8)
+
+
+def test_tp2_concurrent_requests(tp_client):
+    import threading
+
+    n = 4
+    events = [threading.Event() for _ in range(n)]
+    reqs = []
+    for i in range(n):
+        req = EngineRequest(
+            PROMPT[: 8 + i],
+            SamplingParams(max_tokens=5, ignore_eos=True),
+            on_finish=lambda r, e=events[i]: e.set(),
+        )
+        tp_client.add_request(req)
+        reqs.append(req)
+    for e in events:
+        assert e.wait(timeout=120.0)
+    for req in reqs:
+        assert req.state == "finished"
+        assert len(req.out_ids) == 5
+
+
+def test_tp2_abort(tp_client):
+    req = EngineRequest(PROMPT, SamplingParams(max_tokens=10_000, ignore_eos=True))
+    tp_client.add_request(req)
+    tp_client.abort_request(req)
+    assert req.state in ("finished", "failed")
+    assert req.finish_reason == "aborted"
+
+
+def test_registry_tp_spec_creates_group():
+    from llmapigateway_amd.config.loader import EngineSpec
+    from llmapigateway_amd.engine.registry import EngineRegistry, _TPGroupHandle
+
+    reg = EngineRegistry()
+    spec = EngineSpec(model="tiny-llama", tp=2, max_batch_size=4, kv_block_size=16)
+    # CPU path: TPEngineClient defaults num_blocks=256 per rank
+    handle = reg.get_engine(spec)
+    try:
+        assert isinstance(handle, _TPGroupHandle)
+        assert handle.engine.tp == 2
+        # second lookup returns the cached group
+        assert reg.get_engine(spec) is handle
+    finally:
+        handle.stop()
+        reg._engines.clear()
