@@ -34,6 +34,10 @@ os.environ["PYTORCH_TUNABLEOP_TUNING"] = "1"
 os.environ["PYTORCH_TUNABLEOP_FILENAME"] = OUT_TMP
 os.environ.setdefault("PYTORCH_TUNABLEOP_MAX_TUNING_DURATION_MS", "100")
 os.environ.setdefault("PYTORCH_TUNABLEOP_MAX_TUNING_ITERATIONS", "30")
+# Rotate through a >L2-sized pool of A/B/C buffers while benchmarking each
+# solution: the engine streams 0.5 GB of cold weights per layer, so a
+# hot-cache micro-bench picks cache-exploiting solutions that lose in situ.
+os.environ.setdefault("PYTORCH_TUNABLEOP_ROTATING_BUFFER_SIZE", "1024")
 # do not read any pre-existing table: tune from scratch
 os.environ["PYTORCH_TUNABLEOP_TUNING_AFTER_READ"] = "1"
 
