@@ -9,7 +9,9 @@ flat varlen batch (ForwardBatch) in one of two modes:
 - decode: one token per running sequence; ops.attention_decode over the
   paged cache via block tables.
 
-Projections run on hipBLASLt/rocBLAS through F.linear (library GEMMs); the
+Projections run through ops.linear: decode-shaped batches (M<=256) hit the
+hand-written weight-streaming gfx950 GEMM (ops/csrc/gemm_skinny.hip),
+prefill batches the TunableOp-tuned hipBLASLt/rocBLAS library GEMMs; the
 fused hot ops (rmsnorm+residual, rope, swiglu, attention, sampling) are the
 hand-written CDNA4 kernels behind llmapigateway_amd.ops.
 
@@ -160,7 +162,7 @@ class LlamaModel:
             else:
                 x, residual = ops.rmsnorm_residual(h, residual, layer["input_norm"], c.rms_eps)
 
-            qkv = F.linear(x, layer["qkv"])
+            qkv = ops.linear(x, layer["qkv"])
             q, k, v = qkv.split([c.q_size, c.kv_size, c.kv_size], dim=-1)
             q = q.view(T, c.num_heads, c.head_dim)
             k = k.view(T, c.num_kv_heads, c.head_dim)
@@ -178,14 +180,14 @@ class LlamaModel:
                     q, k_caches[i], v_caches[i], batch.block_tables, batch.context_lens, self.scale
                 )
 
-            h = self._maybe_all_reduce(F.linear(attn.reshape(T, c.q_size), layer["o"]))
+            h = self._maybe_all_reduce(ops.linear(attn.reshape(T, c.q_size), layer["o"]))
 
             x, residual = ops.rmsnorm_residual(h, residual, layer["post_norm"], c.rms_eps)
             h = self._maybe_all_reduce(
-                F.linear(ops.swiglu(F.linear(x, layer["gate_up"])), layer["down"])
+                ops.linear(ops.swiglu(ops.linear(x, layer["gate_up"])), layer["down"])
             )
 
         x, _ = ops.rmsnorm_residual(h, residual, self.final_norm, c.rms_eps)
         if batch.logits_indices is not None:
             x = x[batch.logits_indices]
-        return F.linear(x, self.lm_head).float()
+        return ops.linear(x, self.lm_head).float()

@@ -182,6 +182,65 @@ def attention_decode(
     return reference.attention_decode(q, k_cache, v_cache, block_tables, context_lens, scale)
 
 
+# The streaming kernel wins in the latency regime (tiny M: single-request
+# decode, where the library's tile/launch overhead dominates) and ties on
+# lm_head; at M>=64 hipBLASLt/rocBLAS reach ~60% of the per-CU fetch
+# ceiling and win — measured in profiles/gemm_skinny_probe (tools/gemm_probe.py).
+_SKINNY_MAX_M = 16
+_SKINNY_MAX_N = 28672  # lm_head-sized N re-reads X too often; library wins
+# per-device split-K fp32 slab scratch — each workgroup fully overwrites
+# its slab stripe, so no zeroing is needed and the address is stable
+# across hipGraph replays.
+_skinny_ws: dict = {}
+
+
+def _skinny_nsk(N: int, K: int) -> int:
+    """Split-K factor: fill ~2 workgroups per CU (512 total), keep >= 8
+    k-steps per slice. Mirrored by the launcher's validation only."""
+    tiles = N // 64
+    nsk = max(1, -(-512 // tiles))
+    nsk = min(nsk, max(1, (K // 32) // 8), 8)
+    return nsk
+
+
+_skinny_ws_retired: list = []  # keep old slabs alive for captured graphs
+
+
+def _skinny_scratch(device, numel: int):
+    ws = _skinny_ws.get(device.index)
+    if ws is None or ws.numel() < numel:
+        if ws is not None:
+            _skinny_ws_retired.append(ws)  # a hipGraph may still replay into it
+        ws = torch.empty(numel, dtype=torch.float32, device=device)
+        _skinny_ws[device.index] = ws
+    return ws
+
+
+def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """y = x @ w.T. Decode-shaped bf16 GEMMs (M <= 256) go through the
+    weight-streaming gfx950 kernel (csrc/gemm_skinny.hip); everything else
+    through the TunableOp-tuned library GEMMs."""
+    M = x.shape[0]
+    if (
+        x.is_cuda
+        and x.dtype == torch.bfloat16
+        and w.dtype == torch.bfloat16
+        and 0 < M <= _SKINNY_MAX_M
+        and w.shape[0] % 64 == 0
+        and w.shape[0] <= _SKINNY_MAX_N
+        and w.shape[1] % 32 == 0
+        and x.is_contiguous()
+        and w.is_contiguous()
+    ):
+        N, K = w.shape
+        y = torch.empty((M, N), dtype=torch.bfloat16, device=x.device)
+        nsk = _skinny_nsk(N, K)
+        ws = _skinny_scratch(x.device, nsk * M * N) if nsk > 1 else None
+        _native().gemm_skinny(y, x, w, ws, nsk)
+        return y
+    return torch.nn.functional.linear(x, w)
+
+
 def sample(
     logits: torch.Tensor,
     temperature: torch.Tensor,

@@ -20,6 +20,7 @@ hipError_t launch_kv_cache_write(const void*, const void*, void*, void*, const i
 hipError_t launch_attention_decode(void*, const void*, const void*, const void*, const int*, const int*, float, int, int, int, int, int, int, int64_t, hipStream_t);
 hipError_t launch_attention_prefill(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, float, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
 hipError_t launch_sample(int64_t*, const float*, const float*, const float*, float*, int*, int, int, hipStream_t);
+hipError_t launch_gemm_skinny(void*, float*, const void*, const void*, int, int, int, int, hipStream_t);
 }
 
 namespace {
@@ -170,6 +171,26 @@ void sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temperature,
                             B, logits.size(1), current_stream()));
 }
 
+void gemm_skinny(torch::Tensor y, torch::Tensor x, torch::Tensor w,
+                 c10::optional<torch::Tensor> workspace, int64_t nsk) {
+    check_bf16(x, "x");
+    check_bf16(w, "w");
+    check_bf16(y, "y");
+    TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && y.is_contiguous());
+    TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && y.dim() == 2);
+    const int M = x.size(0), K = x.size(1), N = w.size(0);
+    TORCH_CHECK(w.size(1) == K && y.size(0) == M && y.size(1) == N);
+    float* ws = nullptr;
+    if (workspace.has_value() && workspace->defined()) {
+        TORCH_CHECK(workspace->scalar_type() == torch::kFloat32 &&
+                    workspace->numel() >= nsk * (int64_t)M * N,
+                    "workspace must be fp32 with >= nsk*M*N elements");
+        ws = workspace->data_ptr<float>();
+    }
+    CHECK_HIP(launch_gemm_skinny(y.data_ptr(), ws, x.data_ptr(), w.data_ptr(),
+                                 M, N, K, (int)nsk, current_stream()));
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -181,6 +202,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("attention_decode", &attention_decode, "paged decode attention");
     m.def("attention_prefill", &attention_prefill, "varlen causal prefill attention");
     m.def("sample", &sample, "greedy / gumbel-max sampling");
+    m.def("gemm_skinny", &gemm_skinny,
+          "skinny-M weight-streaming GEMM y = x @ w.T (decode projections)");
 
     pybind11::class_<BlockAllocator>(m, "BlockAllocator")
         .def(pybind11::init<int64_t>())

@@ -23,6 +23,7 @@ sources = [
     os.path.join(CSRC, "attention_decode.hip"),
     os.path.join(CSRC, "attention_prefill.hip"),
     os.path.join(CSRC, "sampling.hip"),
+    os.path.join(CSRC, "gemm_skinny.hip"),
 ]
 
 setup(
@@ -33,7 +34,11 @@ setup(
             sources=sources,
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
-                "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+                # mfma-vgpr-form: keep MFMA accumulators in arch VGPRs —
+                # the AGPR form copies every accumulator through
+                # v_accvgpr_read/write pairs each loop iteration
+                "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950",
+                         "-mllvm", "-amdgpu-mfma-vgpr-form"],
             },
         )
     ],
