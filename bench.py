@@ -35,7 +35,7 @@ def parse_args():
     p.add_argument("--steps", type=int, default=3)
     p.add_argument("--warmup", type=int, default=1)
     p.add_argument("--model", default="llama-3-8b")
-    p.add_argument("--batch", type=int, default=64, help="requests per step per GPU")
+    p.add_argument("--batch", type=int, default=256, help="requests per step per GPU")
     p.add_argument("--prompt-len", type=int, default=128)
     p.add_argument("--gen-tokens", type=int, default=64)
     p.add_argument("--kv-blocks", type=int, default=None)

@@ -69,12 +69,13 @@ __global__ void attention_decode_kernel(
     const int* __restrict__ bt_global = block_tables + (size_t)seq * max_blocks;
 #define BT(idx) ((idx) < 1024 ? bt_lds[(idx)] : bt_global[(idx)])
 
-    float m[G], l[G], acc0[G], acc1[G];
+    float m[G], l[G], acc[G][4];
 #pragma unroll
     for (int g = 0; g < G; ++g) {
         m[g] = -INFINITY;
         l[g] = 0.f;
-        acc0[g] = acc1[g] = 0.f;
+#pragma unroll
+        for (int d = 0; d < 4; ++d) acc[g][d] = 0.f;
     }
 
     const int nchunks = (L + WAVE_SIZE - 1) / WAVE_SIZE;
@@ -125,56 +126,57 @@ __global__ void attention_decode_kernel(
             const float factor = (m[g] == -INFINITY) ? 0.f : __expf(m[g] - m_new);
             const float csum = wave_reduce_sum(p);
             l[g] = l[g] * factor + csum;
-            acc0[g] *= factor;
-            acc1[g] *= factor;
+#pragma unroll
+            for (int d = 0; d < 4; ++d) acc[g][d] *= factor;
             m[g] = m_new;
             p_lds[wave][g][lane] = p;
         }
         // wave-synchronous LDS use: private per wave, no barrier needed
 
-        // --- phase B: lane = dim pair; stream V rows coalesced. 4 rows per
-        // iteration: 4 independent loads in flight (the serial 1-row loop
-        // was latency-bound on the per-row L2 round trip) ---
-        const int chunk_keys = min(WAVE_SIZE, L - c * WAVE_SIZE);
-        int j = 0;
-        for (; j + 4 <= chunk_keys; j += 4) {
-            uint32_t vp[4];
+        // --- phase B: half-wave per key, lane owns a dim QUAD (8 B
+        // dwordx2 loads: half the instructions and latency batches of the
+        // earlier dim-pair/dword form). Lane l covers key j + (l>>5),
+        // dims (l&31)*4..+3; 8 keys per iteration = 4 loads in flight.
+        // Keys beyond the chunk have p == 0 (masked in phase A), so only
+        // the ADDRESS needs clamping to stay inside the paged cache. ---
+        const int dbase = (lane & 31) * 4;
+        const int khalf = lane >> 5;
+#pragma unroll 2
+        for (int j = 0; j < WAVE_SIZE; j += 8) {
+            uint2 vp[4];
 #pragma unroll
             for (int u = 0; u < 4; ++u) {
-                const int p2 = c * WAVE_SIZE + j + u;
-                const int vb = BT(p2 / block_size);
-                const int vo = p2 % block_size;
-                vp[u] = reinterpret_cast<const uint32_t*>(
-                    v_cache + (((size_t)vb * Hkv + kvh) * block_size + vo) * D)[lane];
+                const int pos = min(c * WAVE_SIZE + j + 2 * u + khalf, L - 1);
+                const int vb = BT(pos / block_size);
+                const int vo = pos % block_size;
+                vp[u] = *reinterpret_cast<const uint2*>(
+                    v_cache + (((size_t)vb * Hkv + kvh) * block_size + vo) * D +
+                    dbase);
             }
 #pragma unroll
             for (int u = 0; u < 4; ++u) {
-                float v0, v1;
-                unpack2(vp[u], v0, v1);
+                float v0, v1, v2, v3;
+                unpack2(vp[u].x, v0, v1);
+                unpack2(vp[u].y, v2, v3);
+                const int key = j + 2 * u + khalf;
 #pragma unroll
                 for (int g = 0; g < G; ++g) {
-                    const float pj = p_lds[wave][g][j + u];
-                    acc0[g] = fmaf(pj, v0, acc0[g]);
-                    acc1[g] = fmaf(pj, v1, acc1[g]);
+                    const float pj = p_lds[wave][g][key];
+                    acc[g][0] = fmaf(pj, v0, acc[g][0]);
+                    acc[g][1] = fmaf(pj, v1, acc[g][1]);
+                    acc[g][2] = fmaf(pj, v2, acc[g][2]);
+                    acc[g][3] = fmaf(pj, v3, acc[g][3]);
                 }
             }
         }
-        for (; j < chunk_keys; ++j) {
-            const int p2 = c * WAVE_SIZE + j;
-            const int vb = BT(p2 / block_size);
-            const int vo = p2 % block_size;
-            const uint32_t vpair = reinterpret_cast<const uint32_t*>(
-                v_cache + (((size_t)vb * Hkv + kvh) * block_size + vo) * D)[lane];
-            float v0, v1;
-            unpack2(vpair, v0, v1);
-#pragma unroll
-            for (int g = 0; g < G; ++g) {
-                const float pj = p_lds[wave][g][j];
-                acc0[g] = fmaf(pj, v0, acc0[g]);
-                acc1[g] = fmaf(pj, v1, acc1[g]);
-            }
-        }
     }
+
+    // fold the two half-wave key subsets (lane and lane^32 hold the same
+    // dim quad over disjoint keys)
+#pragma unroll
+    for (int g = 0; g < G; ++g)
+#pragma unroll
+        for (int d = 0; d < 4; ++d) acc[g][d] += __shfl_xor(acc[g][d], 32, WAVE_SIZE);
 
     // ---- cross-wave flash combine ----
 #pragma unroll
@@ -183,8 +185,11 @@ __global__ void attention_decode_kernel(
             m_lds[wave][g] = m[g];
             l_lds[wave][g] = l[g];
         }
-        acc_lds[wave][g][2 * lane] = acc0[g];
-        acc_lds[wave][g][2 * lane + 1] = acc1[g];
+        if (lane < 32) {
+#pragma unroll
+            for (int d = 0; d < 4; ++d)
+                acc_lds[wave][g][(lane & 31) * 4 + d] = acc[g][d];
+        }
     }
     __syncthreads();
 
