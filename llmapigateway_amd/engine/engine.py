@@ -84,7 +84,22 @@ class EngineRequest:
         self.on_token = on_token
         self.on_finish = on_finish
         self.created = time.monotonic()
+        self.prefill_start_time: Optional[float] = None
         self.first_token_time: Optional[float] = None
+        self.finished_time: Optional[float] = None
+
+    def timings(self) -> Dict[str, float]:
+        """Per-phase timing (ms) for tracing/usage reporting."""
+        out: Dict[str, float] = {}
+        if self.prefill_start_time:
+            out["queue_ms"] = round((self.prefill_start_time - self.created) * 1e3, 2)
+        if self.first_token_time:
+            out["ttft_ms"] = round((self.first_token_time - self.created) * 1e3, 2)
+        if self.finished_time and self.first_token_time and len(self.out_ids) > 1:
+            dt = self.finished_time - self.first_token_time
+            if dt > 0:
+                out["decode_tok_per_s"] = round((len(self.out_ids) - 1) / dt, 1)
+        return out
 
     @property
     def num_tokens(self) -> int:
@@ -208,6 +223,7 @@ class LLMEngine:
             self.waiting.popleft()
             req.block_table = self.kv.manager.allocate(need)
             req.state = "running"
+            req.prefill_start_time = time.monotonic()
             admitted.append(req)
         return admitted
 
@@ -427,6 +443,7 @@ class LLMEngine:
             return
         req.state = "failed" if reason == "error" else "finished"
         req.finish_reason = reason
+        req.finished_time = time.monotonic()
         self.stats["finished" if req.state == "finished" else "failed"] += 1
         if req.block_table:
             self.kv.manager.free(req.block_table)

@@ -272,6 +272,8 @@ class EngineRegistry:
             "prompt_tokens": len(req.prompt_ids),
             "completion_tokens": len(req.out_ids),
             "total_tokens": len(req.prompt_ids) + len(req.out_ids),
+            # per-phase tracing (extra field; OpenAI clients ignore it)
+            "timings": req.timings() if hasattr(req, "timings") else {},
         }
         return (
             {
@@ -348,6 +350,7 @@ class EngineRegistry:
                             "prompt_tokens": len(fin.prompt_ids),
                             "completion_tokens": len(fin.out_ids),
                             "total_tokens": len(fin.prompt_ids) + len(fin.out_ids),
+                            "timings": fin.timings() if hasattr(fin, "timings") else {},
                         }
                         yield chunk({}, finish=fin.finish_reason or "stop", usage=usage)
                         yield b"data: [DONE]\n\n"

@@ -293,6 +293,7 @@ class TPEngineClient:
                 return
             req.state = state or ("failed" if reason == "error" else "finished")
             req.finish_reason = reason
+            req.finished_time = time.monotonic()
             self.stats["finished" if req.state == "finished" else "failed"] += 1
         if req.on_finish is not None:
             try:
