@@ -23,6 +23,7 @@ from collections import deque
 from dataclasses import dataclass, field
 from typing import Callable, Dict, List, Optional
 
+import numpy as np
 import torch
 
 from .. import ops
@@ -83,6 +84,7 @@ class EngineRequest:
         self.error: Optional[str] = None
         self.on_token = on_token
         self.on_finish = on_finish
+        self.bt_slot: Optional[int] = None  # engine block-table row (running)
         self.created = time.monotonic()
         self.prefill_start_time: Optional[float] = None
         self.first_token_time: Optional[float] = None
@@ -166,6 +168,14 @@ class LLMEngine:
                 max_blocks=(self.max_model_len + block_size - 1) // block_size,
             )
 
+        # host-side persistent block tables: one stable row per admitted
+        # request, updated only when a block is appended — the decode loop
+        # gathers rows with one vectorized numpy fancy-index instead of
+        # rebuilding 256 small tensors per step
+        maxb = (self.max_model_len + block_size - 1) // block_size
+        self._bt_np = np.zeros((max_batch_size, maxb), dtype=np.int32)
+        self._slot_pool = list(range(max_batch_size - 1, -1, -1))
+
         self._gen = torch.Generator(device=self.device).manual_seed(seed ^ 0x5EED)
         self._lock = threading.Lock()
         self._work = threading.Condition(self._lock)
@@ -224,8 +234,15 @@ class LLMEngine:
             req.block_table = self.kv.manager.allocate(need)
             req.state = "running"
             req.prefill_start_time = time.monotonic()
+            req.bt_slot = self._slot_pool.pop()
+            self._bt_np[req.bt_slot, : len(req.block_table)] = req.block_table
             admitted.append(req)
         return admitted
+
+    def _free_slot(self, req: EngineRequest) -> None:
+        if req.bt_slot is not None:
+            self._slot_pool.append(req.bt_slot)
+            req.bt_slot = None
 
     def _preempt_youngest(self) -> bool:
         if not self.running:
@@ -233,6 +250,7 @@ class LLMEngine:
         victim = self.running.pop()
         self.kv.manager.free(victim.block_table)
         victim.block_table = []
+        self._free_slot(victim)
         victim.state = "waiting"
         # re-admission re-prefills prompt + generated so far
         victim.prompt_ids = victim.prompt_ids + victim.out_ids
@@ -312,7 +330,12 @@ class LLMEngine:
             while i < len(self.running):
                 req = self.running[i]
                 try:
+                    before = len(req.block_table)
                     self.kv.manager.extend(req.block_table, req.num_tokens, req.num_tokens + 1)
+                    if len(req.block_table) != before:  # keep the bt row in sync
+                        self._bt_np[req.bt_slot, before : len(req.block_table)] = (
+                            req.block_table[before:]
+                        )
                     i += 1
                 except RuntimeError:
                     if not self._preempt_youngest():
@@ -322,36 +345,33 @@ class LLMEngine:
         if not reqs:
             return 0
 
-        last_tokens = [
-            (req.out_ids[-1] if req.out_ids else req.prompt_ids[-1]) for req in reqs
-        ]
-        pos = [req.num_tokens - 1 for req in reqs]
-        slots = [req.block_table[p // bs] * bs + p % bs for req, p in zip(reqs, pos)]
-        ctx = [p + 1 for p in pos]
+        n = len(reqs)
+        last_tokens = np.fromiter(
+            ((req.out_ids[-1] if req.out_ids else req.prompt_ids[-1]) for req in reqs),
+            dtype=np.int64, count=n,
+        )
+        pos = np.fromiter((req.num_tokens - 1 for req in reqs), dtype=np.int64, count=n)
+        slot_rows = np.fromiter((req.bt_slot for req in reqs), dtype=np.intp, count=n)
+        tables_np = self._bt_np[slot_rows]  # [n, maxb] vectorized gather
+        blk = pos // bs
+        slots = tables_np[np.arange(n), blk].astype(np.int64) * bs + pos % bs
+        ctx = (pos + 1).astype(np.int32)
 
         logits = None
         if self.graph_runner is not None:
             try:
-                logits = self.graph_runner.run(
-                    last_tokens, pos, slots, [r.block_table for r in reqs], ctx
-                )
+                logits = self.graph_runner.run(last_tokens, pos, slots, tables_np, ctx)
             except Exception:
                 logger.exception("hipGraph decode failed; falling back to eager")
                 self.graph_runner = None
         if logits is None:
-            max_blocks = max(len(req.block_table) for req in reqs)
-            tables = torch.zeros(len(reqs), max_blocks, dtype=torch.int32, device=device)
-            for i, req in enumerate(reqs):
-                tables[i, : len(req.block_table)] = torch.tensor(
-                    req.block_table, dtype=torch.int32
-                )
             batch = ForwardBatch(
                 kind="decode",
-                token_ids=torch.tensor(last_tokens, dtype=torch.long, device=device),
-                positions=torch.tensor(pos, dtype=torch.long, device=device),
-                slot_mapping=torch.tensor(slots, dtype=torch.long, device=device),
-                block_tables=tables,
-                context_lens=torch.tensor(ctx, dtype=torch.int32, device=device),
+                token_ids=torch.from_numpy(last_tokens).to(device),
+                positions=torch.from_numpy(pos).to(device),
+                slot_mapping=torch.from_numpy(slots).to(device),
+                block_tables=torch.from_numpy(tables_np).to(device),
+                context_lens=torch.from_numpy(ctx).to(device),
                 logits_indices=None,
             )
             logits = self.model.forward(batch, self.kv.k_caches, self.kv.v_caches)
@@ -448,6 +468,7 @@ class LLMEngine:
         if req.block_table:
             self.kv.manager.free(req.block_table)
             req.block_table = []
+        self._free_slot(req)
         if req.on_finish is not None:
             try:
                 req.on_finish(req)

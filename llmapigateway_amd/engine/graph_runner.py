@@ -95,24 +95,25 @@ class DecodeGraphRunner:
 
     def run(
         self,
-        tokens: List[int],
-        positions: List[int],
-        slots: List[int],
-        block_tables: List[List[int]],
-        context_lens: List[int],
+        tokens,          # np.int64 [n]
+        positions,       # np.int64 [n]
+        slots,           # np.int64 [n]
+        block_tables,    # np.int32 [n, <=max_blocks]
+        context_lens,    # np.int32 [n]
     ) -> torch.Tensor:
         n = len(tokens)
         bucket = self.bucket_for(n)
 
-        self.h_token_ids[:n] = torch.tensor(tokens, dtype=torch.long)
-        self.h_positions[:n] = torch.tensor(positions, dtype=torch.long)
-        self.h_slot_mapping[:n] = torch.tensor(slots, dtype=torch.long)
-        self.h_slot_mapping[n:bucket] = -1
-        self.h_context_lens[:n] = torch.tensor(context_lens, dtype=torch.int32)
-        self.h_context_lens[n:bucket] = 1
-        self.h_block_tables[:bucket].zero_()
-        for i, bt in enumerate(block_tables):
-            self.h_block_tables[i, : len(bt)] = torch.tensor(bt, dtype=torch.int32)
+        # fill pinned staging via numpy views (no per-request tensors)
+        self.h_token_ids.numpy()[:n] = tokens
+        self.h_positions.numpy()[:n] = positions
+        sm = self.h_slot_mapping.numpy()
+        sm[:n] = slots
+        sm[n:bucket] = -1
+        cl = self.h_context_lens.numpy()
+        cl[:n] = context_lens
+        cl[n:bucket] = 1  # padding rows read 1 stale key; output discarded
+        self.h_block_tables.numpy()[:n, : block_tables.shape[1]] = block_tables
 
         self.token_ids[:bucket].copy_(self.h_token_ids[:bucket], non_blocking=True)
         self.positions[:bucket].copy_(self.h_positions[:bucket], non_blocking=True)
