@@ -286,32 +286,30 @@ class LLMEngine:
     def _prefill_step(self, reqs: List[EngineRequest]) -> int:
         device = self.device
         bs = self.kv.block_size
-        token_ids, positions, slots, cu, logits_idx = [], [], [], [0], []
-        max_len = 0
-        for req in reqs:
-            L = len(req.prompt_ids)
-            token_ids.extend(req.prompt_ids)
-            positions.extend(range(L))
-            slots.extend(
-                req.block_table[p // bs] * bs + p % bs for p in range(L)
-            )
-            cu.append(cu[-1] + L)
-            logits_idx.append(cu[-1] - 1)
-            max_len = max(max_len, L)
+        lens = [len(req.prompt_ids) for req in reqs]
+        cu = np.zeros(len(reqs) + 1, dtype=np.int32)
+        np.cumsum(lens, out=cu[1:])
+        token_ids = np.concatenate([np.asarray(r.prompt_ids, dtype=np.int64) for r in reqs])
+        positions = np.concatenate([np.arange(L, dtype=np.int64) for L in lens])
+        slots = np.empty(int(cu[-1]), dtype=np.int64)
+        for i, req in enumerate(reqs):
+            p = np.arange(lens[i], dtype=np.int64)
+            bt = np.asarray(req.block_table, dtype=np.int64)
+            slots[cu[i] : cu[i + 1]] = bt[p // bs] * bs + p % bs
+        logits_idx = (cu[1:] - 1).astype(np.int64)
+        max_len = max(lens)
 
-        tile_seq, tile_off = ops.build_prefill_tiles(
-            [len(r.prompt_ids) for r in reqs], device
-        )
+        tile_seq, tile_off = ops.build_prefill_tiles(lens, device)
         batch = ForwardBatch(
             kind="prefill",
-            token_ids=torch.tensor(token_ids, dtype=torch.long, device=device),
-            positions=torch.tensor(positions, dtype=torch.long, device=device),
-            slot_mapping=torch.tensor(slots, dtype=torch.long, device=device),
-            cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=device),
+            token_ids=torch.from_numpy(token_ids).to(device),
+            positions=torch.from_numpy(positions).to(device),
+            slot_mapping=torch.from_numpy(slots).to(device),
+            cu_seqlens=torch.from_numpy(cu).to(device),
             max_seqlen=max_len,
             tile_seq=tile_seq,
             tile_off=tile_off,
-            logits_indices=torch.tensor(logits_idx, dtype=torch.long, device=device),
+            logits_indices=torch.from_numpy(logits_idx).to(device),
         )
         logits = self.model.forward(batch, self.kv.k_caches, self.kv.v_caches)
         tokens = self._sample(logits, reqs)
