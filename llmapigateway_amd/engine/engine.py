@@ -168,6 +168,19 @@ class LLMEngine:
                 max_blocks=(self.max_model_len + block_size - 1) // block_size,
             )
 
+        # deferred sampling (GPU): the sampled-token fetch of decode step s
+        # resolves while step s+1 is already enqueued — tokens feed the
+        # next step device-side, host delivery runs one step late. Flushed
+        # at every boundary: prefill, preemption, length cap, idle.
+        self.async_sampling = self.device.type == "cuda"
+        self._pending: Optional[tuple] = None  # (reqs, n, event)
+        self._pend_tokens_dev: Optional[torch.Tensor] = None
+        self._pend_pinned: Optional[torch.Tensor] = None
+        if self.async_sampling:
+            self._pend_pinned = torch.empty(
+                max_batch_size, dtype=torch.long, pin_memory=True
+            )
+
         # host-side persistent block tables: one stable row per admitted
         # request, updated only when a block is appended — the decode loop
         # gathers rows with one vectorized numpy fancy-index instead of
@@ -259,6 +272,17 @@ class LLMEngine:
         logger.warning("Preempted %s (KV blocks exhausted)", victim.id)
         return True
 
+    # ---- deferred-sampling flush ----
+    def _flush_pending(self) -> None:
+        if self._pending is None:
+            return
+        reqs, n, event = self._pending
+        self._pending = None
+        event.synchronize()
+        tokens = self._pend_pinned[:n].tolist()
+        with self._lock:
+            self._deliver(reqs, tokens)
+
     # ---- the step ----
     def step(self) -> int:
         """Run one engine iteration. Returns number of tokens produced."""
@@ -266,11 +290,14 @@ class LLMEngine:
             admitted = self._admit()
         try:
             if admitted:
+                self._flush_pending()  # running set is about to change
                 produced = self._prefill_step(admitted)
             else:
                 with self._lock:
-                    if not self.running:
-                        return 0
+                    idle = not self.running
+                if idle:
+                    self._flush_pending()  # deliver the final in-flight step
+                    return 0
                 produced = self._decode_step()
         except Exception as e:
             logger.exception("engine step failed")
@@ -323,37 +350,74 @@ class LLMEngine:
         device = self.device
         bs = self.kv.block_size
         with self._lock:
+            reqs_now = list(self.running)
+        if not reqs_now:
+            self._flush_pending()
+            return 0
+
+        # deferred-sampling bookkeeping: if the previous decode step's
+        # tokens are still in flight, they belong to EXACTLY this request
+        # list; any mismatch or limit-crossing request forces a flush
+        if self._pending is not None:
+            same = self._pending[0] == reqs_now
+            capped = same and any(
+                len(r.out_ids) + 1 >= r.params.max_tokens
+                or r.num_tokens + 2 > self.max_model_len
+                for r in reqs_now
+            )
+            if not same or capped:
+                self._flush_pending()
+                with self._lock:
+                    reqs_now = list(self.running)
+                if not reqs_now:
+                    return 0
+        inflight = 1 if self._pending is not None else 0
+
+        with self._lock:
             # ensure every running seq has a block for the incoming token
             i = 0
             while i < len(self.running):
                 req = self.running[i]
+                nt = req.num_tokens + inflight
                 try:
                     before = len(req.block_table)
-                    self.kv.manager.extend(req.block_table, req.num_tokens, req.num_tokens + 1)
+                    self.kv.manager.extend(req.block_table, nt, nt + 1)
                     if len(req.block_table) != before:  # keep the bt row in sync
                         self._bt_np[req.bt_slot, before : len(req.block_table)] = (
                             req.block_table[before:]
                         )
                     i += 1
                 except RuntimeError:
+                    if inflight:
+                        break  # flush below, then retry the whole step
                     if not self._preempt_youngest():
                         raise
                     # if we preempted the request we were extending, skip it
+            else:
+                i = -1  # completed without break
             reqs = list(self.running)
+        if i != -1:  # needed preemption while tokens were in flight
+            self._flush_pending()
+            return self._decode_step()
         if not reqs:
             return 0
 
         n = len(reqs)
-        last_tokens = np.fromiter(
-            ((req.out_ids[-1] if req.out_ids else req.prompt_ids[-1]) for req in reqs),
-            dtype=np.int64, count=n,
+        pos = np.fromiter(
+            (req.num_tokens + inflight - 1 for req in reqs), dtype=np.int64, count=n
         )
-        pos = np.fromiter((req.num_tokens - 1 for req in reqs), dtype=np.int64, count=n)
         slot_rows = np.fromiter((req.bt_slot for req in reqs), dtype=np.intp, count=n)
         tables_np = self._bt_np[slot_rows]  # [n, maxb] vectorized gather
         blk = pos // bs
         slots = tables_np[np.arange(n), blk].astype(np.int64) * bs + pos % bs
         ctx = (pos + 1).astype(np.int32)
+        if inflight:
+            last_tokens = self._pend_tokens_dev  # device int64 [n], aligned
+        else:
+            last_tokens = np.fromiter(
+                ((req.out_ids[-1] if req.out_ids else req.prompt_ids[-1]) for req in reqs),
+                dtype=np.int64, count=n,
+            )
 
         logits = None
         if self.graph_runner is not None:
@@ -363,9 +427,14 @@ class LLMEngine:
                 logger.exception("hipGraph decode failed; falling back to eager")
                 self.graph_runner = None
         if logits is None:
+            tok_t = (
+                last_tokens
+                if isinstance(last_tokens, torch.Tensor)
+                else torch.from_numpy(last_tokens).to(device)
+            )
             batch = ForwardBatch(
                 kind="decode",
-                token_ids=torch.from_numpy(last_tokens).to(device),
+                token_ids=tok_t,
                 positions=torch.from_numpy(pos).to(device),
                 slot_mapping=torch.from_numpy(slots).to(device),
                 block_tables=torch.from_numpy(tables_np).to(device),
@@ -373,28 +442,43 @@ class LLMEngine:
                 logits_indices=None,
             )
             logits = self.model.forward(batch, self.kv.k_caches, self.kv.v_caches)
-        tokens = self._sample(logits, reqs)
-        self.stats["decode_tokens"] += len(reqs)
+
+        tokens_dev = self._sample_dev(logits, reqs)
+        self.stats["decode_tokens"] += n
+        if self.async_sampling:
+            # previous pending was either flushed or belongs to these same
+            # reqs and is already delivered-by-flush above when needed
+            self._flush_pending()
+            self._pend_pinned[:n].copy_(tokens_dev, non_blocking=True)
+            ev = torch.cuda.Event()
+            ev.record()
+            self._pending = (reqs, n, ev)
+            self._pend_tokens_dev = tokens_dev
+            return n
+        tokens = tokens_dev.tolist()
         with self._lock:
             self._deliver(reqs, tokens)
-        return len(reqs)
+        return n
 
     # ---- sampling ----
-    def _sample(self, logits: torch.Tensor, reqs: List[EngineRequest]) -> List[int]:
+    def _sample_dev(self, logits: torch.Tensor, reqs: List[EngineRequest]) -> torch.Tensor:
+        """Sample next tokens; returns an int64 device tensor (no host sync)."""
+        any_temp = any(r.params.temperature > 0 for r in reqs)
         temps = torch.tensor(
             [r.params.temperature for r in reqs], dtype=torch.float32, device=logits.device
         )
-        need_noise = bool((temps > 0).any())
         filtered = logits
-        if need_noise:
+        if any_temp:
             filtered = self._apply_topk_topp(logits.clone(), reqs)
             noise = torch.rand(
                 logits.shape, generator=self._gen, device=logits.device, dtype=torch.float32
             )
         else:
             noise = None
-        out = ops.sample(filtered, temps, noise)
-        return out.tolist()
+        return ops.sample(filtered, temps, noise)
+
+    def _sample(self, logits: torch.Tensor, reqs: List[EngineRequest]) -> List[int]:
+        return self._sample_dev(logits, reqs).tolist()
 
     def _apply_topk_topp(self, logits: torch.Tensor, reqs: List[EngineRequest]) -> torch.Tensor:
         needs = [
@@ -459,6 +543,8 @@ class LLMEngine:
     def _finish(self, req: EngineRequest, reason: str) -> None:
         if req.state in ("finished", "failed"):
             return
+        if req in self.running:  # e.g. abort of a running request
+            self.running.remove(req)
         req.state = "failed" if reason == "error" else "finished"
         req.finish_reason = reason
         req.finished_time = time.monotonic()

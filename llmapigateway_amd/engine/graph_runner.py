@@ -95,7 +95,7 @@ class DecodeGraphRunner:
 
     def run(
         self,
-        tokens,          # np.int64 [n]
+        tokens,          # np.int64 [n] | int64 CUDA tensor (deferred sampling)
         positions,       # np.int64 [n]
         slots,           # np.int64 [n]
         block_tables,    # np.int32 [n, <=max_blocks]
@@ -105,7 +105,10 @@ class DecodeGraphRunner:
         bucket = self.bucket_for(n)
 
         # fill pinned staging via numpy views (no per-request tensors)
-        self.h_token_ids.numpy()[:n] = tokens
+        if isinstance(tokens, torch.Tensor):
+            self.token_ids[:n].copy_(tokens[:n])  # device-to-device
+        else:
+            self.h_token_ids.numpy()[:n] = tokens
         self.h_positions.numpy()[:n] = positions
         sm = self.h_slot_mapping.numpy()
         sm[:n] = slots
@@ -115,7 +118,8 @@ class DecodeGraphRunner:
         cl[n:bucket] = 1  # padding rows read 1 stale key; output discarded
         self.h_block_tables.numpy()[:n, : block_tables.shape[1]] = block_tables
 
-        self.token_ids[:bucket].copy_(self.h_token_ids[:bucket], non_blocking=True)
+        if not isinstance(tokens, torch.Tensor):
+            self.token_ids[:bucket].copy_(self.h_token_ids[:bucket], non_blocking=True)
         self.positions[:bucket].copy_(self.h_positions[:bucket], non_blocking=True)
         self.slot_mapping[:bucket].copy_(self.h_slot_mapping[:bucket], non_blocking=True)
         self.context_lens[:bucket].copy_(self.h_context_lens[:bucket], non_blocking=True)
