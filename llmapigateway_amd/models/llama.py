@@ -167,8 +167,10 @@ class LlamaModel:
             q = q.view(T, c.num_heads, c.head_dim)
             k = k.view(T, c.num_kv_heads, c.head_dim)
             v = v.view(T, c.num_kv_heads, c.head_dim)
-            ops.rope_inplace(q, k, batch.positions, self.cos_sin)
-            ops.kv_cache_write(k, v, k_caches[i], v_caches[i], batch.slot_mapping)
+            ops.rope_and_kv_write(
+                q, k, v, k_caches[i], v_caches[i],
+                batch.positions, self.cos_sin, batch.slot_mapping,
+            )
 
             if batch.kind == "prefill":
                 attn = ops.attention_prefill(

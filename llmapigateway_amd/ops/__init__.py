@@ -106,6 +106,24 @@ def swiglu(x: torch.Tensor) -> torch.Tensor:
     return reference.swiglu(x)
 
 
+def rope_and_kv_write(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    positions: torch.Tensor,
+    cos_sin: torch.Tensor,
+    slot_mapping: torch.Tensor,
+) -> None:
+    """Fused rope_inplace(q, k) + kv_cache_write(k, v): one launch on GPU."""
+    if q.is_cuda:
+        _native().rope_kv_write(q, k, v, k_cache, v_cache, positions, cos_sin, slot_mapping)
+        return
+    reference.rope_inplace(q, k, positions, cos_sin)
+    reference.kv_cache_write(k, v, k_cache, v_cache, slot_mapping)
+
+
 def kv_cache_write(
     k: torch.Tensor,
     v: torch.Tensor,
@@ -259,8 +277,10 @@ __all__ = [
     "rmsnorm",
     "rmsnorm_residual",
     "rope_inplace",
+    "rope_and_kv_write",
     "swiglu",
     "kv_cache_write",
+    "linear",
     "attention_prefill",
     "attention_decode",
     "sample",

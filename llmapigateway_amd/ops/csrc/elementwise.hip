@@ -238,6 +238,84 @@ extern "C" hipError_t launch_swiglu(
 }
 
 // ---------------------------------------------------------------------------
+// Fused RoPE + KV-cache scatter: rotate q/k in place, then write the roped
+// k and v into the paged cache. One launch instead of two — at decode
+// batch sizes these kernels are dispatch-ramp-bound (~5 us floor each at
+// 256 one-per-CU workgroups), so a fused launch saves ~5 us per layer.
+// Same-workgroup global k writes are drained by __syncthreads() before the
+// cache-copy phase reads them back (same CU, own L1 — coherent).
+// ---------------------------------------------------------------------------
+
+__global__ void rope_kv_kernel(
+    bf16* __restrict__ q,
+    bf16* __restrict__ k,
+    const bf16* __restrict__ v,
+    bf16* __restrict__ k_cache,
+    bf16* __restrict__ v_cache,
+    const int64_t* __restrict__ positions,     // [T]
+    const float* __restrict__ cos_sin,         // [max_pos, D]
+    const int64_t* __restrict__ slot_mapping,  // [T]
+    int64_t q_stride,
+    int64_t k_stride,
+    int64_t v_stride,
+    int Hq,
+    int Hkv,
+    int D,
+    int block_size) {
+    const int t = blockIdx.x;
+    const int half = D / 2;
+    const int64_t pos = positions[t];
+    const float* cs = cos_sin + pos * D;
+
+    const int pairs = (Hq + Hkv) * half;
+    for (int p = threadIdx.x; p < pairs; p += blockDim.x) {
+        const int head = p / half;
+        const int d = p % half;
+        bf16* base = (head < Hq)
+                         ? q + (size_t)t * q_stride + (size_t)head * D
+                         : k + (size_t)t * k_stride + (size_t)(head - Hq) * D;
+        const float c = cs[d];
+        const float s = cs[half + d];
+        const float x1 = bf2f(base[d]);
+        const float x2 = bf2f(base[half + d]);
+        base[d] = f2bf(x1 * c - x2 * s);
+        base[half + d] = f2bf(x2 * c + x1 * s);
+    }
+
+    const int64_t slot = slot_mapping[t];
+    if (slot < 0) return;  // uniform per block: every thread sees the same t
+    __syncthreads();       // k row writes above must land before the copy
+
+    const int64_t block = slot / block_size;
+    const int64_t off = slot % block_size;
+    const int nvec = (Hkv * D) / 8;
+    const uint4* ksrc = reinterpret_cast<const uint4*>(k + (size_t)t * k_stride);
+    const uint4* vsrc = reinterpret_cast<const uint4*>(v + (size_t)t * v_stride);
+    for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+        const int h = (i * 8) / D;
+        const int d = (i * 8) % D;
+        const size_t dst = (((size_t)block * Hkv + h) * block_size + off) * D + d;
+        reinterpret_cast<uint4*>(k_cache + dst)[0] = ksrc[i];
+        reinterpret_cast<uint4*>(v_cache + dst)[0] = vsrc[i];
+    }
+}
+
+extern "C" hipError_t launch_rope_kv(
+    void* q, void* k, const void* v, void* k_cache, void* v_cache,
+    const int64_t* positions, const float* cos_sin,
+    const int64_t* slot_mapping, int T, int64_t q_stride, int64_t k_stride,
+    int64_t v_stride, int Hq, int Hkv, int D, int block_size,
+    hipStream_t stream) {
+    if (D % 8 != 0) return hipErrorInvalidValue;
+    rope_kv_kernel<<<T, 256, 0, stream>>>(
+        (bf16*)q, (bf16*)k, (const bf16*)v, (bf16*)k_cache, (bf16*)v_cache,
+        positions, cos_sin, slot_mapping, q_stride, k_stride, v_stride, Hq,
+        Hkv, D, block_size);
+    HIP_CHECK_LAST();
+    return hipSuccess;
+}
+
+// ---------------------------------------------------------------------------
 // KV-cache scatter: write k/v rows [T, Hkv, D] (strided in T) into the paged
 // cache [num_blocks, Hkv, block_size, D] at slot_mapping[t].
 // One block per token; vectorized 16 B per lane.

@@ -17,6 +17,7 @@ hipError_t launch_rmsnorm_residual(void*, const void*, void*, const void*, float
 hipError_t launch_rope(void*, void*, const int64_t*, const float*, int, int64_t, int64_t, int, int, int, hipStream_t);
 hipError_t launch_swiglu(void*, const void*, int, int, hipStream_t);
 hipError_t launch_kv_cache_write(const void*, const void*, void*, void*, const int64_t*, int, int64_t, int64_t, int, int, int, hipStream_t);
+hipError_t launch_rope_kv(void*, void*, const void*, void*, void*, const int64_t*, const float*, const int64_t*, int, int64_t, int64_t, int64_t, int, int, int, int, hipStream_t);
 hipError_t launch_attention_decode(void*, const void*, const void*, const void*, const int*, const int*, float, int, int, int, int, int, int, int64_t, hipStream_t);
 hipError_t launch_attention_prefill(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, float, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
 hipError_t launch_sample(int64_t*, const float*, const float*, const float*, float*, int*, int, int, hipStream_t);
@@ -105,6 +106,30 @@ void kv_cache_write(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
         k.data_ptr(), v.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
         slot_mapping.data_ptr<int64_t>(), T, k.stride(0), v.stride(0), Hkv,
         block_size, D, current_stream()));
+}
+
+void rope_kv_write(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                   torch::Tensor k_cache, torch::Tensor v_cache,
+                   torch::Tensor positions, torch::Tensor cos_sin,
+                   torch::Tensor slot_mapping) {
+    check_bf16(q, "q");
+    check_bf16(k, "k");
+    check_bf16(k_cache, "k_cache");
+    TORCH_CHECK(q.dim() == 3 && k.dim() == 3 && v.dim() == 3 && k_cache.dim() == 4);
+    TORCH_CHECK(q.stride(2) == 1 && k.stride(2) == 1 && v.stride(2) == 1);
+    TORCH_CHECK(q.stride(1) == q.size(2) && k.stride(1) == k.size(2) &&
+                v.stride(1) == v.size(2));
+    TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
+    TORCH_CHECK(positions.scalar_type() == torch::kInt64);
+    TORCH_CHECK(slot_mapping.scalar_type() == torch::kInt64);
+    TORCH_CHECK(cos_sin.scalar_type() == torch::kFloat32 && cos_sin.is_contiguous());
+    const int T = q.size(0);
+    CHECK_HIP(launch_rope_kv(
+        q.data_ptr(), k.data_ptr(), v.data_ptr(), k_cache.data_ptr(),
+        v_cache.data_ptr(), positions.data_ptr<int64_t>(),
+        cos_sin.data_ptr<float>(), slot_mapping.data_ptr<int64_t>(), T,
+        q.stride(0), k.stride(0), v.stride(0), q.size(1), k_cache.size(1),
+        q.size(2), k_cache.size(2), current_stream()));
 }
 
 void attention_decode(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
@@ -199,6 +224,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("rope_inplace", &rope_inplace, "rotate-half RoPE in place");
     m.def("swiglu", &swiglu, "fused SiLU-gate multiply");
     m.def("kv_cache_write", &kv_cache_write, "paged KV cache scatter");
+    m.def("rope_kv_write", &rope_kv_write, "fused RoPE + paged KV scatter");
     m.def("attention_decode", &attention_decode, "paged decode attention");
     m.def("attention_prefill", &attention_prefill, "varlen causal prefill attention");
     m.def("sample", &sample, "greedy / gumbel-max sampling");

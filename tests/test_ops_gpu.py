@@ -267,3 +267,26 @@ def test_gemm_skinny_graph_replay_rearms_workspace():
         got = ops.linear(x, w)
         err = (got.float() - ref).abs().max().item()
         assert err / (ref.abs().max().item() + 1e-3) < 0.02
+
+
+def test_rope_and_kv_write_matches_separate_ops():
+    torch.manual_seed(7)
+    T, Hq, Hkv, D, BS, NB = 9, 8, 2, 128, 16, 8
+    qkv = torch.randn(T, (Hq + 2 * Hkv) * D, dtype=torch.bfloat16, device=DEV)
+    q = qkv[:, : Hq * D].view(T, Hq, D)
+    k = qkv[:, Hq * D : (Hq + Hkv) * D].view(T, Hkv, D)
+    v = qkv[:, (Hq + Hkv) * D :].view(T, Hkv, D)
+    q2, k2, v2 = q.clone().contiguous(), k.clone().contiguous(), v.clone().contiguous()
+    kc = torch.zeros(NB, Hkv, BS, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.zeros_like(kc)
+    kc2, vc2 = kc.clone(), vc.clone()
+    pos = torch.arange(T, dtype=torch.int64, device=DEV)
+    slots = torch.tensor([3, 17, 40, -1, 9, 100, 55, 2, 77], dtype=torch.int64, device=DEV)
+    cs = ops.build_rope_cache(64, D, 10000.0, device=DEV)
+
+    ops.rope_and_kv_write(q, k, v, kc, vc, pos, cs, slots)
+    ops.rope_inplace(q2, k2, pos, cs)
+    ops.kv_cache_write(k2, v2, kc2, vc2, slots)
+    torch.cuda.synchronize()
+    assert torch.equal(q.contiguous(), q2) and torch.equal(k.contiguous(), k2)
+    assert torch.equal(kc, kc2) and torch.equal(vc, vc2)
