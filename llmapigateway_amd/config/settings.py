@@ -50,7 +50,7 @@ class Settings:
     cors_origins_raw: str = "*"
     # MI355X engine knobs (no reference equivalent — local-engine additions)
     engine_kv_block_size: int = 64
-    engine_max_batch_size: int = 64
+    engine_max_batch_size: int = 256
     engine_hbm_fraction: float = 0.90  # fraction of free HBM given to KV cache
     engine_use_hipgraph: bool = True
     extra: dict = field(default_factory=dict)
@@ -75,7 +75,7 @@ class Settings:
             log_chat_messages=_as_bool(env.get("LOG_CHAT_MESSAGES"), False),
             cors_origins_raw=env.get("CORS_ALLOW_ORIGINS", "*"),
             engine_kv_block_size=int(env.get("ENGINE_KV_BLOCK_SIZE", "64")),
-            engine_max_batch_size=int(env.get("ENGINE_MAX_BATCH_SIZE", "64")),
+            engine_max_batch_size=int(env.get("ENGINE_MAX_BATCH_SIZE", "256")),
             engine_hbm_fraction=float(env.get("ENGINE_HBM_FRACTION", "0.90")),
             engine_use_hipgraph=_as_bool(env.get("ENGINE_USE_HIPGRAPH"), True),
         )
