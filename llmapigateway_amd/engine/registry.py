@@ -191,6 +191,18 @@ class EngineRegistry:
             return None, f"Engine for provider '{provider_name}' failed to start: {e}"
         engine = handle.engine
 
+        # admission control: fail fast into the fallback chain instead of
+        # queueing unboundedly under overload (the q300 loadgen run showed
+        # requests waiting >100 s before erroring without this)
+        depth = len(getattr(engine, "waiting", None) or []) + len(
+            getattr(engine, "_reqs", None) or []
+        )
+        if depth >= self.settings.engine_max_queue:
+            return None, (
+                f"Engine for provider '{provider_name}' is overloaded "
+                f"(queue depth {depth} >= {self.settings.engine_max_queue})"
+            )
+
         messages = payload.get("messages")
         if messages:
             prompt_text = self.tokenizer.render_chat(messages)
@@ -321,10 +333,26 @@ class EngineRegistry:
                 kind, value = first_kind, first_value
                 while True:
                     if kind == "token":
+                        # coalesce every token already queued into ONE SSE
+                        # chunk: per-token chunks saturate the event loop
+                        # near ~10K tokens/s (json+send per token)
+                        toks = [value]
+                        nxt = None
+                        while True:
+                            try:
+                                k2, v2 = queue.get_nowait()
+                            except asyncio.QueueEmpty:
+                                break
+                            if k2 == "token":
+                                toks.append(v2)
+                            else:
+                                nxt = (k2, v2)
+                                break
+                        text = tokenizer.decode(toks)
                         if stops and not stopped:
                             # hold back enough text to cleanly cut a stop
                             # string before it reaches the client
-                            pending += tokenizer.decode([value])
+                            pending += text
                             cut = -1
                             for s in stops:
                                 i = pending.find(s)
@@ -340,7 +368,11 @@ class EngineRegistry:
                                 pending = pending[len(pending) - holdback :]
                                 yield chunk({"content": emit})
                         elif not stopped:
-                            yield chunk({"content": tokenizer.decode([value])})
+                            if text:
+                                yield chunk({"content": text})
+                        if nxt is not None:
+                            kind, value = nxt
+                            continue
                     else:
                         fin: EngineRequest = value
                         if pending and not stopped:

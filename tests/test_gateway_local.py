@@ -185,3 +185,20 @@ def test_local_models_listed(client):
     ids = [m["id"] for m in data]
     assert "local/simple" in ids
     assert "tiny-llama" in ids  # engine presets via local fallback provider
+
+
+def test_engine_overload_fails_fast():
+    """Admission cap: a full engine queue returns an error (fallback-able)
+    instead of queueing unboundedly (SURVEY §5 failure detection)."""
+    import asyncio
+
+    from llmapigateway_amd.config.loader import EngineSpec
+    from llmapigateway_amd.config.settings import Settings
+    from llmapigateway_amd.engine.registry import EngineRegistry
+
+    settings = Settings(engine_max_queue=0)  # every request is "overload"
+    reg = EngineRegistry(settings)
+    spec = EngineSpec(model="tiny-llama", max_batch_size=2, kv_block_size=16)
+    payload = {"model": "tiny-llama", "messages": [{"role": "user", "content": "hi"}]}
+    resp, err = asyncio.run(reg.make_request("p", spec, payload, is_streaming=False))
+    assert resp is None and "overloaded" in err
