@@ -189,6 +189,10 @@ class LLMEngine:
         self._bt_np = np.zeros((max_batch_size, maxb), dtype=np.int32)
         self._slot_pool = list(range(max_batch_size - 1, -1, -1))
 
+        # optional batched token-event sink (set by the gateway registry):
+        # called under the engine lock with [(req, token), ...] per step
+        self.batch_notifier: Optional[Callable[[list], None]] = None
+
         self._gen = torch.Generator(device=self.device).manual_seed(seed ^ 0x5EED)
         self._lock = threading.Lock()
         self._work = threading.Condition(self._lock)
@@ -509,6 +513,10 @@ class LLMEngine:
         now = time.monotonic()
         from .tokenizer import ByteTokenizer
 
+        # pass 1: record tokens; batch-notify consumers in ONE cross-thread
+        # event (the gateway registers batch_notifier — per-token
+        # call_soon_threadsafe wakeups saturate the event loop ~10K/s)
+        events = [] if self.batch_notifier is not None else None
         for req, tok in zip(reqs, tokens):
             if req.state != "running":
                 continue
@@ -522,6 +530,18 @@ class LLMEngine:
                     req.on_token(req, int(tok))
                 except Exception:
                     logger.exception("on_token callback failed for %s", req.id)
+            elif events is not None:
+                events.append((req, int(tok)))
+        if events:
+            try:
+                self.batch_notifier(events)
+            except Exception:
+                logger.exception("batch_notifier failed")
+        # pass 2: finish checks (finish callbacks are queued after the
+        # batched token event, preserving token->finish order per request)
+        for req, tok in zip(reqs, tokens):
+            if req.state != "running":
+                continue
             reason = self._finish_reason(req, int(tok))
             if reason:
                 self.running.remove(req)

@@ -171,6 +171,28 @@ class EngineRegistry:
         st = _FailureState(fail_rate, fail_requests)
         self._failures[provider] = st
 
+    @staticmethod
+    def _ensure_batch_notifier(engine, loop) -> None:
+        """Install (or re-bind after a loop change, e.g. between test
+        asyncio.run calls) the engine's batched token-event dispatcher."""
+        if getattr(engine, "_notifier_loop", None) is loop:
+            return
+
+        def dispatch(events):  # runs on the event loop
+            for req, tok in events:
+                q = getattr(req, "_aq", None)
+                if q is not None:
+                    q.put_nowait(("token", tok))
+
+        def notifier(events):
+            try:
+                loop.call_soon_threadsafe(dispatch, events)
+            except RuntimeError:  # loop closed; requests fail via timeout
+                pass
+
+        engine.batch_notifier = notifier
+        engine._notifier_loop = loop
+
     # ---- the request path ----
     async def make_request(
         self,
@@ -229,7 +251,14 @@ class EngineRegistry:
         def on_finish(req: EngineRequest) -> None:
             loop.call_soon_threadsafe(queue.put_nowait, ("finish", req))
 
-        req = EngineRequest(prompt_ids, params, on_token=on_token, on_finish=on_finish)
+        if hasattr(engine, "batch_notifier"):
+            # single-process engine: one cross-thread wakeup per STEP for
+            # all requests (engine._deliver batches), not one per token
+            self._ensure_batch_notifier(engine, loop)
+            req = EngineRequest(prompt_ids, params, on_finish=on_finish)
+            req._aq = queue
+        else:  # TP worker-group client: token rate is per-group, keep simple
+            req = EngineRequest(prompt_ids, params, on_token=on_token, on_finish=on_finish)
         try:
             engine.add_request(req)
         except Exception as e:
