@@ -60,12 +60,14 @@ DECODE_MS = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 192, 256]
 PREFILL_MS = [512, 1024, 2048, 4096, 8192]
 
 
-def shapes_for(model_name: str):
+def shapes_for(model_name: str, tp: int = 1):
     c = get_model_config(model_name)
+    if tp > 1:
+        c = c.scaled_for_tp(tp)
     h = c.hidden_size
     proj = [
-        (c.q_size + 2 * c.kv_size, h),   # qkv
-        (h, c.q_size),                   # o
+        (c.q_size + 2 * c.kv_size, h),   # qkv (column-sharded under TP)
+        (h, c.q_size),                   # o (row-sharded)
         (2 * c.intermediate_size, h),    # gate_up
         (h, c.intermediate_size),        # down
     ]
@@ -83,6 +85,10 @@ def shapes_for(model_name: str):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--models", nargs="+", default=["llama-3-8b", "mistral-7b"])
+    ap.add_argument(
+        "--tp-models", nargs="+", default=[],
+        help="model:tp entries to tune per-rank sharded shapes, e.g. llama-3-70b:8",
+    )
     ap.add_argument("--out", default=os.path.join(REPO, "llmapigateway_amd", "ops"))
     args = ap.parse_args()
 
@@ -92,6 +98,12 @@ def main():
     todo = []
     for m in args.models:
         for s in shapes_for(m):
+            if s not in seen:
+                seen.add(s)
+                todo.append(s)
+    for ent in args.tp_models:
+        m, _, tp = ent.partition(":")
+        for s in shapes_for(m, int(tp or 1)):
             if s not in seen:
                 seen.add(s)
                 todo.append(s)
