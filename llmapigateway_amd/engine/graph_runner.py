@@ -87,6 +87,16 @@ class DecodeGraphRunner:
         self.graphs[bucket] = graph
         self.graph_logits[bucket] = logits
 
+    def capture_all(self) -> None:
+        """Capture every bucket up front (serving engines: avoids a
+        multi-second capture stall on the first request of each batch
+        size). The static buffers hold benign defaults (slot=-1 skips KV
+        writes, ctx=1 reads block 0, token 0 embeds)."""
+        for b in self.buckets:
+            if b not in self.graphs:
+                self._capture(b)
+        torch.cuda.synchronize(self.device)
+
     def bucket_for(self, n: int) -> int:
         for b in self.buckets:
             if b >= n:

@@ -153,6 +153,9 @@ class EngineRegistry:
                 hbm_fraction=self.settings.engine_hbm_fraction,
                 num_blocks=None if device.startswith("cuda") else 256,
             )
+            if engine.graph_runner is not None:
+                logger.info("pre-capturing decode hipGraphs for %s", key)
+                engine.graph_runner.capture_all()
             handle = _EngineHandle(engine, key)
             self._engines[key] = handle
             logger.info("created engine %s on %s", key, device)
