@@ -87,3 +87,70 @@ def test_kv_blocks_freed(engine):
     free0 = engine.kv.manager.num_free_blocks
     engine.generate([1, 2, 3, 4], SamplingParams(max_tokens=4, ignore_eos=True))
     assert engine.kv.manager.num_free_blocks == free0
+
+
+def _run_all(engine, reqs):
+    for r in reqs:
+        engine.add_request(r)
+    steps = 0
+    while any(r.state in ("waiting", "running") for r in reqs):
+        engine.step()
+        steps += 1
+        assert steps < 10_000
+    return [r.out_ids for r in reqs]
+
+
+def test_deferred_sampling_varied_lengths_matches_sync():
+    """Requests finishing at different steps force the deferred-sampling
+    flush paths (batch-composition change, length cap); outputs must match
+    a synchronous engine bit-for-bit."""
+    kwargs = dict(
+        model=GPU_TINY, device="cuda:0", dtype=torch.bfloat16,
+        block_size=16, num_blocks=256, seed=0,
+    )
+    e_async = LLMEngine(**kwargs)
+    assert e_async.async_sampling
+    e_sync = LLMEngine(**kwargs)
+    e_sync.async_sampling = False
+
+    def mk():
+        return [
+            EngineRequest(list(range(2, 2 + 7 + i)),
+                          SamplingParams(max_tokens=m, ignore_eos=True))
+            for i, m in enumerate([3, 5, 8, 13])
+        ]
+
+    out_a = _run_all(e_async, mk())
+    out_s = _run_all(e_sync, mk())
+    assert out_a == out_s
+    assert [len(o) for o in out_a] == [3, 5, 8, 13]
+
+
+def test_deferred_sampling_mid_decode_admission():
+    """New prompts admitted between decode steps flush in-flight tokens
+    (prefill boundary) without corrupting earlier requests."""
+    engine = LLMEngine(
+        model=GPU_TINY, device="cuda:0", dtype=torch.bfloat16,
+        block_size=16, num_blocks=256, seed=0,
+    )
+    first = [
+        EngineRequest(list(range(3, 20)), SamplingParams(max_tokens=12, ignore_eos=True))
+        for _ in range(2)
+    ]
+    for r in first:
+        engine.add_request(r)
+    for _ in range(4):
+        engine.step()
+    late = [
+        EngineRequest(list(range(5, 14)), SamplingParams(max_tokens=6, ignore_eos=True))
+        for _ in range(2)
+    ]
+    for r in late:
+        engine.add_request(r)
+    while any(r.state in ("waiting", "running") for r in first + late):
+        engine.step()
+    assert all(len(r.out_ids) == 12 for r in first)
+    assert all(len(r.out_ids) == 6 for r in late)
+    # identical prompts decode identically (greedy, shared weights)
+    assert first[0].out_ids == first[1].out_ids
+    assert late[0].out_ids == late[1].out_ids
