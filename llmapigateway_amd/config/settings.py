@@ -53,6 +53,7 @@ class Settings:
     engine_max_batch_size: int = 256
     engine_hbm_fraction: float = 0.90  # fraction of free HBM given to KV cache
     engine_max_queue: int = 2048       # admission cap: overload fails fast to fallback
+    engine_prefix_caching: bool = True # content-addressed KV reuse of shared prompt prefixes
     engine_use_hipgraph: bool = True
     extra: dict = field(default_factory=dict)
 
@@ -79,6 +80,7 @@ class Settings:
             engine_max_batch_size=int(env.get("ENGINE_MAX_BATCH_SIZE", "256")),
             engine_hbm_fraction=float(env.get("ENGINE_HBM_FRACTION", "0.90")),
             engine_max_queue=int(env.get("ENGINE_MAX_QUEUE", "2048")),
+            engine_prefix_caching=_as_bool(env.get("ENGINE_PREFIX_CACHING"), True),
             engine_use_hipgraph=_as_bool(env.get("ENGINE_USE_HIPGRAPH"), True),
         )
 

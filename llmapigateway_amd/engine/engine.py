@@ -85,6 +85,7 @@ class EngineRequest:
         self.on_token = on_token
         self.on_finish = on_finish
         self.bt_slot: Optional[int] = None  # engine block-table row (running)
+        self.num_cached = 0  # prompt tokens served from the prefix cache
         self.created = time.monotonic()
         self.prefill_start_time: Optional[float] = None
         self.first_token_time: Optional[float] = None
@@ -127,6 +128,7 @@ class LLMEngine:
         tp_group: Optional[object] = None,
         tp_rank: int = 0,
         tp_size: int = 1,
+        prefix_caching: bool = False,
     ):
         full_config = get_model_config(model) if isinstance(model, str) else model
         self.full_config = full_config
@@ -150,7 +152,11 @@ class LLMEngine:
             num_blocks = PagedKVCache.fit_num_blocks(
                 config, block_size, self.device, dtype, hbm_fraction
             )
-        self.kv = PagedKVCache(config, num_blocks, block_size, self.device, dtype)
+        self.kv = PagedKVCache(
+            config, num_blocks, block_size, self.device, dtype,
+            prefix_caching=prefix_caching,
+        )
+        self.prefix_caching = prefix_caching
 
         # hipGraph-captured decode (GPU only; TP group ops are capturable
         # with RCCL but kept off by default under TP until validated)
@@ -248,7 +254,13 @@ class LLMEngine:
             if not self.kv.manager.can_allocate(need + 1):
                 break
             self.waiting.popleft()
-            req.block_table = self.kv.manager.allocate(need)
+            if self.prefix_caching:
+                req.block_table, req.num_cached = (
+                    self.kv.manager.allocate_with_prefix(req.prompt_ids)
+                )
+            else:
+                req.block_table = self.kv.manager.allocate(need)
+                req.num_cached = 0
             req.state = "running"
             req.prefill_start_time = time.monotonic()
             req.bt_slot = self._slot_pool.pop()
@@ -317,18 +329,32 @@ class LLMEngine:
     def _prefill_step(self, reqs: List[EngineRequest]) -> int:
         device = self.device
         bs = self.kv.block_size
-        lens = [len(req.prompt_ids) for req in reqs]
+        # with prefix caching, only the un-cached suffix is prefilled; the
+        # cached context is read from the paged cache by the kernel
+        ncs = [getattr(req, "num_cached", 0) for req in reqs]
+        lens = [len(req.prompt_ids) - nc for req, nc in zip(reqs, ncs)]
         cu = np.zeros(len(reqs) + 1, dtype=np.int32)
         np.cumsum(lens, out=cu[1:])
-        token_ids = np.concatenate([np.asarray(r.prompt_ids, dtype=np.int64) for r in reqs])
-        positions = np.concatenate([np.arange(L, dtype=np.int64) for L in lens])
+        token_ids = np.concatenate(
+            [np.asarray(r.prompt_ids[nc:], dtype=np.int64) for r, nc in zip(reqs, ncs)]
+        )
+        positions = np.concatenate(
+            [np.arange(nc, nc + L, dtype=np.int64) for L, nc in zip(lens, ncs)]
+        )
         slots = np.empty(int(cu[-1]), dtype=np.int64)
         for i, req in enumerate(reqs):
-            p = np.arange(lens[i], dtype=np.int64)
+            p = np.arange(ncs[i], ncs[i] + lens[i], dtype=np.int64)
             bt = np.asarray(req.block_table, dtype=np.int64)
             slots[cu[i] : cu[i + 1]] = bt[p // bs] * bs + p % bs
         logits_idx = (cu[1:] - 1).astype(np.int64)
         max_len = max(lens)
+
+        cached_lens_t = None
+        block_tables_t = None
+        if any(ncs):
+            rows = np.fromiter((req.bt_slot for req in reqs), dtype=np.intp, count=len(reqs))
+            block_tables_t = torch.from_numpy(self._bt_np[rows]).to(device)
+            cached_lens_t = torch.from_numpy(np.asarray(ncs, dtype=np.int32)).to(device)
 
         tile_seq, tile_off = ops.build_prefill_tiles(lens, device)
         batch = ForwardBatch(
@@ -340,11 +366,18 @@ class LLMEngine:
             max_seqlen=max_len,
             tile_seq=tile_seq,
             tile_off=tile_off,
+            block_tables=block_tables_t,
+            cached_lens=cached_lens_t,
             logits_indices=torch.from_numpy(logits_idx).to(device),
         )
         logits = self.model.forward(batch, self.kv.k_caches, self.kv.v_caches)
         tokens = self._sample(logits, reqs)
         self.stats["prefill_tokens"] += len(token_ids)
+        if self.prefix_caching:
+            # KV for the suffix is now written (stream-ordered before any
+            # later forward): make the full prompt blocks reusable
+            for req in reqs:
+                self.kv.manager.register_prefix(req.prompt_ids, req.block_table)
         with self._lock:
             self.running.extend(reqs)
             self._deliver(reqs, tokens)

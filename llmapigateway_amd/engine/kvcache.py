@@ -56,11 +56,30 @@ def make_block_allocator(num_blocks: int):
 
 
 class BlockManager:
-    """Per-sequence block-table bookkeeping on top of the allocator."""
+    """Per-sequence block-table bookkeeping on top of the allocator.
 
-    def __init__(self, num_blocks: int, block_size: int):
+    With ``prefix_caching=True``, FULL prompt blocks are content-addressed
+    by a (parent_block, token_bytes) chain key: a new request whose prompt
+    shares a cached prefix reuses those blocks (refcounted) and only
+    prefills the suffix; refcount-0 cached blocks stay resident (data
+    retained) in an LRU and are evicted only under allocation pressure.
+    Shared blocks are never written again: only FULL prompt blocks are
+    registered, and generation always appends into a fresh block.
+    """
+
+    def __init__(self, num_blocks: int, block_size: int, prefix_caching: bool = False):
         self.block_size = block_size
         self.allocator = make_block_allocator(num_blocks)
+        self.prefix_caching = prefix_caching
+        # cached-prefix state (block ids are allocator-owned while cached)
+        self._table: dict = {}      # (parent_block, chunk_bytes) -> block_id
+        self._block_key: dict = {}  # block_id -> its table key
+        self._refs: dict = {}       # block_id -> request refcount
+        from collections import OrderedDict
+
+        self._evictable: "OrderedDict[int, None]" = OrderedDict()
+        self.stats_prefix_hits = 0
+        self.stats_prefix_tokens = 0
 
     @property
     def num_free_blocks(self) -> int:
@@ -70,19 +89,100 @@ class BlockManager:
         return (num_tokens + self.block_size - 1) // self.block_size
 
     def can_allocate(self, num_tokens: int) -> bool:
-        return self.blocks_needed(num_tokens) <= self.allocator.num_free()
+        avail = self.allocator.num_free() + len(self._evictable)
+        return self.blocks_needed(num_tokens) <= avail
+
+    def _alloc_raw(self, n: int) -> List[int]:
+        while self.allocator.num_free() < n and self._evictable:
+            blk, _ = self._evictable.popitem(last=False)  # oldest first
+            key = self._block_key.pop(blk)
+            del self._table[key]
+            self._refs.pop(blk, None)
+            self.allocator.free([blk])
+        return list(self.allocator.allocate(n))
 
     def allocate(self, num_tokens: int) -> List[int]:
-        return list(self.allocator.allocate(self.blocks_needed(num_tokens)))
+        return self._alloc_raw(self.blocks_needed(num_tokens))
+
+    # ---- prefix caching ----
+    def _chunk_key(self, parent: int, prompt_ids, i: int):
+        bs = self.block_size
+        import numpy as _np
+
+        chunk = _np.asarray(prompt_ids[i * bs : (i + 1) * bs], dtype=_np.int64)
+        return (parent, chunk.tobytes())
+
+    def _ref(self, blk: int) -> None:
+        self._refs[blk] = self._refs.get(blk, 0) + 1
+        self._evictable.pop(blk, None)
+
+    def _unref(self, blk: int) -> None:
+        self._refs[blk] -= 1
+        if self._refs[blk] == 0:
+            self._evictable[blk] = None  # LRU tail (data stays resident)
+
+    def allocate_with_prefix(self, prompt_ids) -> tuple:
+        """Returns (block_table, num_cached_tokens)."""
+        bs = self.block_size
+        cached: List[int] = []
+        parent = -1
+        max_full = (len(prompt_ids) - 1) // bs  # leave >= 1 token to prefill
+        for i in range(max_full):
+            blk = self._table.get(self._chunk_key(parent, prompt_ids, i))
+            if blk is None:
+                break
+            cached.append(blk)
+            parent = blk
+        total = self.blocks_needed(len(prompt_ids))
+        try:
+            private = self._alloc_raw(total - len(cached))
+        except RuntimeError:
+            raise
+        for blk in cached:
+            self._ref(blk)
+        if cached:
+            self.stats_prefix_hits += 1
+            self.stats_prefix_tokens += len(cached) * bs
+        return cached + private, len(cached) * bs
+
+    def register_prefix(self, prompt_ids, block_table: List[int]) -> None:
+        """Register a request's FULL prompt blocks (KV now written) so later
+        prompts can reuse them. Call after the prefill forward is issued."""
+        if not self.prefix_caching:
+            return
+        bs = self.block_size
+        parent = -1
+        for i in range(len(prompt_ids) // bs):
+            key = self._chunk_key(parent, prompt_ids, i)
+            existing = self._table.get(key)
+            if existing is not None:
+                parent = existing
+                continue
+            blk = block_table[i]
+            if blk in self._block_key:
+                # this block is already registered under a different chain
+                # (it IS a cached block we reused) — just walk on
+                parent = blk
+                continue
+            self._table[key] = blk
+            self._block_key[blk] = key
+            self._refs[blk] = self._refs.get(blk, 0) + 1  # owner's reference
+            parent = blk
 
     def extend(self, block_table: List[int], old_tokens: int, new_tokens: int) -> None:
         need = self.blocks_needed(new_tokens) - len(block_table)
         if need > 0:
-            block_table.extend(self.allocator.allocate(need))
+            block_table.extend(self._alloc_raw(need))
 
     def free(self, block_table: List[int]) -> None:
-        if block_table:
-            self.allocator.free(list(block_table))
+        plain = []
+        for blk in block_table:
+            if blk in self._block_key:
+                self._unref(blk)
+            else:
+                plain.append(blk)
+        if plain:
+            self.allocator.free(plain)
 
 
 class PagedKVCache:
@@ -93,6 +193,7 @@ class PagedKVCache:
         block_size: int,
         device: torch.device | str,
         dtype: torch.dtype = torch.bfloat16,
+        prefix_caching: bool = False,
     ):
         self.config = config
         self.block_size = block_size
@@ -104,7 +205,7 @@ class PagedKVCache:
         self.v_caches = [
             torch.zeros(shape, dtype=dtype, device=device) for _ in range(config.num_layers)
         ]
-        self.manager = BlockManager(num_blocks, block_size)
+        self.manager = BlockManager(num_blocks, block_size, prefix_caching=prefix_caching)
 
     @staticmethod
     def block_bytes(config: ModelConfig, block_size: int, dtype: torch.dtype) -> int:

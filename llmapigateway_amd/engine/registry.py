@@ -152,6 +152,7 @@ class EngineRegistry:
                 max_batch_size=spec.max_batch_size or self.settings.engine_max_batch_size,
                 hbm_fraction=self.settings.engine_hbm_fraction,
                 num_blocks=None if device.startswith("cuda") else 256,
+                prefix_caching=self.settings.engine_prefix_caching,
             )
             if engine.graph_runner is not None:
                 logger.info("pre-capturing decode hipGraphs for %s", key)

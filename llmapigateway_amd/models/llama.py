@@ -48,6 +48,9 @@ class ForwardBatch:
     # decode:
     block_tables: Optional[torch.Tensor] = None  # [B, max_blocks] int32
     context_lens: Optional[torch.Tensor] = None  # [B] int32
+    # prefill with cached prefix (prefix caching): per-seq number of
+    # positions already in the paged cache (block_tables set too)
+    cached_lens: Optional[torch.Tensor] = None   # [B] int32
     # rows of the flat batch at which logits are needed (last token per seq)
     logits_indices: Optional[torch.Tensor] = None  # [B] int64
 
@@ -176,6 +179,10 @@ class LlamaModel:
                 attn = ops.attention_prefill(
                     q, k, v, batch.cu_seqlens, batch.max_seqlen, self.scale,
                     tile_seq=batch.tile_seq, tile_off=batch.tile_off,
+                    k_cache=k_caches[i] if batch.cached_lens is not None else None,
+                    v_cache=v_caches[i] if batch.cached_lens is not None else None,
+                    block_tables=batch.block_tables,
+                    cached_lens=batch.cached_lens,
                 )
             else:
                 attn = ops.attention_decode(

@@ -163,7 +163,15 @@ def attention_prefill(
     causal: bool = True,
     tile_seq: Optional[torch.Tensor] = None,
     tile_off: Optional[torch.Tensor] = None,
+    k_cache: Optional[torch.Tensor] = None,
+    v_cache: Optional[torch.Tensor] = None,
+    block_tables: Optional[torch.Tensor] = None,
+    cached_lens: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
+    """Varlen causal prefill. With cached_lens set, each sequence also
+    attends (unmasked) to its first cached_lens[i] positions read from the
+    paged KV cache via block_tables — the prefix-caching / chunked-prefill
+    context phase."""
     if scale is None:
         scale = float(q.shape[-1]) ** -0.5
     if q.is_cuda:
@@ -175,10 +183,15 @@ def attention_prefill(
             tile_seq, tile_off = build_prefill_tiles(seqlens, q.device)
         out = torch.empty_like(q)
         _native().attention_prefill(
-            out, q, k, v, cu_seqlens.int(), tile_seq, tile_off, float(scale)
+            out, q, k, v, cu_seqlens.int(), tile_seq, tile_off, float(scale),
+            k_cache, v_cache, block_tables, cached_lens,
         )
         return out
-    return reference.attention_prefill(q, k, v, cu_seqlens, scale, causal)
+    return reference.attention_prefill(
+        q, k, v, cu_seqlens, scale, causal,
+        k_cache=k_cache, v_cache=v_cache,
+        block_tables=block_tables, cached_lens=cached_lens,
+    )
 
 
 def attention_decode(

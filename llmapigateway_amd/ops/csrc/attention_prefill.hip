@@ -45,7 +45,16 @@ __global__ void attention_prefill_kernel(
     int Hkv,
     int64_t q_stride,
     int64_t k_stride,
-    int64_t v_stride) {
+    int64_t v_stride,
+    // cached-context phase (prefix caching / chunked prefill): per-seq
+    // prior KV already resident in the paged cache; every fresh q row
+    // attends to ALL of it (no mask — cached positions precede the tile)
+    const bf16* __restrict__ k_cache,        // [NB, Hkv, BS, D] or null
+    const bf16* __restrict__ v_cache,
+    const int* __restrict__ block_tables,    // [B, max_blocks] or null
+    const int* __restrict__ cached_lens,     // [B] or null
+    int block_size,
+    int max_blocks) {
     const int tile = blockIdx.x;
     const int head = blockIdx.y;
     const int kvh = head / (Hq / Hkv);
@@ -53,6 +62,7 @@ __global__ void attention_prefill_kernel(
     const int tile0 = tile_off[tile];
     const int seq_start = cu_seqlens[seq];
     const int seq_len = cu_seqlens[seq + 1] - seq_start;
+    const int cached = (cached_lens != nullptr) ? cached_lens[seq] : 0;
 
     const int tid = threadIdx.x;
     const int lane = tid & (WAVE_SIZE - 1);
@@ -91,6 +101,104 @@ __global__ void attention_prefill_kernel(
 #pragma unroll
     for (int d = 0; d < 8; ++d) accO[d] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
+    // ---- phase 0: cached context from the paged KV cache (unmasked) ----
+    const int n_cached_tiles = (cached + KVTILE - 1) / KVTILE;
+    const int* bt = (block_tables != nullptr)
+                        ? block_tables + (size_t)seq * max_blocks
+                        : nullptr;
+    for (int jt = 0; jt < n_cached_tiles; ++jt) {
+        const int key0 = jt * KVTILE;
+        const int keys_here = min(KVTILE, cached - key0);
+
+        __syncthreads();
+        for (int i = tid; i < KVTILE * (PF_D / 8); i += PF_WAVES * WAVE_SIZE) {
+            const int kk = i / (PF_D / 8);
+            const int d0 = (i % (PF_D / 8)) * 8;
+            uint4 kraw = {0, 0, 0, 0}, vraw = {0, 0, 0, 0};
+            if (kk < keys_here) {
+                const int pos = key0 + kk;
+                const size_t base =
+                    (((size_t)bt[pos / block_size] * Hkv + kvh) * block_size +
+                     pos % block_size) *
+                    PF_D;
+                kraw = *reinterpret_cast<const uint4*>(k_cache + base + d0);
+                vraw = *reinterpret_cast<const uint4*>(v_cache + base + d0);
+            }
+            *reinterpret_cast<uint4*>(&k_lds[kk][d0]) = kraw;
+            const bf16* v8 = reinterpret_cast<const bf16*>(&vraw);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) vt_lds[d0 + j][kk] = v8[j];
+        }
+        __syncthreads();
+
+        f32x4 s[4];
+#pragma unroll
+        for (int sub = 0; sub < 4; ++sub) {
+            s[sub] = (f32x4){0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int kc = 0; kc < 4; ++kc) {
+                const bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+                    &k_lds[sub * 16 + lrow][kc * 32 + lk * 8]);
+                s[sub] = mfma16(q_frag[kc], bfrag, s[sub]);
+            }
+        }
+
+        float p_val[4][4];
+        float row_max[4], row_sum[4];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int qrow = tile0 + wave * 16 + lk * 4 + r;
+            float mx = NEG_INF;
+#pragma unroll
+            for (int sub = 0; sub < 4; ++sub) {
+                const int key = key0 + sub * 16 + lrow;
+                float val = s[sub][r] * scale;
+                // only key-validity masking: all cached keys precede
+                // every fresh q row, so no causal term
+                if (key >= cached || qrow >= seq_len) val = NEG_INF;
+                p_val[sub][r] = val;
+                mx = fmaxf(mx, val);
+            }
+            row_max[r] = group16_reduce_max(mx);
+        }
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const float m_new = fmaxf(m_st[r], row_max[r]);
+            const float factor = (m_st[r] <= NEG_INF) ? 0.f : __expf(m_st[r] - m_new);
+            float sum = 0.f;
+#pragma unroll
+            for (int sub = 0; sub < 4; ++sub) {
+                const float pv = (m_new <= NEG_INF) ? 0.f : __expf(p_val[sub][r] - m_new);
+                p_val[sub][r] = pv;
+                sum += pv;
+            }
+            row_sum[r] = group16_reduce_sum(sum);
+            l_st[r] = l_st[r] * factor + row_sum[r];
+            m_st[r] = m_new;
+#pragma unroll
+            for (int d = 0; d < 8; ++d) accO[d][r] *= factor;
+        }
+
+#pragma unroll
+        for (int sub = 0; sub < 4; ++sub)
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+                p_lds[wave][lk * 4 + r][sub * 16 + lrow] = f2bf(p_val[sub][r]);
+
+#pragma unroll
+        for (int dsub = 0; dsub < 8; ++dsub) {
+#pragma unroll
+            for (int kc = 0; kc < 2; ++kc) {
+                const bf16x8 pa = *reinterpret_cast<const bf16x8*>(
+                    &p_lds[wave][lrow][kc * 32 + lk * 8]);
+                const bf16x8 vb = *reinterpret_cast<const bf16x8*>(
+                    &vt_lds[dsub * 16 + lrow][kc * 32 + lk * 8]);
+                accO[dsub] = mfma16(pa, vb, accO[dsub]);
+            }
+        }
+    }
+
+    // ---- fresh KV (causal within the new tokens) ----
     const int kv_limit = min(seq_len, tile0 + QTILE);  // causal: keys < limit
     const int n_kv_tiles = (kv_limit + KVTILE - 1) / KVTILE;
 
@@ -209,13 +317,17 @@ extern "C" hipError_t launch_attention_prefill(
     void* out, const void* q, const void* k, const void* v,
     const int* cu_seqlens, const int* tile_seq, const int* tile_off,
     int ntiles, float scale, int Hq, int Hkv, int D, int64_t q_stride,
-    int64_t k_stride, int64_t v_stride, hipStream_t stream) {
+    int64_t k_stride, int64_t v_stride, const void* k_cache,
+    const void* v_cache, const int* block_tables, const int* cached_lens,
+    int block_size, int max_blocks, hipStream_t stream) {
     if (D != PF_D) return hipErrorNotSupported;
     dim3 grid(ntiles, Hq);
     dim3 block(PF_WAVES * WAVE_SIZE);
     attention_prefill_kernel<<<grid, block, 0, stream>>>(
         (bf16*)out, (const bf16*)q, (const bf16*)k, (const bf16*)v, cu_seqlens,
-        tile_seq, tile_off, scale, Hq, Hkv, q_stride, k_stride, v_stride);
+        tile_seq, tile_off, scale, Hq, Hkv, q_stride, k_stride, v_stride,
+        (const bf16*)k_cache, (const bf16*)v_cache, block_tables, cached_lens,
+        block_size, max_blocks);
     HIP_CHECK_LAST();
     return hipSuccess;
 }

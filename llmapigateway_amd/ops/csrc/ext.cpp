@@ -19,7 +19,7 @@ hipError_t launch_swiglu(void*, const void*, int, int, hipStream_t);
 hipError_t launch_kv_cache_write(const void*, const void*, void*, void*, const int64_t*, int, int64_t, int64_t, int, int, int, hipStream_t);
 hipError_t launch_rope_kv(void*, void*, const void*, void*, void*, const int64_t*, const float*, const int64_t*, int, int64_t, int64_t, int64_t, int, int, int, int, hipStream_t);
 hipError_t launch_attention_decode(void*, const void*, const void*, const void*, const int*, const int*, float, int, int, int, int, int, int, int64_t, hipStream_t);
-hipError_t launch_attention_prefill(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, float, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
+hipError_t launch_attention_prefill(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, float, int, int, int, int64_t, int64_t, int64_t, const void*, const void*, const int*, const int*, int, int, hipStream_t);
 hipError_t launch_sample(int64_t*, const float*, const float*, const float*, float*, int*, int, int, hipStream_t);
 hipError_t launch_gemm_skinny(void*, float*, const void*, const void*, int, int, int, int, hipStream_t);
 }
@@ -156,7 +156,11 @@ void attention_decode(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
 void attention_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                        torch::Tensor v, torch::Tensor cu_seqlens,
                        torch::Tensor tile_seq, torch::Tensor tile_off,
-                       double scale) {
+                       double scale,
+                       c10::optional<torch::Tensor> k_cache,
+                       c10::optional<torch::Tensor> v_cache,
+                       c10::optional<torch::Tensor> block_tables,
+                       c10::optional<torch::Tensor> cached_lens) {
     check_bf16(q, "q");
     check_bf16(out, "out");
     TORCH_CHECK(out.is_contiguous());
@@ -167,11 +171,30 @@ void attention_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
     TORCH_CHECK(tile_seq.scalar_type() == torch::kInt32 &&
                 tile_off.scalar_type() == torch::kInt32);
     const int ntiles = tile_seq.size(0);
+    const void* kc = nullptr;
+    const void* vc = nullptr;
+    const int* bt = nullptr;
+    const int* cl = nullptr;
+    int block_size = 0, max_blocks = 0;
+    if (cached_lens.has_value() && cached_lens->defined()) {
+        TORCH_CHECK(k_cache.has_value() && block_tables.has_value());
+        TORCH_CHECK(cached_lens->scalar_type() == torch::kInt32);
+        TORCH_CHECK(block_tables->scalar_type() == torch::kInt32 &&
+                    block_tables->is_contiguous());
+        TORCH_CHECK(k_cache->is_contiguous() && v_cache->is_contiguous());
+        kc = k_cache->data_ptr();
+        vc = v_cache->data_ptr();
+        bt = block_tables->data_ptr<int>();
+        cl = cached_lens->data_ptr<int>();
+        block_size = k_cache->size(2);
+        max_blocks = block_tables->size(1);
+    }
     CHECK_HIP(launch_attention_prefill(
         out.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
         cu_seqlens.data_ptr<int>(), tile_seq.data_ptr<int>(),
         tile_off.data_ptr<int>(), ntiles, (float)scale, q.size(1), k.size(1),
-        q.size(2), q.stride(0), k.stride(0), v.stride(0), current_stream()));
+        q.size(2), q.stride(0), k.stride(0), v.stride(0), kc, vc, bt, cl,
+        block_size, max_blocks, current_stream()));
 }
 
 void sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temperature,

@@ -98,8 +98,14 @@ def attention_prefill(
     cu_seqlens: torch.Tensor,
     scale: Optional[float] = None,
     causal: bool = True,
+    k_cache: Optional[torch.Tensor] = None,
+    v_cache: Optional[torch.Tensor] = None,
+    block_tables: Optional[torch.Tensor] = None,
+    cached_lens: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
-    """Varlen causal attention with GQA over fresh (non-cached) K/V."""
+    """Varlen causal attention with GQA over fresh K/V; with cached_lens,
+    each sequence also attends (unmasked) to its cached prefix gathered
+    from the paged cache (prefix caching / chunked prefill)."""
     T, Hq, D = q.shape
     Hkv = k.shape[1]
     group = Hq // Hkv
@@ -109,13 +115,32 @@ def attention_prefill(
     cu = cu_seqlens.tolist()
     for i in range(len(cu) - 1):
         s, e = cu[i], cu[i + 1]
+        L = e - s
         qi = q[s:e].float()  # [L, Hq, D]
         ki = k[s:e].float().repeat_interleave(group, dim=1)
         vi = v[s:e].float().repeat_interleave(group, dim=1)
+        nc = int(cached_lens[i]) if cached_lens is not None else 0
+        if nc > 0:
+            bs = k_cache.shape[2]
+            rows_k, rows_v = [], []
+            for pos in range(nc):
+                blk = int(block_tables[i, pos // bs])
+                rows_k.append(k_cache[blk, :, pos % bs, :])
+                rows_v.append(v_cache[blk, :, pos % bs, :])
+            kc = torch.stack(rows_k).float().repeat_interleave(group, dim=1)
+            vc = torch.stack(rows_v).float().repeat_interleave(group, dim=1)
+            ki = torch.cat([kc, ki], dim=0)
+            vi = torch.cat([vc, vi], dim=0)
         scores = torch.einsum("qhd,khd->hqk", qi, ki) * scale
         if causal:
-            L = e - s
-            mask = torch.triu(torch.ones(L, L, dtype=torch.bool, device=q.device), diagonal=1)
+            mask = torch.triu(
+                torch.ones(L, L, dtype=torch.bool, device=q.device), diagonal=1
+            )
+            if nc > 0:  # cached keys precede every fresh row: never masked
+                mask = torch.cat(
+                    [torch.zeros(L, nc, dtype=torch.bool, device=q.device), mask],
+                    dim=1,
+                )
             scores.masked_fill_(mask, float("-inf"))
         p = torch.softmax(scores, dim=-1)
         out[s:e] = torch.einsum("hqk,khd->qhd", p, vi).to(q.dtype)
