@@ -332,11 +332,17 @@ def test_attention_prefill_cached_context(cached, fresh):
 
 def test_engine_prefix_hit_matches_uncached_gpu():
     from llmapigateway_amd.engine import LLMEngine, SamplingParams
+    from llmapigateway_amd.models.configs import ModelConfig
 
+    cfg = ModelConfig(
+        name="gpu-tiny-pc", hidden_size=512, intermediate_size=1024,
+        num_layers=2, num_heads=4, num_kv_heads=2, vocab_size=2048,
+        head_dim=128, rope_theta=10000.0, max_positions=1024,
+    )
     prompt = list(range(7, 87))  # 80 tokens -> 4 full blocks of 16
     def mk(prefix):
         return LLMEngine(
-            model=GPU_TINY, device="cuda:0", dtype=torch.bfloat16,
+            model=cfg, device="cuda:0", dtype=torch.bfloat16,
             block_size=16, num_blocks=128, seed=5, prefix_caching=prefix,
         )
     clean = mk(False)
