@@ -129,6 +129,7 @@ class LLMEngine:
         tp_rank: int = 0,
         tp_size: int = 1,
         prefix_caching: bool = False,
+        prefill_budget: int = 8192,
     ):
         full_config = get_model_config(model) if isinstance(model, str) else model
         self.full_config = full_config
@@ -203,6 +204,8 @@ class LLMEngine:
         self._lock = threading.Lock()
         self._work = threading.Condition(self._lock)
         self.waiting: deque[EngineRequest] = deque()
+        self.prefilling: List[EngineRequest] = []  # chunked prefill in progress
+        self.prefill_budget = prefill_budget
         self.running: List[EngineRequest] = []
         self.stats: Dict[str, float] = {
             "requests": 0,
@@ -233,24 +236,27 @@ class LLMEngine:
 
     def has_work(self) -> bool:
         with self._lock:
-            return bool(self.waiting or self.running)
+            return bool(self.waiting or self.prefilling or self.running)
 
     def wait_for_work(self, timeout: float = 0.2) -> bool:
         with self._work:
-            if self.waiting or self.running:
+            if self.waiting or self.prefilling or self.running:
                 return True
             self._work.wait(timeout)
-            return bool(self.waiting or self.running)
+            return bool(self.waiting or self.prefilling or self.running)
 
     # ---- scheduling ----
     def _admit(self) -> List[EngineRequest]:
+        """Move waiting requests into the prefilling set (blocks for the
+        whole prompt are allocated up front; KV fills chunk by chunk)."""
         admitted: List[EngineRequest] = []
-        budget_tokens = 8192  # prefill-batch token budget per step
-        while self.waiting and len(self.running) + len(admitted) < self.max_batch_size:
+        while (
+            self.waiting
+            and len(self.running) + len(self.prefilling) + len(admitted)
+            < self.max_batch_size
+        ):
             req = self.waiting[0]
             need = len(req.prompt_ids)
-            if admitted and sum(len(r.prompt_ids) for r in admitted) + need > budget_tokens:
-                break
             if not self.kv.manager.can_allocate(need + 1):
                 break
             self.waiting.popleft()
@@ -263,6 +269,7 @@ class LLMEngine:
                 req.num_cached = 0
             req.state = "running"
             req.prefill_start_time = time.monotonic()
+            req.prefill_pos = req.num_cached
             req.bt_slot = self._slot_pool.pop()
             self._bt_np[req.bt_slot, : len(req.block_table)] = req.block_table
             admitted.append(req)
@@ -303,11 +310,12 @@ class LLMEngine:
     def step(self) -> int:
         """Run one engine iteration. Returns number of tokens produced."""
         with self._lock:
-            admitted = self._admit()
+            self.prefilling.extend(self._admit())
+            has_prefill = bool(self.prefilling)
         try:
-            if admitted:
+            if has_prefill:
                 self._flush_pending()  # running set is about to change
-                produced = self._prefill_step(admitted)
+                produced = self._prefill_step()
             else:
                 with self._lock:
                     idle = not self.running
@@ -318,43 +326,65 @@ class LLMEngine:
         except Exception as e:
             logger.exception("engine step failed")
             with self._lock:
-                for req in list(self.running) + admitted:
+                for req in list(self.running) + list(self.prefilling):
                     req.error = f"engine error: {e}"
                     self._finish(req, "error")
                 self.running.clear()
+                self.prefilling.clear()
             raise
         self.stats["steps"] += 1
         return produced
 
-    def _prefill_step(self, reqs: List[EngineRequest]) -> int:
+    def _prefill_step(self) -> int:
+        """One prefill forward over up to prefill_budget tokens of the
+        prefilling set. Each request's chunk attends to its previously
+        written context (prefix-cache hits and earlier chunks alike)
+        through the paged cache; a request whose chunk reaches the end of
+        its prompt samples its first token and joins the running set."""
         device = self.device
         bs = self.kv.block_size
-        # with prefix caching, only the un-cached suffix is prefilled; the
-        # cached context is read from the paged cache by the kernel
-        ncs = [getattr(req, "num_cached", 0) for req in reqs]
-        lens = [len(req.prompt_ids) - nc for req, nc in zip(reqs, ncs)]
+
+        with self._lock:
+            budget = self.prefill_budget
+            work: List[tuple] = []  # (req, start, end, final)
+            for req in self.prefilling:
+                if budget <= 0:
+                    break
+                L = len(req.prompt_ids)
+                chunk = min(L - req.prefill_pos, budget)
+                work.append((req, req.prefill_pos, req.prefill_pos + chunk,
+                             req.prefill_pos + chunk == L))
+                budget -= chunk
+        if not work:
+            return 0
+
+        reqs = [w[0] for w in work]
+        starts = [w[1] for w in work]
+        lens = [w[2] - w[1] for w in work]
         cu = np.zeros(len(reqs) + 1, dtype=np.int32)
         np.cumsum(lens, out=cu[1:])
         token_ids = np.concatenate(
-            [np.asarray(r.prompt_ids[nc:], dtype=np.int64) for r, nc in zip(reqs, ncs)]
+            [np.asarray(r.prompt_ids[s : s + L], dtype=np.int64)
+             for r, s, L in zip(reqs, starts, lens)]
         )
         positions = np.concatenate(
-            [np.arange(nc, nc + L, dtype=np.int64) for L, nc in zip(lens, ncs)]
+            [np.arange(s, s + L, dtype=np.int64) for s, L in zip(starts, lens)]
         )
         slots = np.empty(int(cu[-1]), dtype=np.int64)
-        for i, req in enumerate(reqs):
-            p = np.arange(ncs[i], ncs[i] + lens[i], dtype=np.int64)
+        for i, (req, s, L) in enumerate(zip(reqs, starts, lens)):
+            p = np.arange(s, s + L, dtype=np.int64)
             bt = np.asarray(req.block_table, dtype=np.int64)
             slots[cu[i] : cu[i + 1]] = bt[p // bs] * bs + p % bs
         logits_idx = (cu[1:] - 1).astype(np.int64)
-        max_len = max(lens)
 
         cached_lens_t = None
         block_tables_t = None
-        if any(ncs):
+        if any(starts):
             rows = np.fromiter((req.bt_slot for req in reqs), dtype=np.intp, count=len(reqs))
             block_tables_t = torch.from_numpy(self._bt_np[rows]).to(device)
-            cached_lens_t = torch.from_numpy(np.asarray(ncs, dtype=np.int32)).to(device)
+            cached_lens_t = torch.from_numpy(
+                np.asarray(starts, dtype=np.int32)
+            ).to(device)
 
         tile_seq, tile_off = ops.build_prefill_tiles(lens, device)
         batch = ForwardBatch(
@@ -363,7 +393,7 @@ class LLMEngine:
             positions=torch.from_numpy(positions).to(device),
             slot_mapping=torch.from_numpy(slots).to(device),
             cu_seqlens=torch.from_numpy(cu).to(device),
-            max_seqlen=max_len,
+            max_seqlen=max(lens),
             tile_seq=tile_seq,
             tile_off=tile_off,
             block_tables=block_tables_t,
@@ -371,17 +401,23 @@ class LLMEngine:
             logits_indices=torch.from_numpy(logits_idx).to(device),
         )
         logits = self.model.forward(batch, self.kv.k_caches, self.kv.v_caches)
-        tokens = self._sample(logits, reqs)
+        finals = [w[0] for w in work if w[3]]
+        tokens = self._sample(logits, reqs)  # non-final rows are discarded
+        final_tokens = [t for t, w in zip(tokens, work) if w[3]]
         self.stats["prefill_tokens"] += len(token_ids)
         if self.prefix_caching:
-            # KV for the suffix is now written (stream-ordered before any
-            # later forward): make the full prompt blocks reusable
-            for req in reqs:
+            # KV for this chunk is now written (stream-ordered before any
+            # later forward): make completed full prompt blocks reusable
+            for req in finals:
                 self.kv.manager.register_prefix(req.prompt_ids, req.block_table)
         with self._lock:
-            self.running.extend(reqs)
-            self._deliver(reqs, tokens)
-        return len(reqs)
+            for req, s, e, final in work:
+                req.prefill_pos = e
+                if final:
+                    self.prefilling.remove(req)
+            self.running.extend(finals)
+            self._deliver(finals, final_tokens)
+        return len(finals)
 
     def _decode_step(self) -> int:
         device = self.device
@@ -598,6 +634,8 @@ class LLMEngine:
             return
         if req in self.running:  # e.g. abort of a running request
             self.running.remove(req)
+        if req in self.prefilling:
+            self.prefilling.remove(req)
         req.state = "failed" if reason == "error" else "finished"
         req.finish_reason = reason
         req.finished_time = time.monotonic()

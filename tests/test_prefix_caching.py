@@ -113,3 +113,41 @@ def test_engine_prefix_cache_survives_free_and_reuses():
     # blocks all reclaimed or cached; no leak
     mgr = eng.kv.manager
     assert mgr.num_free_blocks + len(mgr._evictable) == 32
+
+
+# ---- chunked prefill (cached-context phase over the request's own chunks) ----
+
+def test_chunked_prefill_matches_single_shot():
+    long_prompt = list(range(5, 55))  # 50 tokens
+    single = make_engine(prefix=False)
+    ref = single.generate(long_prompt, SamplingParams(max_tokens=5, ignore_eos=True))
+
+    chunked = LLMEngine(
+        model="tiny-llama", device="cpu", dtype=torch.float32,
+        block_size=16, num_blocks=64, seed=3, prefill_budget=16,
+    )
+    r = chunked.generate(long_prompt, SamplingParams(max_tokens=5, ignore_eos=True))
+    assert r.out_ids == ref.out_ids
+    # 50 tokens at budget 16 -> 4 prefill chunks before the first token
+    assert r.state == "finished"
+
+
+def test_chunked_prefill_concurrent_long_and_short():
+    eng = LLMEngine(
+        model="tiny-llama", device="cpu", dtype=torch.float32,
+        block_size=16, num_blocks=64, seed=3, prefill_budget=24,
+    )
+    from llmapigateway_amd.engine import EngineRequest
+
+    long_req = EngineRequest(list(range(5, 53)), SamplingParams(max_tokens=4, ignore_eos=True))
+    short_req = EngineRequest(list(range(9, 17)), SamplingParams(max_tokens=4, ignore_eos=True))
+    eng.add_request(long_req)
+    eng.add_request(short_req)
+    while any(r.state in ("waiting", "running") for r in (long_req, short_req)):
+        eng.step()
+    assert long_req.state == short_req.state == "finished"
+    assert len(long_req.out_ids) == 4 and len(short_req.out_ids) == 4
+    # equivalence with solo runs
+    solo = make_engine(prefix=False)
+    assert solo.generate(list(range(5, 53)), SamplingParams(max_tokens=4, ignore_eos=True)).out_ids == long_req.out_ids
+    assert solo.generate(list(range(9, 17)), SamplingParams(max_tokens=4, ignore_eos=True)).out_ids == short_req.out_ids
