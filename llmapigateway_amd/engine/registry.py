@@ -458,9 +458,12 @@ class EngineRegistry:
                 row.update(
                     device=str(eng.device),
                     waiting=len(eng.waiting),
+                    prefilling=len(eng.prefilling),
                     running=len(eng.running),
                     kv_blocks_free=eng.kv.manager.num_free_blocks,
                     kv_blocks_total=eng.kv.num_blocks,
+                    prefix_cache_hits=eng.kv.manager.stats_prefix_hits,
+                    prefix_cached_tokens=eng.kv.manager.stats_prefix_tokens,
                 )
             else:  # TP worker group (state lives in the worker processes)
                 row.update(device=f"tp{eng.tp}", in_flight=len(eng._reqs))

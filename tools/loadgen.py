@@ -27,10 +27,15 @@ def pct(xs, p):
     return round(xs[min(len(xs) - 1, int(p / 100 * len(xs)))], 2)
 
 
-async def one_request(client, args, results):
+async def one_request(client, args, results, idx=0):
+    # --vary appends a unique tail so only the shared head can prefix-hit
+    # (a fully repeated prompt makes the prefix cache absorb ~all prefill)
+    content = "x " * args.prompt_words
+    if args.vary:
+        content += f"request number {idx} " * 4
     payload = {
         "model": args.model,
-        "messages": [{"role": "user", "content": "x " * args.prompt_words}],
+        "messages": [{"role": "user", "content": content}],
         "max_tokens": args.max_tokens,
         "stream": True,
         "ignore_eos": True,
@@ -76,6 +81,8 @@ async def main():
     ap.add_argument("--duration", type=float, default=30.0)
     ap.add_argument("--prompt-words", type=int, default=64)
     ap.add_argument("--max-tokens", type=int, default=64)
+    ap.add_argument("--vary", action="store_true",
+                    help="unique prompt tail per request (limits prefix-cache hits)")
     args = ap.parse_args()
 
     headers = {"Authorization": f"Bearer {args.api_key}"} if args.api_key else {}
@@ -91,7 +98,7 @@ async def main():
             delay = target - time.monotonic()
             if delay > 0:
                 await asyncio.sleep(delay)
-            tasks.append(asyncio.create_task(one_request(client, args, results)))
+            tasks.append(asyncio.create_task(one_request(client, args, results, i)))
         await asyncio.gather(*tasks)
         elapsed = time.monotonic() - t_start
 
