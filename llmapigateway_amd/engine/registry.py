@@ -246,6 +246,18 @@ class EngineRegistry:
         if "ignore_eos" in payload:
             params.ignore_eos = bool(payload["ignore_eos"])
 
+        try:
+            n_choices = int(payload.get("n") or 1)
+        except (TypeError, ValueError):
+            n_choices = 1
+        if n_choices > 16:
+            return None, "n > 16 is not supported"
+        if n_choices > 1:
+            return await self._make_request_multi(
+                provider_name, engine, prompt_ids, params, payload,
+                is_streaming, n_choices,
+            )
+
         loop = asyncio.get_running_loop()
         queue: asyncio.Queue = asyncio.Queue()
 
@@ -290,6 +302,164 @@ class EngineRegistry:
         return (
             self._stream_response(
                 req, engine, queue, kind, value, completion_id, created, model_name
+            ),
+            None,
+        )
+
+    async def _make_request_multi(
+        self, provider_name, engine, prompt_ids, params, payload,
+        is_streaming, n,
+    ):
+        """OpenAI `n` choices: n engine requests share the prompt (the
+        prefix cache dedups its KV). Greedy runs return identical choices
+        (matching upstream behavior); temperature>0 diverges per choice.
+        Stop-string holdback trimming is not applied to multi-choice
+        streams (the engine still terminates on the stop string)."""
+        import dataclasses
+
+        loop = asyncio.get_running_loop()
+        shared: asyncio.Queue = asyncio.Queue()
+
+        class _Tagged:
+            __slots__ = ("q", "i")
+
+            def __init__(self, q, i):
+                self.q, self.i = q, i
+
+            def put_nowait(self, item):
+                self.q.put_nowait((self.i, item))
+
+        use_batch = hasattr(engine, "batch_notifier")
+        if use_batch:
+            self._ensure_batch_notifier(engine, loop)
+        reqs: List[EngineRequest] = []
+        for i in range(n):
+            def mk_finish(idx):
+                def onf(req):
+                    loop.call_soon_threadsafe(shared.put_nowait, (idx, ("finish", req)))
+                return onf
+
+            p_i = dataclasses.replace(params, stop=list(params.stop))
+            if use_batch:
+                req = EngineRequest(prompt_ids, p_i, on_finish=mk_finish(i))
+                req._aq = _Tagged(shared, i)
+            else:
+                def mk_tok(idx):
+                    def ont(req, tok):
+                        loop.call_soon_threadsafe(shared.put_nowait, (idx, ("token", tok)))
+                    return ont
+
+                req = EngineRequest(
+                    prompt_ids, p_i, on_token=mk_tok(i), on_finish=mk_finish(i)
+                )
+            try:
+                engine.add_request(req)
+            except Exception as e:
+                for r in reqs:
+                    engine.abort_request(r)
+                return None, f"Engine '{provider_name}' rejected request: {e}"
+            reqs.append(req)
+
+        model_name = payload.get("model", "unknown")
+        completion_id = f"chatcmpl-{uuid.uuid4().hex[:24]}"
+        created = int(time.time())
+        tokenizer = self.tokenizer
+
+        try:
+            first = await asyncio.wait_for(shared.get(), timeout=300.0)
+        except asyncio.TimeoutError:
+            for r in reqs:
+                engine.abort_request(r)
+            return None, f"Engine '{provider_name}' timed out before first token"
+        if first[1][0] == "finish" and first[1][1].state == "failed":
+            for r in reqs:
+                engine.abort_request(r)
+            return None, first[1][1].error or "engine failure"
+
+        def agg_usage():
+            comp = sum(len(r.out_ids) for r in reqs)
+            return {
+                "prompt_tokens": len(prompt_ids),
+                "completion_tokens": comp,
+                "total_tokens": len(prompt_ids) + comp,
+            }
+
+        if not is_streaming:
+            remaining = n
+            ev = first
+            while True:
+                _, (kind, value) = ev
+                if kind == "finish":
+                    remaining -= 1
+                    if remaining == 0:
+                        break
+                ev = await shared.get()
+            choices = []
+            for i, req in enumerate(reqs):
+                text = tokenizer.decode(req.out_ids)
+                for s in req.params.stop:
+                    cut = text.find(s)
+                    if cut >= 0:
+                        text = text[:cut]
+                choices.append(
+                    {
+                        "index": i,
+                        "message": {"role": "assistant", "content": text},
+                        "finish_reason": req.finish_reason or "stop",
+                    }
+                )
+            return (
+                {
+                    "id": completion_id,
+                    "object": "chat.completion",
+                    "created": created,
+                    "model": model_name,
+                    "choices": choices,
+                    "usage": agg_usage(),
+                },
+                None,
+            )
+
+        def chunk(idx, delta, finish=None, usage=None) -> bytes:
+            obj: Dict[str, Any] = {
+                "id": completion_id,
+                "object": "chat.completion.chunk",
+                "created": created,
+                "model": model_name,
+                "choices": [{"index": idx, "delta": delta, "finish_reason": finish}],
+            }
+            if usage is not None:
+                obj["usage"] = usage
+            return b"data: " + json.dumps(obj, separators=(",", ":")).encode() + b"\n\n"
+
+        async def gen():
+            try:
+                for i in range(n):
+                    yield chunk(i, {"role": "assistant", "content": ""})
+                remaining = n
+                ev = first
+                while True:
+                    idx, (kind, value) = ev
+                    if kind == "token":
+                        yield chunk(idx, {"content": tokenizer.decode([value])})
+                    else:
+                        remaining -= 1
+                        yield chunk(
+                            idx, {}, finish=value.finish_reason or "stop",
+                            usage=agg_usage() if remaining == 0 else None,
+                        )
+                        if remaining == 0:
+                            yield b"data: [DONE]\n\n"
+                            return
+                    ev = await shared.get()
+            finally:
+                for r in reqs:
+                    if r.state in ("waiting", "running"):
+                        engine.abort_request(r)
+
+        return (
+            StreamingResponse(
+                gen(), media_type="text/event-stream", headers=dict(STREAM_HEADERS)
             ),
             None,
         )

@@ -202,3 +202,61 @@ def test_engine_overload_fails_fast():
     payload = {"model": "tiny-llama", "messages": [{"role": "user", "content": "hi"}]}
     resp, err = asyncio.run(reg.make_request("p", spec, payload, is_streaming=False))
     assert resp is None and "overloaded" in err
+
+
+def test_n_choices_nonstream():
+    import asyncio
+
+    from llmapigateway_amd.config.loader import EngineSpec
+    from llmapigateway_amd.engine.registry import EngineRegistry
+
+    reg = EngineRegistry()
+    spec = EngineSpec(model="tiny-llama", max_batch_size=8, kv_block_size=16)
+    payload = {
+        "model": "tiny-llama", "n": 3, "max_tokens": 5, "ignore_eos": True,
+        "messages": [{"role": "user", "content": "hello"}],
+    }
+    resp, err = asyncio.run(reg.make_request("p", spec, payload, is_streaming=False))
+    assert err is None
+    assert [c["index"] for c in resp["choices"]] == [0, 1, 2]
+    # greedy: all choices identical (matches upstream-provider behavior)
+    texts = [c["message"]["content"] for c in resp["choices"]]
+    assert texts[0] == texts[1] == texts[2] and texts[0]
+    assert resp["usage"]["completion_tokens"] == 15
+
+
+def test_n_choices_streaming():
+    import asyncio
+    import json as _json
+
+    from llmapigateway_amd.config.loader import EngineSpec
+    from llmapigateway_amd.engine.registry import EngineRegistry
+
+    reg = EngineRegistry()
+    spec = EngineSpec(model="tiny-llama", max_batch_size=8, kv_block_size=16)
+    payload = {
+        "model": "tiny-llama", "n": 2, "max_tokens": 4, "ignore_eos": True,
+        "messages": [{"role": "user", "content": "hello"}],
+    }
+
+    async def run():
+        resp, err = await reg.make_request("p", spec, payload, is_streaming=True)
+        assert err is None
+        frames = []
+        async for raw in resp.body_iterator:
+            frames.append(raw)
+        return frames
+
+    frames = asyncio.run(run())
+    assert frames[-1] == b"data: [DONE]\n\n"
+    seen = {0: "", 1: ""}
+    finishes = 0
+    for f in frames[:-1]:
+        obj = _json.loads(f[6:])
+        ch = obj["choices"][0]
+        if ch["delta"].get("content"):
+            seen[ch["index"]] += ch["delta"]["content"]
+        if ch["finish_reason"]:
+            finishes += 1
+    assert finishes == 2
+    assert seen[0] and seen[0] == seen[1]  # greedy: identical streams
