@@ -65,6 +65,29 @@ MODEL_PRESETS = {
         vocab_size=128256,
         rope_theta=500000.0,
     ),
+    # MHA family (GQA group = 1)
+    "llama-2-7b": ModelConfig(
+        name="llama-2-7b",
+        hidden_size=4096,
+        intermediate_size=11008,
+        num_layers=32,
+        num_heads=32,
+        num_kv_heads=32,
+        vocab_size=32000,
+        rope_theta=10000.0,
+        max_positions=4096,
+    ),
+    # odd GQA group (G = 7) exercises the templated decode kernel
+    "qwen2-7b": ModelConfig(
+        name="qwen2-7b",
+        hidden_size=3584,
+        intermediate_size=18944,
+        num_layers=28,
+        num_heads=28,
+        num_kv_heads=4,
+        vocab_size=152064,
+        rope_theta=1000000.0,
+    ),
     "mistral-7b": ModelConfig(
         name="mistral-7b",
         hidden_size=4096,

@@ -234,10 +234,14 @@ extern "C" hipError_t launch_attention_decode(
         (bf16*)out, (const bf16*)q, (const bf16*)k_cache,                      \
         (const bf16*)v_cache, block_tables, context_lens, scale, Hq, Hkv,      \
         block_size, max_blocks, q_stride)
-    switch (G) {
+    switch (G) {  // every GQA ratio up to MAX_G (e.g. qwen2-7b has G=7)
         case 1: LAUNCH_G(1); break;
         case 2: LAUNCH_G(2); break;
+        case 3: LAUNCH_G(3); break;
         case 4: LAUNCH_G(4); break;
+        case 5: LAUNCH_G(5); break;
+        case 6: LAUNCH_G(6); break;
+        case 7: LAUNCH_G(7); break;
         case 8: LAUNCH_G(8); break;
         default: return hipErrorInvalidValue;
     }

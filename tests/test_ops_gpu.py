@@ -146,6 +146,8 @@ def test_attention_prefill_strided_qkv():
         (1, 4, 4, 16, [1]),
         (4, 8, 2, 16, [5, 16, 33, 200]),
         (8, 32, 8, 64, [7, 64, 65, 100, 128, 250, 300, 512]),
+        (3, 4, 4, 16, [9, 70, 130]),    # MHA (G=1, llama-2 family)
+        (3, 28, 4, 16, [9, 70, 130]),   # odd GQA group (G=7, qwen2 family)
     ],
 )
 def test_attention_decode(B, Hq, Hkv, BS, lens):
