@@ -194,6 +194,24 @@ def attention_prefill(
     )
 
 
+_SPLIT_SPAN = 2048  # keep in sync with SPLIT_SPAN in attention_decode.hip
+_decode_ws: dict = {}
+_decode_ws_retired: list = []
+
+
+def _decode_scratch(device, n_acc: int, n_ml: int):
+    ws = _decode_ws.get(device.index)
+    if ws is None or ws[0].numel() < n_acc or ws[1].numel() < n_ml:
+        if ws is not None:
+            _decode_ws_retired.append(ws)  # captured graphs may still use it
+        ws = (
+            torch.empty(n_acc, dtype=torch.float32, device=device),
+            torch.empty(n_ml, dtype=torch.float32, device=device),
+        )
+        _decode_ws[device.index] = ws
+    return ws
+
+
 def attention_decode(
     q: torch.Tensor,
     k_cache: torch.Tensor,
@@ -206,9 +224,27 @@ def attention_decode(
         scale = float(q.shape[-1]) ** -0.5
     if q.is_cuda:
         out = torch.empty_like(q)
-        _native().attention_decode(
-            out, q, k_cache, v_cache, block_tables, context_lens, float(scale)
-        )
+        # flash-decode context splits when (B x Hkv) underfills the 256 CUs
+        # but the block table allows long contexts; derived only from
+        # capture-stable shapes so hipGraph replays stay valid
+        B, Hq = q.shape[0], q.shape[1]
+        max_ctx = block_tables.shape[1] * k_cache.shape[2]
+        nsplit = 1
+        if B * k_cache.shape[1] < 192 and max_ctx > _SPLIT_SPAN:
+            nsplit = min(8, -(-max_ctx // _SPLIT_SPAN))
+        if nsplit > 1:
+            pa, pm = _decode_scratch(
+                q.device, B * Hq * nsplit * q.shape[2], B * Hq * nsplit * 2
+            )
+            _native().attention_decode(
+                out, q, k_cache, v_cache, block_tables, context_lens,
+                float(scale), pa, pm, nsplit,
+            )
+        else:
+            _native().attention_decode(
+                out, q, k_cache, v_cache, block_tables, context_lens,
+                float(scale), None, None, 1,
+            )
         return out
     return reference.attention_decode(q, k_cache, v_cache, block_tables, context_lens, scale)
 

@@ -23,8 +23,15 @@
 #define MAX_G 8
 #define NWAVES 4
 #define DECODE_D 128
+#define SPLIT_SPAN 2048  // context per split (flash-decode partials)
 
-template <int G>
+// SPLIT=false: one workgroup per (seq, kv-head) writes out directly.
+// SPLIT=true: grid.z context-splits write un-normalized partials
+// (m, l, acc) to scratch; decode_combine_kernel merges them. Used when
+// B*Hkv underfills the chip but the context can be long (small-batch
+// long-context decode); the split count is derived from max_blocks so
+// it is hipGraph-capture-stable (idle splits exit on short contexts).
+template <int G, bool SPLIT>
 __launch_bounds__(NWAVES* WAVE_SIZE)
 __global__ void attention_decode_kernel(
     bf16* __restrict__ out,                 // [B, Hq, D]
@@ -38,12 +45,31 @@ __global__ void attention_decode_kernel(
     int Hkv,
     int block_size,
     int max_blocks,
-    int64_t q_stride) {
+    int64_t q_stride,
+    float* __restrict__ part_acc,           // [B, Hq, NSPLIT, D] (SPLIT)
+    float* __restrict__ part_ml,            // [B, Hq, NSPLIT, 2] (SPLIT)
+    int nsplit) {
     const int seq = blockIdx.x;
     const int kvh = blockIdx.y;
+    const int split = SPLIT ? blockIdx.z : 0;
     const int D = DECODE_D;
     const int L = context_lens[seq];
     if (L <= 0) return;
+    const int span0 = SPLIT ? split * SPLIT_SPAN : 0;
+    if (SPLIT && span0 >= L) {  // idle split: publish an empty partial
+        const int lane0 = threadIdx.x & (WAVE_SIZE - 1);
+        if (threadIdx.x < WAVE_SIZE) {
+#pragma unroll
+            for (int g = 0; g < G; ++g) {
+                const size_t pb =
+                    (((size_t)seq * Hq + kvh * G + g) * nsplit + split);
+                if (lane0 < 2) part_ml[pb * 2 + lane0] = lane0 == 0 ? -INFINITY : 0.f;
+                for (int d = lane0; d < D; d += WAVE_SIZE) part_acc[pb * D + d] = 0.f;
+            }
+        }
+        return;
+    }
+    const int span1 = SPLIT ? min(L, span0 + SPLIT_SPAN) : L;
 
     const int tid = threadIdx.x;
     const int lane = tid & (WAVE_SIZE - 1);
@@ -77,10 +103,11 @@ __global__ void attention_decode_kernel(
         for (int d = 0; d < 4; ++d) acc[g][d] = 0.f;
     }
 
-    const int nchunks = (L + WAVE_SIZE - 1) / WAVE_SIZE;
-    for (int c = wave; c < nchunks; c += NWAVES) {
+    const int c0 = span0 / WAVE_SIZE;  // SPLIT_SPAN is a multiple of 64
+    const int nchunks = (span1 + WAVE_SIZE - 1) / WAVE_SIZE;
+    for (int c = c0 + wave; c < nchunks; c += NWAVES) {
         const int pos = c * WAVE_SIZE + lane;
-        const bool valid = pos < L;
+        const bool valid = pos < span1;
 
         // --- phase A: lane = key ---
         float s[G];
@@ -145,7 +172,7 @@ __global__ void attention_decode_kernel(
             uint2 vp[4];
 #pragma unroll
             for (int u = 0; u < 4; ++u) {
-                const int pos = min(c * WAVE_SIZE + j + 2 * u + khalf, L - 1);
+                const int pos = min(c * WAVE_SIZE + j + 2 * u + khalf, span1 - 1);
                 const int vb = BT(pos / block_size);
                 const int vo = pos % block_size;
                 vp[u] = *reinterpret_cast<const uint2*>(
@@ -205,35 +232,78 @@ __global__ void attention_decode_kernel(
                 f[w] = (m_lds[w][g] == -INFINITY) ? 0.f : __expf(m_lds[w][g] - gm);
                 gl += f[w] * l_lds[w][g];
             }
-            const float inv_l = (gl > 0.f) ? 1.f / gl : 0.f;
             float o0 = 0.f, o1 = 0.f;
 #pragma unroll
             for (int w = 0; w < NWAVES; ++w) {
                 o0 += f[w] * acc_lds[w][g][2 * lane];
                 o1 += f[w] * acc_lds[w][g][2 * lane + 1];
             }
-            reinterpret_cast<uint32_t*>(
-                out + (size_t)seq * Hq * DECODE_D + (size_t)(kvh * G + g) * DECODE_D)[lane] =
-                pack2(o0 * inv_l, o1 * inv_l);
+            if (SPLIT) {
+                // un-normalized partial + (m, l) for the combine kernel
+                const size_t pb =
+                    (((size_t)seq * Hq + kvh * G + g) * nsplit + split);
+                part_acc[pb * DECODE_D + 2 * lane] = o0;
+                part_acc[pb * DECODE_D + 2 * lane + 1] = o1;
+                if (lane == 0) {
+                    part_ml[pb * 2] = gm;
+                    part_ml[pb * 2 + 1] = gl;
+                }
+            } else {
+                const float inv_l = (gl > 0.f) ? 1.f / gl : 0.f;
+                reinterpret_cast<uint32_t*>(
+                    out + (size_t)seq * Hq * DECODE_D +
+                    (size_t)(kvh * G + g) * DECODE_D)[lane] =
+                    pack2(o0 * inv_l, o1 * inv_l);
+            }
         }
     }
+}
+
+// merge the per-split partials: out[b,h] = sum_z exp(m_z - M) acc_z / L
+__global__ void decode_combine_kernel(
+    bf16* __restrict__ out,            // [B, Hq, D]
+    const float* __restrict__ part_acc,  // [B, Hq, NSPLIT, D]
+    const float* __restrict__ part_ml,   // [B, Hq, NSPLIT, 2]
+    int nsplit) {
+    const int bh = blockIdx.x;  // seq * Hq + head
+    const int lane = threadIdx.x;  // 64 lanes, 2 dims each
+    const size_t base = (size_t)bh * nsplit;
+    float gm = -INFINITY;
+    for (int z = 0; z < nsplit; ++z) gm = fmaxf(gm, part_ml[(base + z) * 2]);
+    float gl = 0.f, o0 = 0.f, o1 = 0.f;
+    for (int z = 0; z < nsplit; ++z) {
+        const float mz = part_ml[(base + z) * 2];
+        const float f = (mz == -INFINITY) ? 0.f : __expf(mz - gm);
+        gl += f * part_ml[(base + z) * 2 + 1];
+        o0 += f * part_acc[(base + z) * DECODE_D + 2 * lane];
+        o1 += f * part_acc[(base + z) * DECODE_D + 2 * lane + 1];
+    }
+    const float inv_l = (gl > 0.f) ? 1.f / gl : 0.f;
+    reinterpret_cast<uint32_t*>(out + (size_t)bh * DECODE_D)[lane] =
+        pack2(o0 * inv_l, o1 * inv_l);
 }
 
 extern "C" hipError_t launch_attention_decode(
     void* out, const void* q, const void* k_cache, const void* v_cache,
     const int* block_tables, const int* context_lens, float scale, int B,
     int Hq, int Hkv, int block_size, int max_blocks, int D, int64_t q_stride,
-    hipStream_t stream) {
+    float* part_acc, float* part_ml, int nsplit, hipStream_t stream) {
     if (D != DECODE_D) return hipErrorNotSupported;
     if (Hq % Hkv != 0) return hipErrorInvalidValue;
+    if (nsplit > 1 && (part_acc == nullptr || part_ml == nullptr))
+        return hipErrorInvalidValue;
     const int G = Hq / Hkv;
-    dim3 grid(B, Hkv);
+    dim3 grid(B, Hkv, nsplit > 1 ? nsplit : 1);
     dim3 block(NWAVES * WAVE_SIZE);
-#define LAUNCH_G(GV)                                                           \
-    attention_decode_kernel<GV><<<grid, block, 0, stream>>>(                   \
+#define LAUNCH_G2(GV, SPLIT)                                                   \
+    attention_decode_kernel<GV, SPLIT><<<grid, block, 0, stream>>>(            \
         (bf16*)out, (const bf16*)q, (const bf16*)k_cache,                      \
-        (const bf16*)v_cache, block_tables, context_lens, scale, Hq, Hkv,      \
-        block_size, max_blocks, q_stride)
+        (const bf16*)v_cache, block_tables, context_lens, scale, Hq, Hkv,     \
+        block_size, max_blocks, q_stride, part_acc, part_ml, nsplit)
+#define LAUNCH_G(GV)                                                           \
+    do {                                                                       \
+        if (nsplit > 1) LAUNCH_G2(GV, true); else LAUNCH_G2(GV, false);        \
+    } while (0)
     switch (G) {  // every GQA ratio up to MAX_G (e.g. qwen2-7b has G=7)
         case 1: LAUNCH_G(1); break;
         case 2: LAUNCH_G(2); break;
@@ -246,6 +316,11 @@ extern "C" hipError_t launch_attention_decode(
         default: return hipErrorInvalidValue;
     }
 #undef LAUNCH_G
+#undef LAUNCH_G2
+    if (nsplit > 1) {
+        decode_combine_kernel<<<dim3(B * Hq), dim3(WAVE_SIZE), 0, stream>>>(
+            (bf16*)out, part_acc, part_ml, nsplit);
+    }
     HIP_CHECK_LAST();
     return hipSuccess;
 }

@@ -18,7 +18,7 @@ hipError_t launch_rope(void*, void*, const int64_t*, const float*, int, int64_t,
 hipError_t launch_swiglu(void*, const void*, int, int, hipStream_t);
 hipError_t launch_kv_cache_write(const void*, const void*, void*, void*, const int64_t*, int, int64_t, int64_t, int, int, int, hipStream_t);
 hipError_t launch_rope_kv(void*, void*, const void*, void*, void*, const int64_t*, const float*, const int64_t*, int, int64_t, int64_t, int64_t, int, int, int, int, hipStream_t);
-hipError_t launch_attention_decode(void*, const void*, const void*, const void*, const int*, const int*, float, int, int, int, int, int, int, int64_t, hipStream_t);
+hipError_t launch_attention_decode(void*, const void*, const void*, const void*, const int*, const int*, float, int, int, int, int, int, int, int64_t, float*, float*, int, hipStream_t);
 hipError_t launch_attention_prefill(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, float, int, int, int, int64_t, int64_t, int64_t, const void*, const void*, const int*, const int*, int, int, hipStream_t);
 hipError_t launch_sample(int64_t*, const float*, const float*, const float*, float*, int*, int, int, hipStream_t);
 hipError_t launch_gemm_skinny(void*, float*, const void*, const void*, int, int, int, int, hipStream_t);
@@ -134,7 +134,9 @@ void rope_kv_write(torch::Tensor q, torch::Tensor k, torch::Tensor v,
 
 void attention_decode(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
                       torch::Tensor v_cache, torch::Tensor block_tables,
-                      torch::Tensor context_lens, double scale) {
+                      torch::Tensor context_lens, double scale,
+                      c10::optional<torch::Tensor> part_acc,
+                      c10::optional<torch::Tensor> part_ml, int64_t nsplit) {
     check_bf16(q, "q");
     check_bf16(out, "out");
     TORCH_CHECK(out.is_contiguous());
@@ -147,10 +149,22 @@ void attention_decode(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
     const int Hkv = k_cache.size(1);
     const int block_size = k_cache.size(2);
     const int max_blocks = block_tables.size(1);
+    float* pa = nullptr;
+    float* pm = nullptr;
+    if (nsplit > 1) {
+        TORCH_CHECK(part_acc.has_value() && part_ml.has_value());
+        TORCH_CHECK(part_acc->scalar_type() == torch::kFloat32 &&
+                    part_acc->numel() >= (int64_t)B * Hq * nsplit * D);
+        TORCH_CHECK(part_ml->scalar_type() == torch::kFloat32 &&
+                    part_ml->numel() >= (int64_t)B * Hq * nsplit * 2);
+        pa = part_acc->data_ptr<float>();
+        pm = part_ml->data_ptr<float>();
+    }
     CHECK_HIP(launch_attention_decode(
         out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
         block_tables.data_ptr<int>(), context_lens.data_ptr<int>(), (float)scale,
-        B, Hq, Hkv, block_size, max_blocks, D, q.stride(0), current_stream()));
+        B, Hq, Hkv, block_size, max_blocks, D, q.stride(0), pa, pm, (int)nsplit,
+        current_stream()));
 }
 
 void attention_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
