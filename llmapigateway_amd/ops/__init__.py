@@ -249,12 +249,13 @@ def attention_decode(
     return reference.attention_decode(q, k_cache, v_cache, block_tables, context_lens, scale)
 
 
-# The streaming kernel wins in the latency regime (tiny M: single-request
-# decode, where the library's tile/launch overhead dominates) and ties on
-# lm_head; at M>=64 hipBLASLt/rocBLAS reach ~60% of the per-CU fetch
-# ceiling and win — measured in profiles/gemm_skinny_probe (tools/gemm_probe.py).
-_SKINNY_MAX_M = 16
-_SKINNY_MAX_N = 28672  # lm_head-sized N re-reads X too often; library wins
+# The 8-wave MF=2 depth-4 pipeline matches the tuned library on every
+# decode shape and beats it at M=1 (1.14x) and gate_up M=256 (1.03x) —
+# profiles/r01_gemm_skinny_probe.md. Dispatch all decode-sized
+# projections through it; lm_head (N>28672) stays on the library (tie,
+# and its X re-read factor is ~2000x).
+_SKINNY_MAX_M = 256
+_SKINNY_MAX_N = 28672
 # per-device split-K fp32 slab scratch — each workgroup fully overwrites
 # its slab stripe, so no zeroing is needed and the address is stable
 # across hipGraph replays.

@@ -52,8 +52,10 @@ __device__ __forceinline__ bf16x8 gs_gload(const bf16* p) {
 // two k-steps of a row, and nt drops the line before the second 64 B hit,
 // doubling the W fetch traffic.
 
+// blockDim = nwaves*64 with nwaves = ceil(M / (16*MF)) (<= 8): MF=2 with 8
+// waves covers M=256 while the register-feasible depth-4 pipeline applies
 template <int MF, bool SPLITK>
-__launch_bounds__(GS_WAVES* WAVE_SIZE)
+__launch_bounds__(8 * WAVE_SIZE)
 __global__ void gemm_skinny_kernel(
     bf16* __restrict__ y,        // [M, N] (!SPLITK)
     float* __restrict__ yw,      // [nsk, M, N] fp32 slabs (SPLITK)
@@ -231,9 +233,13 @@ extern "C" hipError_t launch_gemm_skinny(
     if (nsk < 1) return hipErrorInvalidValue;
     if (nsk > 1 && workspace == nullptr) return hipErrorInvalidValue;
     const int tiles = N / GS_NT;
-    const int mf = ceil_div_i(M, 16 * GS_WAVES);
+    // smallest MF whose wave count fits 8 waves: deeper pipelines only fit
+    // registers at MF<=2, so prefer more waves over more rows per wave
+    int mf = 1;
+    while (ceil_div_i(M, 16 * mf) > 8) mf *= 2;
+    const int nwaves = ceil_div_i(M, 16 * mf);
     dim3 grid(tiles, nsk);
-    dim3 block(GS_WAVES * WAVE_SIZE);
+    dim3 block(nwaves * WAVE_SIZE);
 #define GS_LAUNCH(MFV, SPLIT)                                                  \
     gemm_skinny_kernel<MFV, SPLIT><<<grid, block, 0, stream>>>(                \
         (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K, nsk)
