@@ -249,12 +249,14 @@ def attention_decode(
     return reference.attention_decode(q, k_cache, v_cache, block_tables, context_lens, scale)
 
 
-# The 8-wave MF=2 depth-4 pipeline matches the tuned library on every
-# decode shape and beats it at M=1 (1.14x) and gate_up M=256 (1.03x) —
-# profiles/r01_gemm_skinny_probe.md. Dispatch all decode-sized
-# projections through it; lm_head (N>28672) stays on the library (tie,
-# and its X re-read factor is ~2000x).
-_SKINNY_MAX_M = 256
+# The 8-wave MF=2 depth-4 pipeline ties the tuned library when W is
+# L3-resident (tools/gemm_probe.py rotates too little data for >100 MB
+# weights, so its M>=64 rows are L3-warm and misled an M<=256 dispatch:
+# in-engine, where the 16 GB weight cycle is always L3-cold, batch-256
+# decode dropped 272 -> 129 req/s). In the true cold regime the library
+# keeps winning at M >= 64; the custom kernel stays dispatched for the
+# latency regime only.
+_SKINNY_MAX_M = 16
 _SKINNY_MAX_N = 28672
 # per-device split-K fp32 slab scratch — each workgroup fully overwrites
 # its slab stripe, so no zeroing is needed and the address is stable

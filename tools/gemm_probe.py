@@ -26,7 +26,12 @@ SHAPES = [
 
 
 def bench(fn, iters=50):
-    # rotate through 8 weight copies so W is cold in L2 (engine-realistic)
+    # rotate weight copies so W is cold in cache. CAVEAT: the 256 MB
+    # rotation cap leaves >100 MB weights resident in the 256 MB Infinity
+    # Cache (L3) — such rows measure the L3-warm regime, NOT the engine's
+    # L3-cold weight cycle. Cross-check any dispatch decision against an
+    # in-engine bench (this bit us once: an L3-warm "parity" result
+    # regressed batch-256 decode 2x in situ).
     torch.cuda.synchronize()
     t0 = time.monotonic()
     for i in range(iters):
