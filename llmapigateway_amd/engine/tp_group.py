@@ -195,7 +195,7 @@ class TPEngineClient:
     ):
         from ..models.configs import get_model_config
 
-        self.full_config = get_model_config(model)
+        self.full_config = get_model_config(model) if isinstance(model, str) else model
         self.max_model_len = max_model_len or self.full_config.max_positions
         self.tp = tp
         ctx = mp.get_context("spawn")
