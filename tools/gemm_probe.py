@@ -49,7 +49,7 @@ def main():
     print(f"{'M':>4} {'N':>7} {'K':>6} {'custom us':>10} {'library us':>11} {'speedup':>8}")
     for (M, N, K) in SHAPES:
         x = (torch.randn(M, K, device=dev) * 0.5).bfloat16()
-        ncopies = max(1, min(8, (256 << 20) // (N * K * 2)))
+        ncopies = max(1, min(16, (2048 << 20) // (N * K * 2)))  # rotate >2 GB: defeat the 256 MB L3
         ws = [(torch.randn(N, K, device=dev) * 0.02).bfloat16() for _ in range(ncopies)]
         # correctness spot check
         ref = x.float() @ ws[0].float().T
