@@ -256,7 +256,7 @@ def attention_decode(
 # decode dropped 272 -> 129 req/s). In the true cold regime the library
 # keeps winning at M >= 64; the custom kernel stays dispatched for the
 # latency regime only.
-_SKINNY_MAX_M = 16
+_SKINNY_MAX_M = int(os.environ.get("SKINNY_GEMM_MAX_M", "16"))
 _SKINNY_MAX_N = 28672
 # per-device split-K fp32 slab scratch — each workgroup fully overwrites
 # its slab stripe, so no zeroing is needed and the address is stable
@@ -265,10 +265,12 @@ _skinny_ws: dict = {}
 
 
 def _skinny_nsk(N: int, K: int) -> int:
-    """Split-K factor: fill ~2 workgroups per CU (512 total), keep >= 8
-    k-steps per slice. Mirrored by the launcher's validation only."""
+    """Split-K factor: target exactly 256 workgroups (one 8-wave WG fills
+    a CU; extra WGs only queue) — more splits just multiply the fp32 slab
+    traffic, which at the old 512-WG target added ~60% to the weight
+    stream and halved in-engine decode throughput."""
     tiles = N // 64
-    nsk = max(1, -(-512 // tiles))
+    nsk = max(1, -(-256 // tiles))
     nsk = min(nsk, max(1, (K // 32) // 8), 8)
     return nsk
 
