@@ -17,6 +17,7 @@ from __future__ import annotations
 
 import itertools
 import logging
+import os
 import threading
 import time
 from collections import deque
@@ -162,7 +163,11 @@ class LLMEngine:
         # hipGraph-captured decode (GPU only; TP group ops are capturable
         # with RCCL but kept off by default under TP until validated)
         if use_hipgraph is None:
-            use_hipgraph = self.device.type == "cuda" and tp_size == 1
+            use_hipgraph = (
+                self.device.type == "cuda"
+                and tp_size == 1
+                and not os.environ.get("LLMAPI_NO_HIPGRAPH")
+            )
         self.graph_runner = None
         if use_hipgraph and self.device.type == "cuda":
             from .graph_runner import DecodeGraphRunner
