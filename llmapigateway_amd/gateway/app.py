@@ -14,7 +14,7 @@ from contextlib import asynccontextmanager
 from pathlib import Path
 from typing import Optional
 
-from fastapi import FastAPI
+from fastapi import FastAPI, Response
 from fastapi.middleware.cors import CORSMiddleware
 from fastapi.responses import RedirectResponse
 from fastapi.staticfiles import StaticFiles
@@ -106,5 +106,32 @@ def create_app(
     @app.get("/health")
     async def health():
         return {"status": "ok"}
+
+    from .metrics import CONTENT_TYPE_LATEST, GatewayMetrics
+
+    app.state.metrics = GatewayMetrics()
+
+    @app.get("/metrics")
+    async def metrics():
+        m = app.state.metrics
+        registry = getattr(getattr(app.state, "dispatcher", None), "engine_registry", None)
+        m.refresh_engine_gauges(registry)
+        if registry is not None:
+            try:
+                gen = sum(r.get("decode_tokens", 0) for r in registry.stats())
+                pre = sum(r.get("prefill_tokens", 0) for r in registry.stats())
+                pc = sum(r.get("prefix_cached_tokens", 0) for r in registry.stats())
+                # counters must be monotonic: advance by the delta since last scrape
+                for counter, total in (
+                    (m.generated_tokens_total, gen),
+                    (m.prefill_tokens_total, pre),
+                    (m.prefix_cached_tokens_total, pc),
+                ):
+                    delta = total - counter._value.get()
+                    if delta > 0:
+                        counter.inc(delta)
+            except Exception:  # pragma: no cover
+                pass
+        return Response(content=m.render(), media_type=CONTENT_TYPE_LATEST)
 
     return app

@@ -26,6 +26,7 @@ import asyncio
 import copy
 import logging
 import os
+import time
 from typing import Any, Dict
 
 from fastapi import APIRouter, HTTPException, Request
@@ -93,6 +94,9 @@ async def chat_completions(request: Request):
         sequence = sequence[start:] + sequence[:start]
         logger.info("Rotation: starting at index %d for '%s'", start, requested_model)
 
+    metrics = getattr(request.app.state, "metrics", None)
+    t_request = time.monotonic()
+    had_fallback = False
     last_error = "No providers were attempted."
     for entry in sequence:
         provider_name = entry.provider
@@ -146,6 +150,14 @@ async def chat_completions(request: Request):
                     state = request.scope.setdefault("state", {})
                     state["served_provider"] = provider_name
                     state["served_model"] = provider_model
+                    if metrics is not None:
+                        metrics.requests_total.labels(
+                            status="fallback_success" if had_fallback else "success"
+                        ).inc()
+                        # for local engines make_request returns after the
+                        # first token; for proxies after stream priming —
+                        # either way this is time-to-first-byte
+                        metrics.ttft_seconds.observe(time.monotonic() - t_request)
                     logger.info(
                         "Success: model '%s' on provider '%s'%s",
                         provider_model,
@@ -156,6 +168,9 @@ async def chat_completions(request: Request):
                 last_error = (
                     f"Model {provider_model} failed with provider '{provider_name}': {error}"
                 )
+                had_fallback = True
+                if metrics is not None:
+                    metrics.fallback_attempts_total.labels(provider=provider_name).inc()
                 logger.warning(last_error)
 
             if retry_count > 0 and 0 < retry_delay < 120:
@@ -168,6 +183,8 @@ async def chat_completions(request: Request):
                 await asyncio.sleep(retry_delay)
             retry_count -= 1
 
+    if metrics is not None:
+        metrics.requests_total.labels(status="error").inc()
     logger.error("All providers failed for '%s'. Last error: %s", requested_model, last_error)
     raise HTTPException(
         status_code=503,

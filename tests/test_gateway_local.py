@@ -260,3 +260,32 @@ def test_n_choices_streaming():
             finishes += 1
     assert finishes == 2
     assert seen[0] and seen[0] == seen[1]  # greedy: identical streams
+
+
+def test_metrics_endpoint(tmp_repo):
+    import json as _json
+    from fastapi.testclient import TestClient
+
+    (tmp_repo / "providers.json").write_text(_json.dumps(
+        [{"local-tiny": {"baseUrl": "local://tiny-llama?device=0", "apikey": ""}}]
+    ))
+    (tmp_repo / "models_fallback_rules.json").write_text(_json.dumps(
+        [{"gateway_model_name": "llmgateway/tiny",
+          "fallback_models": [{"provider": "local-tiny", "model": "tiny-llama"}]}]
+    ))
+    from llmapigateway_amd.gateway.app import create_app
+
+    app = create_app()
+    with TestClient(app) as client:
+        r = client.post(
+            "/v1/chat/completions",
+            json={"model": "llmgateway/tiny", "max_tokens": 4, "ignore_eos": True,
+                  "messages": [{"role": "user", "content": "hi"}]},
+        )
+        assert r.status_code == 200
+        m = client.get("/metrics")
+        assert m.status_code == 200
+        text = m.text
+        assert 'gateway_chat_requests_total{status="success"} 1.0' in text
+        assert "engine_generated_tokens_total 3.0" in text  # 1 prefill-sampled + 3 decode
+        assert "gateway_ttft_seconds_bucket" in text
