@@ -42,6 +42,8 @@ class SamplingParams:
     temperature: float = 0.0
     top_p: float = 1.0
     top_k: int = 0
+    presence_penalty: float = 0.0   # flat penalty on already-generated tokens
+    frequency_penalty: float = 0.0  # per-occurrence penalty (output tokens)
     max_tokens: int = 128
     seed: Optional[int] = None
     stop: List[str] = field(default_factory=list)
@@ -56,6 +58,8 @@ class SamplingParams:
             temperature=float(payload.get("temperature", 0.0) or 0.0),
             top_p=float(payload.get("top_p", 1.0) or 1.0),
             top_k=int(payload.get("top_k", 0) or 0),
+            presence_penalty=float(payload.get("presence_penalty", 0.0) or 0.0),
+            frequency_penalty=float(payload.get("frequency_penalty", 0.0) or 0.0),
             max_tokens=int(
                 payload.get("max_completion_tokens")
                 or payload.get("max_tokens")
@@ -539,8 +543,35 @@ class LLMEngine:
         return n
 
     # ---- sampling ----
+    def _apply_penalties(self, logits: torch.Tensor, reqs: List[EngineRequest]) -> torch.Tensor:
+        """OpenAI presence/frequency penalties over the OUTPUT tokens so
+        far (with deferred sampling the in-flight token lags one step —
+        an accepted approximation). Returns logits, cloned only if any
+        request uses penalties."""
+        rows, idxs, vals = [], [], []
+        from collections import Counter
+
+        for i, r in enumerate(reqs):
+            pp, fp = r.params.presence_penalty, r.params.frequency_penalty
+            if (pp == 0.0 and fp == 0.0) or not r.out_ids:
+                continue
+            for tok, c in Counter(r.out_ids).items():
+                rows.append(i)
+                idxs.append(tok)
+                vals.append(pp + fp * c)
+        if not rows:
+            return logits
+        logits = logits.clone()
+        dev = logits.device
+        logits[
+            torch.tensor(rows, dtype=torch.long, device=dev),
+            torch.tensor(idxs, dtype=torch.long, device=dev),
+        ] -= torch.tensor(vals, dtype=logits.dtype, device=dev)
+        return logits
+
     def _sample_dev(self, logits: torch.Tensor, reqs: List[EngineRequest]) -> torch.Tensor:
         """Sample next tokens; returns an int64 device tensor (no host sync)."""
+        logits = self._apply_penalties(logits, reqs)
         any_temp = any(r.params.temperature > 0 for r in reqs)
         temps = torch.tensor(
             [r.params.temperature for r in reqs], dtype=torch.float32, device=logits.device

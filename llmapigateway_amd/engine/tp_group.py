@@ -339,6 +339,8 @@ def _params_dict(params) -> dict:
         "temperature": params.temperature,
         "top_p": params.top_p,
         "top_k": params.top_k,
+        "presence_penalty": params.presence_penalty,
+        "frequency_penalty": params.frequency_penalty,
         "max_tokens": params.max_tokens,
         "seed": params.seed,
         "stop": list(params.stop),

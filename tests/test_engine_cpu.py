@@ -141,3 +141,28 @@ def test_on_token_callbacks():
     while req.state in ("waiting", "running"):
         eng.step()
     assert seen[:-1] == req.out_ids and seen[-1] == "done"
+
+
+def test_presence_penalty_blocks_repeats():
+    """A huge presence penalty makes greedy decoding avoid every token it
+    has already generated (OpenAI presence_penalty semantics over output
+    tokens)."""
+    import torch
+
+    from llmapigateway_amd.engine import LLMEngine, SamplingParams
+
+    eng = LLMEngine(
+        model="tiny-llama", device="cpu", dtype=torch.float32,
+        block_size=16, num_blocks=64, seed=0,
+    )
+    base = eng.generate(
+        list(range(5, 25)), SamplingParams(max_tokens=12, ignore_eos=True)
+    )
+    pen = eng.generate(
+        list(range(5, 25)),
+        SamplingParams(max_tokens=12, ignore_eos=True, presence_penalty=1e6),
+    )
+    assert len(pen.out_ids) == 12
+    assert len(set(pen.out_ids)) == 12, "penalized run must not repeat tokens"
+    # sanity: the unpenalized greedy run is allowed to repeat
+    assert base.state == "finished"
