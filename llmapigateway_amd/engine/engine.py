@@ -44,6 +44,7 @@ class SamplingParams:
     top_k: int = 0
     presence_penalty: float = 0.0   # flat penalty on already-generated tokens
     frequency_penalty: float = 0.0  # per-occurrence penalty (output tokens)
+    logit_bias: Dict[int, float] = field(default_factory=dict)
     max_tokens: int = 128
     seed: Optional[int] = None
     stop: List[str] = field(default_factory=list)
@@ -60,6 +61,10 @@ class SamplingParams:
             top_k=int(payload.get("top_k", 0) or 0),
             presence_penalty=float(payload.get("presence_penalty", 0.0) or 0.0),
             frequency_penalty=float(payload.get("frequency_penalty", 0.0) or 0.0),
+            logit_bias={
+                int(k): float(v)
+                for k, v in (payload.get("logit_bias") or {}).items()
+            },
             max_tokens=int(
                 payload.get("max_completion_tokens")
                 or payload.get("max_tokens")
@@ -553,12 +558,15 @@ class LLMEngine:
 
         for i, r in enumerate(reqs):
             pp, fp = r.params.presence_penalty, r.params.frequency_penalty
-            if (pp == 0.0 and fp == 0.0) or not r.out_ids:
-                continue
-            for tok, c in Counter(r.out_ids).items():
+            if (pp != 0.0 or fp != 0.0) and r.out_ids:
+                for tok, c in Counter(r.out_ids).items():
+                    rows.append(i)
+                    idxs.append(tok)
+                    vals.append(pp + fp * c)
+            for tok, bias in r.params.logit_bias.items():
                 rows.append(i)
                 idxs.append(tok)
-                vals.append(pp + fp * c)
+                vals.append(-bias)  # subtracted below: negate to ADD the bias
         if not rows:
             return logits
         logits = logits.clone()
@@ -582,6 +590,17 @@ class LLMEngine:
             noise = torch.rand(
                 logits.shape, generator=self._gen, device=logits.device, dtype=torch.float32
             )
+            # per-request seeds (OpenAI `seed`): deterministic per output
+            # position, independent of batch composition
+            for i, r in enumerate(reqs):
+                if r.params.seed is not None and r.params.temperature > 0:
+                    g = torch.Generator(device=logits.device).manual_seed(
+                        (int(r.params.seed) << 20) ^ len(r.out_ids)
+                    )
+                    noise[i] = torch.rand(
+                        logits.shape[-1], generator=g, device=logits.device,
+                        dtype=torch.float32,
+                    )
         else:
             noise = None
         return ops.sample(filtered, temps, noise)

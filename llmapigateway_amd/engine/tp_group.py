@@ -341,6 +341,7 @@ def _params_dict(params) -> dict:
         "top_k": params.top_k,
         "presence_penalty": params.presence_penalty,
         "frequency_penalty": params.frequency_penalty,
+        "logit_bias": dict(params.logit_bias),
         "max_tokens": params.max_tokens,
         "seed": params.seed,
         "stop": list(params.stop),

@@ -166,3 +166,48 @@ def test_presence_penalty_blocks_repeats():
     assert len(set(pen.out_ids)) == 12, "penalized run must not repeat tokens"
     # sanity: the unpenalized greedy run is allowed to repeat
     assert base.state == "finished"
+
+
+def test_logit_bias_forces_token():
+    import torch
+
+    from llmapigateway_amd.engine import LLMEngine, SamplingParams
+
+    eng = LLMEngine(
+        model="tiny-llama", device="cpu", dtype=torch.float32,
+        block_size=16, num_blocks=64, seed=0,
+    )
+    r = eng.generate(
+        list(range(5, 25)),
+        SamplingParams(max_tokens=5, ignore_eos=True, logit_bias={123: 1e9}),
+    )
+    assert r.out_ids == [123] * 5
+
+
+def test_seeded_sampling_reproducible_across_batches():
+    """Same request + seed -> same tokens regardless of batch composition."""
+    import torch
+
+    from llmapigateway_amd.engine import EngineRequest, LLMEngine, SamplingParams
+
+    def run(extra):
+        eng = LLMEngine(
+            model="tiny-llama", device="cpu", dtype=torch.float32,
+            block_size=16, num_blocks=64, seed=0,
+        )
+        reqs = [EngineRequest(
+            list(range(5, 25)),
+            SamplingParams(max_tokens=6, ignore_eos=True, temperature=0.9, seed=77),
+        )]
+        for _ in range(extra):  # batch-mates change the shared noise draw
+            reqs.append(EngineRequest(
+                list(range(9, 29)),
+                SamplingParams(max_tokens=6, ignore_eos=True, temperature=0.9),
+            ))
+        for q in reqs:
+            eng.add_request(q)
+        while any(q.state in ("waiting", "running") for q in reqs):
+            eng.step()
+        return reqs[0].out_ids
+
+    assert run(0) == run(3)
