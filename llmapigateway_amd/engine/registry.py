@@ -162,6 +162,33 @@ class EngineRegistry:
             logger.info("created engine %s on %s", key, device)
             return handle
 
+    def prune(self, active_specs: List[EngineSpec]) -> None:
+        """Stop and release engines no longer referenced by any local
+        provider entry (config reload): a removed 70B engine would
+        otherwise pin its weights in HBM indefinitely."""
+        keep = {self._engine_key(spec) for spec in active_specs}
+        with self._create_lock:
+            dead = [k for k in self._engines if k not in keep]
+            for key in dead:
+                handle = self._engines.pop(key)
+                logger.info("pruning engine %s (no longer configured)", key)
+                eng = handle.engine
+                for req in (
+                    list(getattr(eng, "waiting", []))
+                    + list(getattr(eng, "prefilling", []))
+                    + list(getattr(eng, "running", []))
+                ):
+                    try:  # fail in-flight work loudly instead of stranding it
+                        eng.abort_request(req)
+                    except Exception:
+                        pass
+                try:
+                    handle.stop()
+                except Exception:
+                    logger.exception("failed to stop engine %s", key)
+        if dead and torch.cuda.is_available():
+            torch.cuda.empty_cache()
+
     def _failure_state(self, provider: str, spec: EngineSpec) -> _FailureState:
         st = self._failures.get(provider)
         if st is None:

@@ -111,4 +111,12 @@ async def save_providers(request: Request):
         raise HTTPException(
             status_code=500, detail=f"Saved but reload failed: {'; '.join(reload_errs)}"
         )
+    # release engines whose provider entry no longer references them — a
+    # removed/changed local provider would otherwise keep its GPU-resident
+    # weights (up to 141 GB) alive forever
+    registry = getattr(request.app.state.dispatcher, "engine_registry", None)
+    if registry is not None:
+        registry.prune(
+            [p.engine_spec for p in config_loader.providers_config.values() if p.is_local]
+        )
     return {"status": "ok", "providers": len(config_loader.providers_config)}

@@ -289,3 +289,23 @@ def test_metrics_endpoint(tmp_repo):
         assert 'gateway_chat_requests_total{status="success"} 1.0' in text
         assert "engine_generated_tokens_total 3.0" in text  # 1 prefill-sampled + 3 decode
         assert "gateway_ttft_seconds_bucket" in text
+
+
+def test_registry_prune_releases_removed_engines():
+    import asyncio
+
+    from llmapigateway_amd.config.loader import EngineSpec
+    from llmapigateway_amd.engine.registry import EngineRegistry
+
+    reg = EngineRegistry()
+    spec_a = EngineSpec(model="tiny-llama", device=0, max_batch_size=4, kv_block_size=16)
+    spec_b = EngineSpec(model="tiny-llama", device=1, max_batch_size=4, kv_block_size=16)
+    # CPU: both resolve to cpu engines but distinct keys (device in key)
+    ha = reg.get_engine(spec_a)
+    hb = reg.get_engine(spec_b)
+    assert len(reg._engines) == 2
+    reg.prune([spec_a])
+    assert len(reg._engines) == 1
+    assert reg.get_engine(spec_a) is ha
+    assert not hb.thread.is_alive()
+    asyncio.run(reg.aclose())
