@@ -100,7 +100,16 @@ class EngineRegistry:
 
     # ---- engine lifecycle ----
     def _engine_key(self, spec: EngineSpec) -> str:
-        return f"{spec.model}@{spec.device}x{spec.tp}"
+        # identical specs share one engine (replica provider entries on one
+        # GPU); explicit batch/block/dtype overrides make the spec distinct
+        key = f"{spec.model}@{spec.device}x{spec.tp}"
+        if spec.max_batch_size:
+            key += f"/b{spec.max_batch_size}"
+        if spec.kv_block_size:
+            key += f"/k{spec.kv_block_size}"
+        if spec.dtype not in ("bfloat16", None):
+            key += f"/{spec.dtype}"
+        return key
 
     def _resolve_device(self, spec: EngineSpec) -> str:
         if torch.cuda.is_available():
