@@ -309,3 +309,52 @@ def test_registry_prune_releases_removed_engines():
     assert reg.get_engine(spec_a) is ha
     assert not hb.thread.is_alive()
     asyncio.run(reg.aclose())
+
+
+def test_providers_reload_prunes_removed_engine(tmp_repo):
+    """Saving a providers.json that drops a local provider releases its
+    engine (HTTP round-trip through the rules editor)."""
+    import json as _json
+    from fastapi.testclient import TestClient
+
+    (tmp_repo / "providers.json").write_text(_json.dumps(
+        [{"local-tiny": {"baseUrl": "local://tiny-llama?device=0", "apikey": ""}}]
+    ))
+    (tmp_repo / "models_fallback_rules.json").write_text(_json.dumps(
+        [{"gateway_model_name": "llmgateway/tiny",
+          "fallback_models": [{"provider": "local-tiny", "model": "tiny-llama"}]}]
+    ))
+    from llmapigateway_amd.gateway.app import create_app
+
+    app = create_app()
+    with TestClient(app) as client:
+        r = client.post(
+            "/v1/chat/completions",
+            json={"model": "llmgateway/tiny", "max_tokens": 2, "ignore_eos": True,
+                  "messages": [{"role": "user", "content": "hi"}]},
+        )
+        assert r.status_code == 200
+        reg = app.state.dispatcher.engine_registry
+        assert len(reg._engines) == 1
+        handle = next(iter(reg._engines.values()))
+        # dropping the provider while rules still reference it is rejected
+        only_other = _json.dumps(
+            [{"local-other": {"baseUrl": "local://tiny-llama?device=1", "apikey": ""}}]
+        )
+        resp = client.post("/v1/config/providers", content=only_other.encode())
+        assert resp.status_code == 400
+        assert len(reg._engines) == 1  # nothing pruned on a rejected save
+        # proper migration: add the new provider, repoint the rules, drop the old
+        both = _json.dumps([
+            {"local-tiny": {"baseUrl": "local://tiny-llama?device=0", "apikey": ""}},
+            {"local-other": {"baseUrl": "local://tiny-llama?device=1", "apikey": ""}},
+        ])
+        assert client.post("/v1/config/providers", content=both.encode()).status_code == 200
+        new_rules = _json.dumps(
+            [{"gateway_model_name": "llmgateway/tiny",
+              "fallback_models": [{"provider": "local-other", "model": "tiny-llama"}]}]
+        )
+        assert client.post("/v1/config/models-rules", content=new_rules.encode()).status_code == 200
+        assert client.post("/v1/config/providers", content=only_other.encode()).status_code == 200
+        assert len(reg._engines) == 0  # old engine pruned
+        assert not handle.thread.is_alive()
