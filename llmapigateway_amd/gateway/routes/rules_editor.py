@@ -117,6 +117,6 @@ async def save_providers(request: Request):
     registry = getattr(request.app.state.dispatcher, "engine_registry", None)
     if registry is not None:
         registry.prune(
-            [p.engine_spec for p in config_loader.providers_config.values() if p.is_local]
+            [p.engine_spec() for p in config_loader.providers_config.values() if p.is_local]
         )
     return {"status": "ok", "providers": len(config_loader.providers_config)}
