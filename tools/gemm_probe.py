@@ -12,6 +12,11 @@ import torch.nn.functional as F
 
 from llmapigateway_amd import ops
 
+# force the custom kernel for the "custom" column regardless of the
+# production dispatch threshold (without this, M > _SKINNY_MAX_M silently
+# measures the library against itself — which fooled us once)
+ops._SKINNY_MAX_M = 1_000_000
+
 SHAPES = [
     (256, 6144, 4096),    # qkv
     (256, 4096, 4096),    # o
