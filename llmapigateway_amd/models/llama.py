@@ -132,6 +132,21 @@ class LlamaModel:
         self.final_norm = torch.ones(hidden, dtype=self.dtype, device=self.device)
         self.lm_head = self.embed if c.tie_embeddings else mk(self.full_config.vocab_size, hidden, std)
 
+        # k-major swizzled twins of the decode projections for the
+        # weight-streaming skinny GEMM (contiguous 4 KB B-tile streams;
+        # ops/csrc/gemm_skinny.hip). Measured: 30-40% faster than the
+        # plain-layout custom kernel but still behind the tuned library
+        # at M>=64 cold (profiles/r01_gemm_skinny_probe.md), so the extra
+        # weight copy is opt-in.
+        import os as _os
+
+        if self.device.type == "cuda" and _os.environ.get("LLMAPI_SWZ_WEIGHTS"):
+            for layer in self.layers:
+                for name in ("qkv", "o", "gate_up", "down"):
+                    w = layer[name]
+                    if w.shape[0] % 64 == 0 and w.shape[1] % 32 == 0:
+                        layer[name + "_swz"] = ops.swizzle_weight(w)
+
     def param_bytes(self) -> int:
         total = self.embed.numel() + self.final_norm.numel()
         if self.lm_head is not self.embed:
@@ -165,7 +180,7 @@ class LlamaModel:
             else:
                 x, residual = ops.rmsnorm_residual(h, residual, layer["input_norm"], c.rms_eps)
 
-            qkv = ops.linear(x, layer["qkv"])
+            qkv = ops.linear(x, layer["qkv"], layer.get("qkv_swz"))
             q, k, v = qkv.split([c.q_size, c.kv_size, c.kv_size], dim=-1)
             q = q.view(T, c.num_heads, c.head_dim)
             k = k.view(T, c.num_kv_heads, c.head_dim)
@@ -189,11 +204,16 @@ class LlamaModel:
                     q, k_caches[i], v_caches[i], batch.block_tables, batch.context_lens, self.scale
                 )
 
-            h = self._maybe_all_reduce(ops.linear(attn.reshape(T, c.q_size), layer["o"]))
+            h = self._maybe_all_reduce(
+                ops.linear(attn.reshape(T, c.q_size), layer["o"], layer.get("o_swz"))
+            )
 
             x, residual = ops.rmsnorm_residual(h, residual, layer["post_norm"], c.rms_eps)
             h = self._maybe_all_reduce(
-                ops.linear(ops.swiglu(ops.linear(x, layer["gate_up"])), layer["down"])
+                ops.linear(
+                    ops.swiglu(ops.linear(x, layer["gate_up"], layer.get("gate_up_swz"))),
+                    layer["down"], layer.get("down_swz"),
+                )
             )
 
         x, _ = ops.rmsnorm_residual(h, residual, self.final_norm, c.rms_eps)

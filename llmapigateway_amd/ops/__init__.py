@@ -288,16 +288,33 @@ def _skinny_scratch(device, numel: int):
     return ws
 
 
-def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+def swizzle_weight(w: torch.Tensor) -> torch.Tensor:
+    """[N, K] -> k-major [K/32, N, 32] for gemm_skinny's contiguous
+    B-tile streams (one 4 KB block per 64-row n-stripe per k-step)."""
+    N, K = w.shape
+    assert K % 32 == 0 and N % 64 == 0
+    return w.view(N, K // 32, 32).permute(1, 0, 2).contiguous()
+
+
+# swizzled weights raise the profitable dispatch ceiling (contiguous
+# streams); plain-layout dispatch stays in the latency regime
+_SKINNY_SWZ_MAX_M = int(os.environ.get("SKINNY_GEMM_SWZ_MAX_M", "256"))
+
+
+def linear(
+    x: torch.Tensor, w: torch.Tensor, w_swz: Optional[torch.Tensor] = None
+) -> torch.Tensor:
     """y = x @ w.T. Decode-shaped bf16 GEMMs (M <= 256) go through the
-    weight-streaming gfx950 kernel (csrc/gemm_skinny.hip); everything else
-    through the TunableOp-tuned library GEMMs."""
+    weight-streaming gfx950 kernel (csrc/gemm_skinny.hip) — with a
+    pre-swizzled weight (models/llama.py builds them) its B-stream is
+    fully contiguous; everything else through the TunableOp-tuned
+    library GEMMs."""
     M = x.shape[0]
     if (
         x.is_cuda
         and x.dtype == torch.bfloat16
         and w.dtype == torch.bfloat16
-        and 0 < M <= _SKINNY_MAX_M
+        and 0 < M
         and w.shape[0] % 64 == 0
         and w.shape[0] <= _SKINNY_MAX_N
         and w.shape[1] % 32 == 0
@@ -305,11 +322,18 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
         and w.is_contiguous()
     ):
         N, K = w.shape
-        y = torch.empty((M, N), dtype=torch.bfloat16, device=x.device)
-        nsk = _skinny_nsk(N, K)
-        ws = _skinny_scratch(x.device, nsk * M * N) if nsk > 1 else None
-        _native().gemm_skinny(y, x, w, ws, nsk)
-        return y
+        if w_swz is not None and M <= _SKINNY_SWZ_MAX_M:
+            y = torch.empty((M, N), dtype=torch.bfloat16, device=x.device)
+            nsk = _skinny_nsk(N, K)
+            ws = _skinny_scratch(x.device, nsk * M * N) if nsk > 1 else None
+            _native().gemm_skinny(y, x, w_swz, ws, nsk, True)
+            return y
+        if M <= _SKINNY_MAX_M:
+            y = torch.empty((M, N), dtype=torch.bfloat16, device=x.device)
+            nsk = _skinny_nsk(N, K)
+            ws = _skinny_scratch(x.device, nsk * M * N) if nsk > 1 else None
+            _native().gemm_skinny(y, x, w, ws, nsk, False)
+            return y
     return torch.nn.functional.linear(x, w)
 
 

@@ -21,7 +21,7 @@ hipError_t launch_rope_kv(void*, void*, const void*, void*, void*, const int64_t
 hipError_t launch_attention_decode(void*, const void*, const void*, const void*, const int*, const int*, float, int, int, int, int, int, int, int64_t, float*, float*, int, hipStream_t);
 hipError_t launch_attention_prefill(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, float, int, int, int, int64_t, int64_t, int64_t, const void*, const void*, const int*, const int*, int, int, hipStream_t);
 hipError_t launch_sample(int64_t*, const float*, const float*, const float*, float*, int*, int, int, hipStream_t);
-hipError_t launch_gemm_skinny(void*, float*, const void*, const void*, int, int, int, int, hipStream_t);
+hipError_t launch_gemm_skinny(void*, float*, const void*, const void*, int, int, int, int, int, hipStream_t);
 }
 
 namespace {
@@ -234,14 +234,23 @@ void sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temperature,
 }
 
 void gemm_skinny(torch::Tensor y, torch::Tensor x, torch::Tensor w,
-                 c10::optional<torch::Tensor> workspace, int64_t nsk) {
+                 c10::optional<torch::Tensor> workspace, int64_t nsk,
+                 bool swizzled) {
     check_bf16(x, "x");
     check_bf16(w, "w");
     check_bf16(y, "y");
     TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && y.is_contiguous());
-    TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && y.dim() == 2);
-    const int M = x.size(0), K = x.size(1), N = w.size(0);
-    TORCH_CHECK(w.size(1) == K && y.size(0) == M && y.size(1) == N);
+    TORCH_CHECK(x.dim() == 2 && y.dim() == 2);
+    const int M = x.size(0), K = x.size(1);
+    int N;
+    if (swizzled) {  // w is [K/32, N, 32]
+        TORCH_CHECK(w.dim() == 3 && w.size(0) == K / 32 && w.size(2) == 32);
+        N = w.size(1);
+    } else {
+        TORCH_CHECK(w.dim() == 2 && w.size(1) == K);
+        N = w.size(0);
+    }
+    TORCH_CHECK(y.size(0) == M && y.size(1) == N);
     float* ws = nullptr;
     if (workspace.has_value() && workspace->defined()) {
         TORCH_CHECK(workspace->scalar_type() == torch::kFloat32 &&
@@ -250,7 +259,8 @@ void gemm_skinny(torch::Tensor y, torch::Tensor x, torch::Tensor w,
         ws = workspace->data_ptr<float>();
     }
     CHECK_HIP(launch_gemm_skinny(y.data_ptr(), ws, x.data_ptr(), w.data_ptr(),
-                                 M, N, K, (int)nsk, current_stream()));
+                                 M, N, K, (int)nsk, swizzled ? 1 : 0,
+                                 current_stream()));
 }
 
 }  // namespace

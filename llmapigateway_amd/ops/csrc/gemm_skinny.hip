@@ -52,15 +52,24 @@ __device__ __forceinline__ bf16x8 gs_gload(const bf16* p) {
 // two k-steps of a row, and nt drops the line before the second 64 B hit,
 // doubling the W fetch traffic.
 
+// scheduling fence: keep the software-pipeline shape — without it the
+// scheduler compresses the depth-4 pipeline back to serial loads (84
+// VGPRs observed) and the kernel runs latency-bound
+#define GS_FENCE __builtin_amdgcn_sched_barrier(0)
+
 // blockDim = nwaves*64 with nwaves = ceil(M / (16*MF)) (<= 8): MF=2 with 8
-// waves covers M=256 while the register-feasible depth-4 pipeline applies
-template <int MF, bool SPLITK>
+// waves covers M=256 while the register-feasible depth-4 pipeline applies.
+// SWZ: W is pre-swizzled k-major ([K/32, N, 32]) so every B-fragment load
+// is part of a contiguous 4 KB (64-row x 64 B) stream per k-step —
+// the [N, K] row-major layout reads 16 rows x 64 B per instruction and
+// caps at ~3.5 TB/s effective.
+template <int MF, bool SPLITK, bool SWZ>
 __launch_bounds__(8 * WAVE_SIZE)
 __global__ void gemm_skinny_kernel(
     bf16* __restrict__ y,        // [M, N] (!SPLITK)
     float* __restrict__ yw,      // [nsk, M, N] fp32 slabs (SPLITK)
     const bf16* __restrict__ x,  // [M, K]
-    const bf16* __restrict__ w,  // [N, K]
+    const bf16* __restrict__ w,  // [N, K] or swizzled [K/32, N, 32]
     int M,
     int N,
     int K,
@@ -89,10 +98,16 @@ __global__ void gemm_skinny_kernel(
         const int r = m_base + f * 16 + lrow;
         ap[f] = x + (size_t)(r < M ? r : 0) * K + k0 + lk * 8;
     }
+    // B pointers: row-major [N, K] strides K per n-row and 32 per k-step;
+    // swizzled [K/32, N, 32] strides 32 per n-row and 32*N per k-step
+    const size_t bstep = SWZ ? (size_t)32 * N : 32;  // elems per k-step
     const bf16* bp[GS_NF];
 #pragma unroll
     for (int n = 0; n < GS_NF; ++n)
-        bp[n] = w + (size_t)(n0 + n * 16 + lrow) * K + k0 + lk * 8;
+        bp[n] = SWZ
+                    ? w + (size_t)(k0 / 32) * N * 32 +
+                          (size_t)(n0 + n * 16 + lrow) * 32 + lk * 8
+                    : w + (size_t)(n0 + n * 16 + lrow) * K + k0 + lk * 8;
 
     const int nsteps = (k1 - k0) / 32;  // K and kslice are 32-aligned
     if (nsteps > 0 && m_base < M) {
@@ -103,12 +118,12 @@ __global__ void gemm_skinny_kernel(
         // stream (measured SQ_WAIT_ANY).
         bf16x8 a0[MF], b0[GS_NF], a1[MF], b1[GS_NF];
         bf16x8 a2[MF], b2[GS_NF], a3[MF], b3[GS_NF];
-#define GS_LOAD(AV, BV, OFF)                                                   \
+#define GS_LOAD(AV, BV, S)                                                     \
     do {                                                                       \
         _Pragma("unroll") for (int f = 0; f < MF; ++f) AV[f] =                 \
-            gs_gload(ap[f] + (OFF));                                           \
+            gs_gload(ap[f] + (S) * 32);                                        \
         _Pragma("unroll") for (int n = 0; n < GS_NF; ++n) BV[n] =              \
-            gs_gload(bp[n] + (OFF));                                           \
+            gs_gload(bp[n] + (S) * (SWZ ? bstep : 32));                        \
     } while (0)
 #define GS_MFMA_ALL(AV, BV)                                                    \
     do {                                                                       \
@@ -116,10 +131,11 @@ __global__ void gemm_skinny_kernel(
             _Pragma("unroll") for (int f = 0; f < MF; ++f) acc[f][n] =         \
                 gs_mfma(AV[f], BV[n], acc[f][n]);                              \
     } while (0)
-#define GS_BUMP(ELEMS)                                                         \
+#define GS_BUMP(STEPS)                                                         \
     do {                                                                       \
-        _Pragma("unroll") for (int f = 0; f < MF; ++f) ap[f] += (ELEMS);       \
-        _Pragma("unroll") for (int n = 0; n < GS_NF; ++n) bp[n] += (ELEMS);    \
+        _Pragma("unroll") for (int f = 0; f < MF; ++f) ap[f] += (STEPS) * 32;  \
+        _Pragma("unroll") for (int n = 0; n < GS_NF; ++n)                      \
+            bp[n] += (STEPS) * (SWZ ? bstep : 32);                             \
     } while (0)
         if constexpr (MF <= 2) {
             // depth-4 pipeline: fits registers at MF<=2 and fully covers
@@ -128,31 +144,38 @@ __global__ void gemm_skinny_kernel(
                 for (int s = 0; s < nsteps; ++s) {
                     GS_LOAD(a0, b0, 0);
                     GS_MFMA_ALL(a0, b0);
-                    GS_BUMP(32);
+                    GS_BUMP(1);
                 }
             } else {
                 GS_LOAD(a0, b0, 0);
-                GS_LOAD(a1, b1, 32);
-                GS_LOAD(a2, b2, 64);
+                GS_LOAD(a1, b1, 1);
+                GS_LOAD(a2, b2, 2);
                 int s = 0;
                 for (; s + 7 <= nsteps; s += 4) {  // prefetches reach s+6
-                    GS_LOAD(a3, b3, 96);
+                    GS_LOAD(a3, b3, 3);
+                    GS_FENCE;
                     GS_MFMA_ALL(a0, b0);
-                    GS_LOAD(a0, b0, 128);
+                    GS_FENCE;
+                    GS_LOAD(a0, b0, 4);
+                    GS_FENCE;
                     GS_MFMA_ALL(a1, b1);
-                    GS_LOAD(a1, b1, 160);
+                    GS_FENCE;
+                    GS_LOAD(a1, b1, 5);
+                    GS_FENCE;
                     GS_MFMA_ALL(a2, b2);
-                    GS_LOAD(a2, b2, 192);
+                    GS_FENCE;
+                    GS_LOAD(a2, b2, 6);
+                    GS_FENCE;
                     GS_MFMA_ALL(a3, b3);
-                    GS_BUMP(128);
+                    GS_BUMP(4);
                 }
                 // tail: r in [3,6]; a0/a1/a2 hold steps s, s+1, s+2
                 const int r = nsteps - s;
-                if (r >= 4) GS_LOAD(a3, b3, 96);
+                if (r >= 4) GS_LOAD(a3, b3, 3);
                 GS_MFMA_ALL(a0, b0);
-                if (r >= 5) GS_LOAD(a0, b0, 128);
+                if (r >= 5) GS_LOAD(a0, b0, 4);
                 GS_MFMA_ALL(a1, b1);
-                if (r >= 6) GS_LOAD(a1, b1, 160);
+                if (r >= 6) GS_LOAD(a1, b1, 5);
                 GS_MFMA_ALL(a2, b2);
                 if (r >= 4) GS_MFMA_ALL(a3, b3);
                 if (r >= 5) GS_MFMA_ALL(a0, b0);
@@ -163,14 +186,17 @@ __global__ void gemm_skinny_kernel(
             GS_LOAD(a0, b0, 0);
             int s = 0;
             for (; s + 2 <= nsteps - 1; s += 2) {
-                GS_LOAD(a1, b1, 32);
+                GS_LOAD(a1, b1, 1);
+                GS_FENCE;
                 GS_MFMA_ALL(a0, b0);
-                GS_LOAD(a0, b0, 64);
+                GS_FENCE;
+                GS_LOAD(a0, b0, 2);
+                GS_FENCE;
                 GS_MFMA_ALL(a1, b1);
-                GS_BUMP(64);
+                GS_BUMP(2);
             }
             if (nsteps - s == 2) {
-                GS_LOAD(a1, b1, 32);
+                GS_LOAD(a1, b1, 1);
                 GS_MFMA_ALL(a0, b0);
                 GS_MFMA_ALL(a1, b1);
             } else {
@@ -227,7 +253,7 @@ __global__ void gemm_skinny_reduce_kernel(
 
 extern "C" hipError_t launch_gemm_skinny(
     void* y, float* workspace, const void* x, const void* w, int M, int N,
-    int K, int nsk, hipStream_t stream) {
+    int K, int nsk, int swz, hipStream_t stream) {
     if (M <= 0 || M > GS_MAX_M) return hipErrorInvalidValue;
     if ((N % GS_NT) != 0 || (K % 32) != 0) return hipErrorInvalidValue;
     if (nsk < 1) return hipErrorInvalidValue;
@@ -240,9 +266,14 @@ extern "C" hipError_t launch_gemm_skinny(
     const int nwaves = ceil_div_i(M, 16 * mf);
     dim3 grid(tiles, nsk);
     dim3 block(nwaves * WAVE_SIZE);
-#define GS_LAUNCH(MFV, SPLIT)                                                  \
-    gemm_skinny_kernel<MFV, SPLIT><<<grid, block, 0, stream>>>(                \
+#define GS_LAUNCH1(MFV, SPLIT, SWZV)                                           \
+    gemm_skinny_kernel<MFV, SPLIT, SWZV><<<grid, block, 0, stream>>>(          \
         (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K, nsk)
+#define GS_LAUNCH(MFV, SPLIT)                                                  \
+    do {                                                                       \
+        if (swz) GS_LAUNCH1(MFV, SPLIT, true);                                 \
+        else GS_LAUNCH1(MFV, SPLIT, false);                                    \
+    } while (0)
     if (nsk == 1) {
         switch (mf) {
             case 1: GS_LAUNCH(1, false); break;
@@ -264,6 +295,7 @@ extern "C" hipError_t launch_gemm_skinny(
                                     stream>>>((bf16*)y, workspace, mn, nsk);
     }
 #undef GS_LAUNCH
+#undef GS_LAUNCH1
     HIP_CHECK_LAST();
     return hipSuccess;
 }

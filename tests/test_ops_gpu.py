@@ -374,3 +374,15 @@ def test_attention_decode_split_context():
     ref = reference.attention_decode(q.cpu(), kc.cpu(), vc.cpu(), perm.cpu(), ctx.cpu())
     err = (to_f32(got) - to_f32(ref)).abs().max().item()
     assert err < 0.05, f"split-path err {err}"
+
+
+@pytest.mark.parametrize("M,N,K", [(7, 4096, 4096), (64, 4096, 14336), (256, 28672, 4096)])
+def test_gemm_skinny_swizzled_matches_plain(M, N, K):
+    torch.manual_seed(M + N)
+    x = (torch.randn(M, K, device=DEV) * 0.5).bfloat16()
+    w = (torch.randn(N, K, device=DEV) * 0.02).bfloat16()
+    wz = ops.swizzle_weight(w)
+    ref = x.float() @ w.float().T
+    got = ops.linear(x, w, wz).float()
+    err = (got - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
+    assert err < 0.02, f"swizzled rel err {err}"

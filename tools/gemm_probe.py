@@ -51,7 +51,7 @@ def main():
     global SHAPES
     if len(sys.argv) == 4:  # probe a single shape (for rocprofv3 --pmc runs)
         SHAPES = [tuple(int(a) for a in sys.argv[1:4])]
-    print(f"{'M':>4} {'N':>7} {'K':>6} {'custom us':>10} {'library us':>11} {'speedup':>8}")
+    print(f"{'M':>4} {'N':>7} {'K':>6} {'custom us':>10} {'swz us':>9} {'library us':>11} {'lib/swz':>8}")
     for (M, N, K) in SHAPES:
         x = (torch.randn(M, K, device=dev) * 0.5).bfloat16()
         ncopies = max(1, min(16, (2048 << 20) // (N * K * 2)))  # rotate >2 GB: defeat the 256 MB L3
@@ -61,11 +61,16 @@ def main():
         got = ops.linear(x, ws[0]).float()
         err = (got - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
         assert err < 0.02, f"skinny GEMM wrong for {(M,N,K)}: rel {err}"
+        wz = [ops.swizzle_weight(w) for w in ws]
+        got_s = ops.linear(x, ws[0], wz[0]).float()
+        err_s = (got_s - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
+        assert err_s < 0.02, f"swizzled GEMM wrong for {(M,N,K)}: rel {err_s}"
         for _ in range(5):
-            ops.linear(x, ws[0]); F.linear(x, ws[0])
+            ops.linear(x, ws[0]); F.linear(x, ws[0]); ops.linear(x, ws[0], wz[0])
         t_c = bench(lambda i: ops.linear(x, ws[i % ncopies]))
+        t_s = bench(lambda i: ops.linear(x, ws[i % ncopies], wz[i % ncopies]))
         t_l = bench(lambda i: F.linear(x, ws[i % ncopies]))
-        print(f"{M:>4} {N:>7} {K:>6} {t_c:>10.1f} {t_l:>11.1f} {t_l / t_c:>8.2f}")
+        print(f"{M:>4} {N:>7} {K:>6} {t_c:>10.1f} {t_s:>9.1f} {t_l:>11.1f} {t_l / t_s:>8.2f}")
 
 
 if __name__ == "__main__":
