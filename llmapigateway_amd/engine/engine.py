@@ -96,6 +96,7 @@ class EngineRequest:
         self.on_finish = on_finish
         self.bt_slot: Optional[int] = None  # engine block-table row (running)
         self.num_cached = 0  # prompt tokens served from the prefix cache
+        self.prefill_pos = 0  # prompt tokens already written to the KV cache
         self.created = time.monotonic()
         self.prefill_start_time: Optional[float] = None
         self.first_token_time: Optional[float] = None
@@ -350,11 +351,14 @@ class LLMEngine:
         return produced
 
     def _prefill_step(self) -> int:
-        """One prefill forward over up to prefill_budget tokens of the
-        prefilling set. Each request's chunk attends to its previously
-        written context (prefix-cache hits and earlier chunks alike)
-        through the paged cache; a request whose chunk reaches the end of
-        its prompt samples its first token and joins the running set."""
+        """One forward over up to prefill_budget prompt tokens of the
+        prefilling set PLUS one decode row for every running sequence (a
+        "mixed" step): decode never stalls behind prefill. Each prefill
+        chunk attends to its previously written context (prefix-cache hits
+        and earlier chunks alike) through the paged cache; a request whose
+        chunk reaches the end of its prompt samples its first token and
+        joins the running set. Deferred-sampling state is always flushed
+        before this step, so decode inputs are host-visible."""
         device = self.device
         bs = self.kv.block_size
 
@@ -369,7 +373,24 @@ class LLMEngine:
                 work.append((req, req.prefill_pos, req.prefill_pos + chunk,
                              req.prefill_pos + chunk == L))
                 budget -= chunk
-        if not work:
+            # decode rows ride along: ensure every running seq has a block
+            # for its incoming token (same preemption policy as decode)
+            i = 0
+            while i < len(self.running):
+                req = self.running[i]
+                try:
+                    before = len(req.block_table)
+                    self.kv.manager.extend(req.block_table, req.num_tokens, req.num_tokens + 1)
+                    if len(req.block_table) != before:
+                        self._bt_np[req.bt_slot, before : len(req.block_table)] = (
+                            req.block_table[before:]
+                        )
+                    i += 1
+                except RuntimeError:
+                    if not self._preempt_youngest():
+                        raise
+            dec_reqs = list(self.running)
+        if not work and not dec_reqs:
             return 0
 
         reqs = [w[0] for w in work]
@@ -400,9 +421,29 @@ class LLMEngine:
                 np.asarray(starts, dtype=np.int32)
             ).to(device)
 
+        # decode rows appended after the prefill rows (mixed step)
+        Tp = int(cu[-1])
+        nd = len(dec_reqs)
+        dec_bt_t = dec_ctx_t = None
+        if nd:
+            d_tokens = np.fromiter(
+                ((r.out_ids[-1] if r.out_ids else r.prompt_ids[-1]) for r in dec_reqs),
+                dtype=np.int64, count=nd,
+            )
+            d_pos = np.fromiter((r.num_tokens - 1 for r in dec_reqs), dtype=np.int64, count=nd)
+            d_rows = np.fromiter((r.bt_slot for r in dec_reqs), dtype=np.intp, count=nd)
+            d_tables = self._bt_np[d_rows]
+            d_slots = d_tables[np.arange(nd), d_pos // bs].astype(np.int64) * bs + d_pos % bs
+            token_ids = np.concatenate([token_ids, d_tokens])
+            positions = np.concatenate([positions, d_pos])
+            slots = np.concatenate([slots, d_slots])
+            logits_idx = np.concatenate([logits_idx, Tp + np.arange(nd, dtype=np.int64)])
+            dec_bt_t = torch.from_numpy(d_tables).to(device)
+            dec_ctx_t = torch.from_numpy((d_pos + 1).astype(np.int32)).to(device)
+
         tile_seq, tile_off = ops.build_prefill_tiles(lens, device)
         batch = ForwardBatch(
-            kind="prefill",
+            kind="mixed" if nd else "prefill",
             token_ids=torch.from_numpy(token_ids).to(device),
             positions=torch.from_numpy(positions).to(device),
             slot_mapping=torch.from_numpy(slots).to(device),
@@ -412,13 +453,19 @@ class LLMEngine:
             tile_off=tile_off,
             block_tables=block_tables_t,
             cached_lens=cached_lens_t,
+            n_prefill_tokens=Tp,
+            dec_block_tables=dec_bt_t,
+            dec_context_lens=dec_ctx_t,
             logits_indices=torch.from_numpy(logits_idx).to(device),
         )
         logits = self.model.forward(batch, self.kv.k_caches, self.kv.v_caches)
         finals = [w[0] for w in work if w[3]]
-        tokens = self._sample(logits, reqs)  # non-final rows are discarded
+        sample_reqs = reqs + dec_reqs
+        tokens = self._sample(logits, sample_reqs)  # non-final prefill rows discarded
         final_tokens = [t for t, w in zip(tokens, work) if w[3]]
-        self.stats["prefill_tokens"] += len(token_ids)
+        dec_tokens = tokens[len(reqs):]
+        self.stats["prefill_tokens"] += Tp
+        self.stats["decode_tokens"] += nd
         if self.prefix_caching:
             # KV for this chunk is now written (stream-ordered before any
             # later forward): make completed full prompt blocks reusable
@@ -429,9 +476,11 @@ class LLMEngine:
                 req.prefill_pos = e
                 if final:
                     self.prefilling.remove(req)
+            if nd:
+                self._deliver(dec_reqs, dec_tokens)
             self.running.extend(finals)
             self._deliver(finals, final_tokens)
-        return len(finals)
+        return len(finals) + nd
 
     def _decode_step(self) -> int:
         device = self.device

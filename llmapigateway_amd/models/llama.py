@@ -36,7 +36,7 @@ from .configs import ModelConfig
 
 @dataclass
 class ForwardBatch:
-    kind: str  # "prefill" | "decode"
+    kind: str  # "prefill" | "decode" | "mixed"
     token_ids: torch.Tensor          # [T] int64
     positions: torch.Tensor          # [T] int64
     slot_mapping: torch.Tensor       # [T] int64 (global KV slots; -1 = don't write)
@@ -53,6 +53,13 @@ class ForwardBatch:
     cached_lens: Optional[torch.Tensor] = None   # [B] int32
     # rows of the flat batch at which logits are needed (last token per seq)
     logits_indices: Optional[torch.Tensor] = None  # [B] int64
+    # mixed steps: rows [0, n_prefill_tokens) are prefill (cu_seqlens etc.
+    # apply to them), rows [n_prefill_tokens, T) are decode rows with their
+    # own tables/lens (decode never shares block_tables with the cached-
+    # prefill phase in a mixed step)
+    n_prefill_tokens: int = 0
+    dec_block_tables: Optional[torch.Tensor] = None  # [Bd, max_blocks] int32
+    dec_context_lens: Optional[torch.Tensor] = None  # [Bd] int32
 
 
 class LlamaModel:
@@ -199,6 +206,21 @@ class LlamaModel:
                     block_tables=batch.block_tables,
                     cached_lens=batch.cached_lens,
                 )
+            elif batch.kind == "mixed":
+                Tp = batch.n_prefill_tokens
+                attn_p = ops.attention_prefill(
+                    q[:Tp], k[:Tp], v[:Tp], batch.cu_seqlens, batch.max_seqlen,
+                    self.scale, tile_seq=batch.tile_seq, tile_off=batch.tile_off,
+                    k_cache=k_caches[i] if batch.cached_lens is not None else None,
+                    v_cache=v_caches[i] if batch.cached_lens is not None else None,
+                    block_tables=batch.block_tables,
+                    cached_lens=batch.cached_lens,
+                )
+                attn_d = ops.attention_decode(
+                    q[Tp:], k_caches[i], v_caches[i],
+                    batch.dec_block_tables, batch.dec_context_lens, self.scale,
+                )
+                attn = torch.cat([attn_p, attn_d], dim=0)
             else:
                 attn = ops.attention_decode(
                     q, k_caches[i], v_caches[i], batch.block_tables, batch.context_lens, self.scale
