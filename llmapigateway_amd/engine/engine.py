@@ -373,23 +373,31 @@ class LLMEngine:
                 work.append((req, req.prefill_pos, req.prefill_pos + chunk,
                              req.prefill_pos + chunk == L))
                 budget -= chunk
-            # decode rows ride along: ensure every running seq has a block
-            # for its incoming token (same preemption policy as decode)
-            i = 0
-            while i < len(self.running):
-                req = self.running[i]
-                try:
-                    before = len(req.block_table)
-                    self.kv.manager.extend(req.block_table, req.num_tokens, req.num_tokens + 1)
-                    if len(req.block_table) != before:
-                        self._bt_np[req.bt_slot, before : len(req.block_table)] = (
-                            req.block_table[before:]
-                        )
-                    i += 1
-                except RuntimeError:
-                    if not self._preempt_youngest():
-                        raise
-            dec_reqs = list(self.running)
+            # decode rows ride along ONLY while some prompt is being
+            # chunked (a whole-prompt prefill is one short step; adding
+            # decode rows to it desyncs batch-uniform workloads and costs
+            # the hipGraph path for no stall-avoidance benefit)
+            chunking = any(
+                (start > req.num_cached) or (not final)
+                for (req, start, end, final) in work
+            )
+            dec_reqs: List[EngineRequest] = []
+            if chunking:
+                i = 0
+                while i < len(self.running):
+                    req = self.running[i]
+                    try:
+                        before = len(req.block_table)
+                        self.kv.manager.extend(req.block_table, req.num_tokens, req.num_tokens + 1)
+                        if len(req.block_table) != before:
+                            self._bt_np[req.bt_slot, before : len(req.block_table)] = (
+                                req.block_table[before:]
+                            )
+                        i += 1
+                    except RuntimeError:
+                        if not self._preempt_youngest():
+                            raise
+                dec_reqs = list(self.running)
         if not work and not dec_reqs:
             return 0
 
