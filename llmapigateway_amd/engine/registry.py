@@ -162,6 +162,7 @@ class EngineRegistry:
                 hbm_fraction=self.settings.engine_hbm_fraction,
                 num_blocks=None if device.startswith("cuda") else 256,
                 prefix_caching=self.settings.engine_prefix_caching,
+                use_hipgraph=None if self.settings.engine_use_hipgraph else False,
             )
             if engine.graph_runner is not None:
                 logger.info("pre-capturing decode hipGraphs for %s", key)
