@@ -7,6 +7,8 @@ Behavior is selected by the requested model name:
 - "errchunk"      -> SSE whose first real data frame carries an error
 - "slowsplit"     -> SSE frames split across odd byte boundaries
 - "flaky:<n>"     -> fails (500) the first n requests per model string, then ok
+- "subonly:<p>"   -> succeeds only when payload provider.order == [p]
+                     (OpenRouter sub-provider ordering); records orders seen
 """
 
 from __future__ import annotations
@@ -18,6 +20,7 @@ from fastapi import FastAPI, Request
 from fastapi.responses import JSONResponse, StreamingResponse
 
 FLAKY_COUNTS: defaultdict[str, int] = defaultdict(int)
+SUB_ORDERS_SEEN: list = []
 
 
 def make_mock_upstream() -> FastAPI:
@@ -52,6 +55,16 @@ def make_mock_upstream() -> FastAPI:
         body = await request.json()
         model = body.get("model", "ok")
         stream = bool(body.get("stream", False))
+
+        if model.startswith("subonly:"):
+            want = model.split(":")[1]
+            order = (body.get("provider") or {}).get("order")
+            SUB_ORDERS_SEEN.append(order)
+            if order != [want]:
+                return JSONResponse(
+                    status_code=502, content={"error": f"sub-provider {order} unavailable"}
+                )
+            model = "ok"
 
         if model.startswith("flaky:"):
             n = int(model.split(":")[1])

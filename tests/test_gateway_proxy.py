@@ -44,7 +44,16 @@ RULES = """
       "fallback_models": [
           { "provider": "mock", "model": "flaky:2", "retry_count": 3, "retry_delay": 1 } ] },
     { "gateway_model_name": "gw/split",
-      "fallback_models": [ { "provider": "mock", "model": "slowsplit" } ] }
+      "fallback_models": [ { "provider": "mock", "model": "slowsplit" } ] },
+    { "gateway_model_name": "gw/suborder",
+      "fallback_models": [
+          { "provider": "mock", "model": "subonly:good",
+            "use_provider_order_as_fallback": true,
+            "providers_order": ["bad", "good"] } ] },
+    { "gateway_model_name": "gw/suborder-pinned",
+      "fallback_models": [
+          { "provider": "mock", "model": "subonly:good",
+            "providers_order": ["bad", "good"] } ] }
 ]
 """
 
@@ -329,3 +338,31 @@ def test_custom_params_injected(client, monkeypatch):
     assert seen["payload"]["reasoning_effort"] == "high"
     assert seen["headers"]["x-demo"] == "1"
     assert seen["payload"]["model"] == "ok"
+
+
+def test_sub_provider_order_as_fallback(tmp_path):
+    """OpenRouter sub-provider ordering (reference chat.py:159-189): with
+    use_provider_order_as_fallback, one attempt per sub-provider in order,
+    each pinned with allow_fallbacks semantics."""
+    from tests.mock_upstream import SUB_ORDERS_SEEN
+
+    client = TestClient(build_app(tmp_path))
+    with client:
+        SUB_ORDERS_SEEN.clear()
+        r = chat(client, "gw/suborder")
+        assert r.status_code == 200
+        assert SUB_ORDERS_SEEN == [["bad"], ["good"]]
+
+
+def test_sub_provider_order_pinned_list(tmp_path):
+    """Without the fallback flag the whole order list is sent in ONE
+    attempt (chat.py:150-156) — the mock only accepts ["good"], so the
+    pinned ["bad", "good"] attempt fails and the request 503s."""
+    from tests.mock_upstream import SUB_ORDERS_SEEN
+
+    client = TestClient(build_app(tmp_path))
+    with client:
+        SUB_ORDERS_SEEN.clear()
+        r = chat(client, "gw/suborder-pinned")
+        assert r.status_code == 503
+        assert SUB_ORDERS_SEEN == [["bad", "good"]]
