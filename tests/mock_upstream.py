@@ -9,6 +9,8 @@ Behavior is selected by the requested model name:
 - "flaky:<n>"     -> fails (500) the first n requests per model string, then ok
 - "subonly:<p>"   -> succeeds only when payload provider.order == [p]
                      (OpenRouter sub-provider ordering); records orders seen
+- "echo"          -> non-streaming; response content carries the request
+                     body + selected headers (param-injection assertions)
 """
 
 from __future__ import annotations
@@ -55,6 +57,18 @@ def make_mock_upstream() -> FastAPI:
         body = await request.json()
         model = body.get("model", "ok")
         stream = bool(body.get("stream", False))
+
+        if model == "echo":
+            return JSONResponse(content={
+                "id": "cmpl-echo", "object": "chat.completion", "model": model,
+                "choices": [{"index": 0, "finish_reason": "stop",
+                             "message": {"role": "assistant", "content": json.dumps({
+                                 "body": body,
+                                 "x_demo": request.headers.get("x-demo"),
+                                 "auth": request.headers.get("authorization"),
+                             })}}],
+                "usage": {"prompt_tokens": 1, "completion_tokens": 1, "total_tokens": 2},
+            })
 
         if model.startswith("subonly:"):
             want = model.split(":")[1]

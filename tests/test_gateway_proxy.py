@@ -50,6 +50,11 @@ RULES = """
           { "provider": "mock", "model": "subonly:good",
             "use_provider_order_as_fallback": true,
             "providers_order": ["bad", "good"] } ] },
+    { "gateway_model_name": "gw/inject",
+      "fallback_models": [
+          { "provider": "mock", "model": "echo",
+            "custom_body_params": {"reasoning_effort": "high", "service_tier": "flex"},
+            "custom_headers": {"x-demo": "42"} } ] },
     { "gateway_model_name": "gw/suborder-pinned",
       "fallback_models": [
           { "provider": "mock", "model": "subonly:good",
@@ -366,3 +371,22 @@ def test_sub_provider_order_pinned_list(tmp_path):
         r = chat(client, "gw/suborder-pinned")
         assert r.status_code == 503
         assert SUB_ORDERS_SEEN == [["bad", "good"]]
+
+
+def test_custom_params_and_headers_injected(tmp_path):
+    """custom_body_params / custom_headers reach the upstream verbatim
+    (reference chat.py:116-123), plus the provider Bearer key."""
+    client = TestClient(build_app(tmp_path))
+    with client:
+        r = client.post(
+            "/v1/chat/completions",
+            json={"model": "gw/inject", "messages": [{"role": "user", "content": "hi"}],
+                  "stream": False},
+        )
+        assert r.status_code == 200
+        echoed = json.loads(r.json()["choices"][0]["message"]["content"])
+        assert echoed["body"]["reasoning_effort"] == "high"
+        assert echoed["body"]["service_tier"] == "flex"
+        assert echoed["body"]["model"] == "echo"  # per-entry model override
+        assert echoed["x_demo"] == "42"
+        assert echoed["auth"] == "Bearer MOCK_KEY"
