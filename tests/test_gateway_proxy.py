@@ -390,3 +390,27 @@ def test_custom_params_and_headers_injected(tmp_path):
         assert echoed["body"]["model"] == "echo"  # per-entry model override
         assert echoed["x_demo"] == "42"
         assert echoed["auth"] == "Bearer MOCK_KEY"
+
+
+def test_models_degrades_when_upstream_unreachable(tmp_path):
+    """Upstream /models failure degrades to the gateway-rule list
+    (reference models.py:239-296 graceful path)."""
+    (tmp_path / "providers.json").write_text(
+        PROVIDERS.replace('"http://mock.test/v1", "apikey": "MOCK_KEY"',
+                          '"http://unreachable.test/v1", "apikey": "MOCK_KEY"')
+    )
+    (tmp_path / "models_fallback_rules.json").write_text(RULES)
+    settings = Settings(fallback_provider="mock", gateway_api_key=None)
+    app = create_app(
+        settings=settings,
+        providers_path=str(tmp_path / "providers.json"),
+        fallback_rules_path=str(tmp_path / "models_fallback_rules.json"),
+        db_dir=str(tmp_path / "db"),
+        log_dir=str(tmp_path / "logs"),
+    )
+    # no mock transport wired: the /models fetch truly fails
+    with TestClient(app) as client:
+        r = client.get("/v1/models")
+        assert r.status_code == 200
+        ids = [m["id"] for m in r.json()["data"]]
+        assert ids and all(i.startswith("gw/") for i in ids)
