@@ -6,11 +6,16 @@ model + paged KV cache; the gateway's fallback loop treats engine errors
 (including injected faults) exactly like upstream HTTP errors — raised
 before the first streamed byte so fallback can engage.
 
-Scheduling policy (vLLM-v0-style): each step() is either one prefill
-forward over newly admitted prompts (their first token samples from that
-same forward) or one decode forward over all running sequences. Decode
-out-of-block conditions preempt the youngest running sequence back to the
-waiting queue (its KV is freed; re-admission re-prefills prompt+generated).
+Scheduling policy: each step() is either a prefill forward (admitted
+prompts advance by up to prefill_budget tokens — chunked for long
+prompts, with decode rows riding along while a prompt is mid-chunk so
+decode never stalls) or a decode forward over all running sequences
+(hipGraph-replayed on GPU with deferred sampling: step s+1 enqueues
+before step s's tokens reach the host). Prompt prefixes shared with
+earlier requests are served from the content-addressed prefix cache and
+only the suffix is prefilled. Decode out-of-block conditions preempt the
+youngest running sequence back to the waiting queue (its KV is freed;
+re-admission re-prefills prompt+generated, re-hitting the prefix cache).
 """
 
 from __future__ import annotations

@@ -14,9 +14,11 @@
 //   phase A: lane = one key position; the lane streams that key's 128-dim
 //            row (16 B vector loads) and dots it against q staged in LDS
 //            (vectorized float4 broadcast reads);
-//   phase B: lanes switch to dim ownership (lane d owns dims 2d, 2d+1) and
-//            stream V rows coalesced (64 lanes x 4 B = a 256 B row).
-// Online softmax per wave, flash-style cross-wave combine in LDS at the end.
+//   phase B: half-wave per key, lane owns a dim quad (dwordx2 loads,
+//            2 keys per instruction), cross-half shfl_xor fold.
+// Online softmax per wave, flash-style cross-wave combine in LDS at the
+// end; long-context small-batch launches split the context across grid.z
+// (flash-decode partials + combine kernel).
 
 #include "common.h"
 

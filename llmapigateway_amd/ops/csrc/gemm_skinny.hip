@@ -9,14 +9,15 @@
 // - grid = (N/64 n-stripes) x (split-K slices). One workgroup owns ALL M
 //   rows of a 64-wide N stripe for its K-slice, so W is never re-read and
 //   the X re-read factor is only N/64 (X lives in the per-XCD L2s).
-// - 4 waves; wave w owns M rows [w*16*MF, ..): MF m-fragments x NF=4
-//   n-fragments of v_mfma_f32_16x16x32_bf16 per 32-deep k-step. All
+// - up to 8 waves; wave w owns M rows [w*16*MF, ..): MF m-fragments x
+//   NF=4 n-fragments of v_mfma_f32_16x16x32_bf16 per 32-deep k-step. All
 //   fragments are 16 B dwordx4 loads straight from global (B/W coalesced
-//   by 16 rows x 64 B; A/X from L2).
-// - software pipeline: k-step s+1's 8 loads issue before s's 16 MFMAs,
-//   double-buffered with an unroll-2 ping-pong so the compiler keeps the
-//   accumulators pinned in AGPRs (a rotating-buffer formulation produced
-//   ~33 v_accvgpr copies per iteration and a full vmcnt(0) drain).
+//   by 16 rows x 64 B, or fully contiguous with the opt-in k-major
+//   swizzle; A/X from L2).
+// - software pipeline up to depth 4, held in shape by sched_barriers
+//   (see profiles/r01_gemm_skinny_probe.md for the measured history).
+//   Production dispatch is the latency regime (M <= 16, ops/__init__.py);
+//   the tuned library keeps M >= 64 (its LDS macro-tiles re-read X less).
 // - split-K writes private fp32 slabs (each workgroup fully writes its
 //   [M x 64] stripe of slab blockIdx.y — no zero-fill, no atomics, no
 //   cross-workgroup visibility hazard); a second tiny kernel reduces the
