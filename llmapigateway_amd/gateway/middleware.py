@@ -207,6 +207,12 @@ class ChatAccountingMiddleware:
 
             provider = state.get("served_provider")
             model = state.get("served_model") or sniffer.model
+            if sniffer.error_detail:  # mid-stream error chunk: log-only,
+                # bytes already reached the client (request_handler.py:125-133)
+                logger.warning(
+                    "Error chunk mid-stream from provider %s: %s",
+                    provider, str(sniffer.error_detail)[:500],
+                )
             if sniffer.usage and self.usage_db is not None:
                 fields = sse.token_usage_fields(sniffer.usage)
                 self.usage_db.insert_usage(model=model, provider=provider, **fields)
