@@ -88,12 +88,15 @@
     const { engines } = await r.json();
     document.getElementById("enginesTable").innerHTML = engines.length
       ? table(
-          ["Engine", "Device", "Model", "Waiting", "Running", "KV free/total",
-           "Requests", "Finished", "Failed", "Prefill toks", "Decode toks"],
+          ["Engine", "Device", "Model", "Waiting", "Prefilling", "Running",
+           "KV free/total", "Requests", "Finished", "Failed",
+           "Prefill toks", "Decode toks", "Prefix hits", "Prefix toks"],
           engines.map((e) => [
-            e.engine, e.device, e.model, e.waiting, e.running,
-            `${e.kv_blocks_free}/${e.kv_blocks_total}`, fmt(e.requests),
-            fmt(e.finished), fmt(e.failed), fmt(e.prefill_tokens), fmt(e.decode_tokens)]))
+            e.engine, e.device, e.model, e.waiting ?? "-", e.prefilling ?? "-",
+            e.running ?? "-",
+            `${e.kv_blocks_free ?? "-"}/${e.kv_blocks_total ?? "-"}`, fmt(e.requests),
+            fmt(e.finished), fmt(e.failed), fmt(e.prefill_tokens), fmt(e.decode_tokens),
+            fmt(e.prefix_cache_hits ?? 0), fmt(e.prefix_cached_tokens ?? 0)]))
       : "<p class='hint'>No local engines running.</p>";
   }
   document.getElementById("refreshEngines").onclick = loadEngines;
