@@ -146,6 +146,7 @@ class LLMEngine:
         tp_size: int = 1,
         prefix_caching: bool = False,
         prefill_budget: int = 8192,
+        tokenizer: Optional[object] = None,
     ):
         full_config = get_model_config(model) if isinstance(model, str) else model
         self.full_config = full_config
@@ -157,6 +158,11 @@ class LLMEngine:
             max_model_len or full_config.max_positions, full_config.max_positions
         )
         self.max_batch_size = max_batch_size
+        if tokenizer is None:
+            from .tokenizer import ByteTokenizer
+
+            tokenizer = ByteTokenizer()
+        self.tokenizer = tokenizer  # stop-string decode + EOS detection
 
         if self.device.type == "cuda":
             torch.cuda.set_device(self.device)
@@ -479,12 +485,14 @@ class LLMEngine:
         dec_tokens = tokens[len(reqs):]
         self.stats["prefill_tokens"] += Tp
         self.stats["decode_tokens"] += nd
-        if self.prefix_caching:
-            # KV for this chunk is now written (stream-ordered before any
-            # later forward): make completed full prompt blocks reusable
-            for req in finals:
-                self.kv.manager.register_prefix(req.prompt_ids, req.block_table)
         with self._lock:
+            if self.prefix_caching:
+                # KV for this chunk is now written (stream-ordered before
+                # any later forward): make completed full prompt blocks
+                # reusable. Under the lock — abort/_finish on gateway
+                # threads mutate the same refcount tables via kv.manager.
+                for req in finals:
+                    self.kv.manager.register_prefix(req.prompt_ids, req.block_table)
             for req, s, e, final in work:
                 req.prefill_pos = e
                 if final:
@@ -592,7 +600,9 @@ class LLMEngine:
             )
             logits = self.model.forward(batch, self.kv.k_caches, self.kv.v_caches)
 
-        tokens_dev = self._sample_dev(logits, reqs)
+        tokens_dev = self._sample_dev(
+            logits, reqs, noise_pos=[len(r.out_ids) + inflight for r in reqs]
+        )
         self.stats["decode_tokens"] += n
         if self.async_sampling:
             # previous pending was either flushed or belongs to these same
@@ -639,8 +649,20 @@ class LLMEngine:
         ] -= torch.tensor(vals, dtype=logits.dtype, device=dev)
         return logits
 
-    def _sample_dev(self, logits: torch.Tensor, reqs: List[EngineRequest]) -> torch.Tensor:
-        """Sample next tokens; returns an int64 device tensor (no host sync)."""
+    def _sample_dev(
+        self,
+        logits: torch.Tensor,
+        reqs: List[EngineRequest],
+        noise_pos: Optional[List[int]] = None,
+    ) -> torch.Tensor:
+        """Sample next tokens; returns an int64 device tensor (no host sync).
+
+        ``noise_pos[i]`` is the output position being sampled for request i —
+        callers on the deferred-sampling decode path pass
+        ``len(out_ids) + inflight`` because the previous step's token has not
+        reached ``out_ids`` yet (keying on bare len(out_ids) would reuse the
+        same seeded noise vector for two consecutive tokens after every
+        flush boundary)."""
         logits = self._apply_penalties(logits, reqs)
         any_temp = any(r.params.temperature > 0 for r in reqs)
         temps = torch.tensor(
@@ -656,8 +678,9 @@ class LLMEngine:
             # position, independent of batch composition
             for i, r in enumerate(reqs):
                 if r.params.seed is not None and r.params.temperature > 0:
+                    pos = noise_pos[i] if noise_pos is not None else len(r.out_ids)
                     g = torch.Generator(device=logits.device).manual_seed(
-                        (int(r.params.seed) << 20) ^ len(r.out_ids)
+                        (int(r.params.seed) << 20) ^ pos
                     )
                     noise[i] = torch.rand(
                         logits.shape[-1], generator=g, device=logits.device,
@@ -697,8 +720,6 @@ class LLMEngine:
     # ---- delivery / lifecycle (call with lock held) ----
     def _deliver(self, reqs: List[EngineRequest], tokens: List[int]) -> None:
         now = time.monotonic()
-        from .tokenizer import ByteTokenizer
-
         # pass 1: record tokens; batch-notify consumers in ONE cross-thread
         # event (the gateway registers batch_notifier — per-token
         # call_soon_threadsafe wakeups saturate the event loop ~10K/s)
@@ -708,7 +729,7 @@ class LLMEngine:
                 continue
             req.out_ids.append(int(tok))
             if req.params.stop:
-                req.text += ByteTokenizer().decode([int(tok)])
+                req.text += self.tokenizer.decode([int(tok)])
             if req.first_token_time is None:
                 req.first_token_time = now
             if req.on_token is not None:
@@ -734,9 +755,7 @@ class LLMEngine:
                 self._finish(req, reason)
 
     def _finish_reason(self, req: EngineRequest, tok: int) -> Optional[str]:
-        from .tokenizer import ByteTokenizer
-
-        if not req.params.ignore_eos and tok == ByteTokenizer.EOS:
+        if not req.params.ignore_eos and tok in self.tokenizer.eos_ids:
             return "stop"
         if req.params.stop and any(s in req.text for s in req.params.stop):
             return "stop"

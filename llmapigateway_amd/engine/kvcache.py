@@ -134,12 +134,18 @@ class BlockManager:
             cached.append(blk)
             parent = blk
         total = self.blocks_needed(len(prompt_ids))
+        # Pin the matched cached blocks FIRST: _alloc_raw evicts refcount-0
+        # cached blocks under pressure, and without the ref it could hand a
+        # matched prefix block back as this request's own suffix block
+        # (read-as-prefix + write-as-suffix aliasing -> silent corruption).
+        for blk in cached:
+            self._ref(blk)
         try:
             private = self._alloc_raw(total - len(cached))
         except RuntimeError:
+            for blk in cached:
+                self._unref(blk)
             raise
-        for blk in cached:
-            self._ref(blk)
         if cached:
             self.stats_prefix_hits += 1
             self.stats_prefix_tokens += len(cached) * bs

@@ -17,6 +17,7 @@ class ByteTokenizer:
 
     def __init__(self, vocab_size: int = 128256):
         self.vocab_size = vocab_size
+        self.eos_ids = frozenset({self.EOS})
 
     def encode(self, text: str, add_bos: bool = True) -> List[int]:
         ids = [self.BOS] if add_bos else []

@@ -329,6 +329,14 @@ class TPEngineClient:
             self.inbox.put(("stop",))
         except Exception:
             pass
+        # fail in-flight requests NOW (mirrors _watch): registry.prune()
+        # calls stop() while clients may still be streaming — without this
+        # they would hang until the first-token timeout
+        with self._lock:
+            pending = list(self._reqs.values())
+        for req in pending:
+            req.error = self._dead
+            self._finish_local(req, "error", "failed")
         for p in self.procs:
             p.join(timeout=10.0)
         self._terminate()

@@ -65,6 +65,42 @@ def test_manager_eviction_under_pressure():
     assert nc3 == 0
 
 
+def test_manager_prefix_hit_never_aliases_suffix_under_pressure():
+    """Regression: matched cached prefix blocks must be
+    pinned BEFORE the private-tail allocation — otherwise under pressure
+    _alloc_raw evicts those refcount-0 blocks and hands one back as the
+    request's own suffix block (read-as-prefix + write-as-suffix alias)."""
+    m = BlockManager(4, 4, prefix_caching=True)
+    p1 = list(range(9))  # 3 blocks: 2 full cacheable + 1 tail
+    t1, _ = m.allocate_with_prefix(p1)
+    m.register_prefix(p1, t1)
+    m.free(t1)  # both full blocks now cached + evictable; allocator has 1 free
+    t2, nc2 = m.allocate_with_prefix(p1)
+    assert nc2 == 8
+    assert len(set(t2)) == len(t2), f"aliased block table {t2}"
+    m.free(t2)
+
+
+def test_manager_prefix_alloc_failure_unpins_cached():
+    """On private-tail allocation failure the matched cached blocks must be
+    unpinned again (left evictable), not leaked with a stale refcount."""
+    m = BlockManager(4, 4, prefix_caching=True)
+    p1 = list(range(9))
+    t1, _ = m.allocate_with_prefix(p1)
+    m.register_prefix(p1, t1)
+    m.free(t1)
+    hold = m.allocate(8)  # take the remaining 2 blocks (evicts nothing? no:
+    # allocator has 1 free + 2 evictable; taking 2 blocks evicts 1 cached)
+    import pytest
+
+    big = list(range(100)) * 2  # needs far more blocks than exist
+    with pytest.raises(RuntimeError):
+        m.allocate_with_prefix(big)
+    # any cached blocks matched during the failed attempt are evictable again
+    assert all(m._refs.get(b, 0) == 0 for b in m._evictable)
+    m.free(hold)
+
+
 def test_manager_full_block_prompt_leaves_suffix():
     m = BlockManager(16, 4, prefix_caching=True)
     prompt = list(range(8))  # exactly 2 blocks: at most 1 may come cached

@@ -84,6 +84,31 @@ def test_tp2_abort(tp_client):
     assert req.finish_reason == "aborted"
 
 
+def test_stop_fails_inflight_requests_immediately():
+    """Regression: registry.prune() stops a TP group while
+    clients are streaming; stop() must fail pending requests right away
+    instead of leaving them to the 300 s first-token timeout."""
+    client = TPEngineClient(
+        model="tiny-llama", tp=2, max_batch_size=8, kv_block_size=16, num_blocks=64,
+        start_timeout=180.0,
+    )
+    import threading
+    import time
+
+    done = threading.Event()
+    req = EngineRequest(
+        PROMPT,
+        SamplingParams(max_tokens=100_000, ignore_eos=True),
+        on_finish=lambda r: done.set(),
+    )
+    client.add_request(req)
+    t0 = time.monotonic()
+    client.stop()
+    assert done.wait(timeout=15.0), "stop() left the in-flight request hanging"
+    assert time.monotonic() - t0 < 15.0
+    assert req.state == "failed" and req.finish_reason == "error"
+
+
 def test_registry_tp_spec_creates_group():
     from llmapigateway_amd.config.loader import EngineSpec
     from llmapigateway_amd.engine.registry import EngineRegistry, _TPGroupHandle
