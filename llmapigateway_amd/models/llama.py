@@ -139,20 +139,26 @@ class LlamaModel:
         self.final_norm = torch.ones(hidden, dtype=self.dtype, device=self.device)
         self.lm_head = self.embed if c.tie_embeddings else mk(self.full_config.vocab_size, hidden, std)
 
-        # k-major swizzled twins of the decode projections for the
-        # weight-streaming skinny GEMM (contiguous 4 KB B-tile streams;
-        # ops/csrc/gemm_skinny.hip). Measured: 30-40% faster than the
-        # plain-layout custom kernel but still behind the tuned library
-        # at M>=64 cold (profiles/r01_gemm_skinny_probe.md), so the extra
-        # weight copy is opt-in.
+        # Fragment-major twins of the decode projections for the
+        # macro-tile LDS-staged GEMM (ops/csrc/gemm_m256.hip) — ON by
+        # default: the twin copy costs ~weights-again for the projections
+        # (~14 GB for llama-3-8b, cheap against 288 GB HBM) and buys the
+        # decode step its dominant GEMM time back from the library
+        # (profiles/r02_gemm_m256_probe.md). LLMAPI_NO_FRAG_WEIGHTS=1
+        # opts out; LLMAPI_SWZ_WEIGHTS=1 selects the legacy k-major twins
+        # (gemm_skinny streaming form) for comparison runs.
         import os as _os
 
-        if self.device.type == "cuda" and _os.environ.get("LLMAPI_SWZ_WEIGHTS"):
+        if self.device.type == "cuda" and not _os.environ.get("LLMAPI_NO_FRAG_WEIGHTS"):
+            legacy = bool(_os.environ.get("LLMAPI_SWZ_WEIGHTS"))
             for layer in self.layers:
                 for name in ("qkv", "o", "gate_up", "down"):
                     w = layer[name]
-                    if w.shape[0] % 64 == 0 and w.shape[1] % 32 == 0:
-                        layer[name + "_swz"] = ops.swizzle_weight(w)
+                    if legacy:
+                        if w.shape[0] % 64 == 0 and w.shape[1] % 32 == 0:
+                            layer[name + "_swz"] = ops.swizzle_weight(w)
+                    elif w.shape[0] % 64 == 0 and w.shape[1] % 64 == 0:
+                        layer[name + "_swz"] = ops.swizzle_weight_frag(w)
 
     def param_bytes(self) -> int:
         total = self.embed.numel() + self.final_norm.numel()

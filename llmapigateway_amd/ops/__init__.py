@@ -296,19 +296,71 @@ def swizzle_weight(w: torch.Tensor) -> torch.Tensor:
     return w.view(N, K // 32, 32).permute(1, 0, 2).contiguous()
 
 
+def swizzle_weight_frag(w: torch.Tensor) -> torch.Tensor:
+    """[N, K] -> fragment-major [K/32, N/16, 64, 8] for gemm_m256.
+
+    Element [k32][n16][lane][e] = W[n16*16 + (lane&15)][k32*32 + (lane>>4)*8
+    + e]: each 1 KiB row is exactly one wave's mfma_f32_16x16x32_bf16
+    B-fragment in lane order, so the kernel's global_load_lds staging is a
+    straight contiguous copy and its ds_read_b128 is lane-linear
+    (conflict-free per the gfx950 b128 lane groups)."""
+    N, K = w.shape
+    assert K % 64 == 0 and N % 64 == 0
+    return (
+        w.view(N // 16, 16, K // 32, 4, 8)
+        .permute(2, 0, 3, 1, 4)
+        .reshape(K // 32, N // 16, 64, 8)
+        .contiguous()
+    )
+
+
+def _m256_nsk(N: int, K: int, nf: int) -> int:
+    """Split-K factor for gemm_m256: fill the 256 CUs (one 8-wave block
+    per CU at the 120 KB LDS ring) when the column-tile count alone
+    cannot. The fp32 slab traffic (2*nsk*M*N*4 B) is the price, so tiles
+    >= 256 never split; smaller-N shapes are load-path/latency-bound and
+    the slab round trip is cheap relative to the CU-fill win
+    (profiles/r02_gemm_m256_probe.md)."""
+    tiles = N // (16 * nf)
+    if tiles >= 256:
+        return 1
+    nsk = -(-256 // tiles)
+    return max(1, min(nsk, (K // 64) // 2, 8))
+
+
 # swizzled weights raise the profitable dispatch ceiling (contiguous
 # streams); plain-layout dispatch stays in the latency regime
 _SKINNY_SWZ_MAX_M = int(os.environ.get("SKINNY_GEMM_SWZ_MAX_M", "256"))
+
+
+def gemm_m256(
+    x: torch.Tensor, w_frag: torch.Tensor, nf: Optional[int] = None,
+    nsk: Optional[int] = None,
+) -> torch.Tensor:
+    """y = x @ w.T with w pre-swizzled fragment-major (swizzle_weight_frag).
+    The macro-tile LDS-staged decode GEMM (csrc/gemm_m256.hip); M <= 256."""
+    M, K = x.shape
+    N = w_frag.shape[1] * 16
+    if nf is None:
+        nf = 8 if N // 64 >= 448 else 4  # wide tiles only where blocks abound
+    if N % (16 * nf) != 0:
+        nf = 4
+    if nsk is None:
+        nsk = _m256_nsk(N, K, nf)
+    y = torch.empty((M, N), dtype=torch.bfloat16, device=x.device)
+    ws = _skinny_scratch(x.device, nsk * M * N) if nsk > 1 else None
+    _native().gemm_m256(y, x, w_frag, ws, nsk, nf)
+    return y
 
 
 def linear(
     x: torch.Tensor, w: torch.Tensor, w_swz: Optional[torch.Tensor] = None
 ) -> torch.Tensor:
     """y = x @ w.T. Decode-shaped bf16 GEMMs (M <= 256) go through the
-    weight-streaming gfx950 kernel (csrc/gemm_skinny.hip) — with a
-    pre-swizzled weight (models/llama.py builds them) its B-stream is
-    fully contiguous; everything else through the TunableOp-tuned
-    library GEMMs."""
+    macro-tile gfx950 kernel (csrc/gemm_m256.hip) when the fragment-major
+    weight twin exists (models/llama.py builds them), through the
+    streaming skinny kernel (csrc/gemm_skinny.hip) in the latency regime
+    otherwise; everything else through the TunableOp-tuned library GEMMs."""
     M = x.shape[0]
     if (
         x.is_cuda
@@ -322,7 +374,15 @@ def linear(
         and w.is_contiguous()
     ):
         N, K = w.shape
-        if w_swz is not None and M <= _SKINNY_SWZ_MAX_M:
+        if (
+            w_swz is not None
+            and w_swz.dim() == 4
+            and M <= 256
+            and K % 64 == 0
+        ):
+            return gemm_m256(x, w_swz)
+        if w_swz is not None and w_swz.dim() == 3 and M <= _SKINNY_SWZ_MAX_M:
+            # legacy k-major twin -> streaming skinny kernel
             y = torch.empty((M, N), dtype=torch.bfloat16, device=x.device)
             nsk = _skinny_nsk(N, K)
             ws = _skinny_scratch(x.device, nsk * M * N) if nsk > 1 else None
@@ -359,6 +419,9 @@ __all__ = [
     "swiglu",
     "kv_cache_write",
     "linear",
+    "gemm_m256",
+    "swizzle_weight",
+    "swizzle_weight_frag",
     "attention_prefill",
     "attention_decode",
     "sample",

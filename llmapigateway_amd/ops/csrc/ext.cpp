@@ -22,6 +22,7 @@ hipError_t launch_attention_decode(void*, const void*, const void*, const void*,
 hipError_t launch_attention_prefill(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, float, int, int, int, int64_t, int64_t, int64_t, const void*, const void*, const int*, const int*, int, int, hipStream_t);
 hipError_t launch_sample(int64_t*, const float*, const float*, const float*, float*, int*, int, int, hipStream_t);
 hipError_t launch_gemm_skinny(void*, float*, const void*, const void*, int, int, int, int, int, hipStream_t);
+hipError_t launch_gemm_m256(void*, float*, const void*, const void*, int, int, int, int, int, hipStream_t);
 }
 
 namespace {
@@ -263,6 +264,33 @@ void gemm_skinny(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                                  current_stream()));
 }
 
+void gemm_m256(torch::Tensor y, torch::Tensor x, torch::Tensor w_frag,
+               c10::optional<torch::Tensor> workspace, int64_t nsk,
+               int64_t nf) {
+    check_bf16(x, "x");
+    check_bf16(w_frag, "w_frag");
+    check_bf16(y, "y");
+    TORCH_CHECK(x.is_contiguous() && w_frag.is_contiguous() && y.is_contiguous());
+    TORCH_CHECK(x.dim() == 2 && y.dim() == 2);
+    const int M = x.size(0), K = x.size(1);
+    // fragment-major twin: [K/32, N/16, 64, 8]
+    TORCH_CHECK(w_frag.dim() == 4 && w_frag.size(0) == K / 32 &&
+                w_frag.size(2) == 64 && w_frag.size(3) == 8,
+                "w_frag must be the fragment-major [K/32, N/16, 64, 8] twin");
+    const int N = w_frag.size(1) * 16;
+    TORCH_CHECK(y.size(0) == M && y.size(1) == N);
+    float* ws = nullptr;
+    if (workspace.has_value() && workspace->defined()) {
+        TORCH_CHECK(workspace->scalar_type() == torch::kFloat32 &&
+                    workspace->numel() >= nsk * (int64_t)M * N,
+                    "workspace must be fp32 with >= nsk*M*N elements");
+        ws = workspace->data_ptr<float>();
+    }
+    CHECK_HIP(launch_gemm_m256(y.data_ptr(), ws, x.data_ptr(),
+                               w_frag.data_ptr(), M, N, K, (int)nsk, (int)nf,
+                               current_stream()));
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -277,6 +305,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("sample", &sample, "greedy / gumbel-max sampling");
     m.def("gemm_skinny", &gemm_skinny,
           "skinny-M weight-streaming GEMM y = x @ w.T (decode projections)");
+    m.def("gemm_m256", &gemm_m256,
+          "macro-tile LDS-staged decode GEMM y = x @ w.T (M <= 256)");
 
     pybind11::class_<BlockAllocator>(m, "BlockAllocator")
         .def(pybind11::init<int64_t>())

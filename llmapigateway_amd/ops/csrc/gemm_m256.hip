@@ -1,0 +1,269 @@
+// Macro-tile decode-projection GEMM for gfx950: Y[M,N] = X[M,K] · W[N,K]^T,
+// bf16, M <= 256 (the continuous-batching decode regime).
+//
+// Round-1's single-pass streaming kernel (gemm_skinny.hip) lost to the
+// tuned library at M >= 64 because its NT=64 stripes re-read X N/64 times
+// through fragment-shaped (16-row x 64 B) loads — the per-CU load path,
+// not HBM, was the wall (profiles/r01_gemm_skinny_probe.md). This kernel
+// is the macro-tile restructure:
+//
+// - One workgroup owns ALL M rows of a BN-wide column stripe for its
+//   K-slice: W is streamed exactly once and X travels the load path once
+//   per block in full 128-B lines.
+// - BOTH operands are staged into LDS with `global_load_lds` (async DMA,
+//   zero staging VGPRs). X tiles land lane-linear with the bank-conflict
+//   XOR applied to the per-lane SOURCE address (the swizzle and the
+//   ds_read XOR are the same involution); W is pre-swizzled host-side
+//   into fragment-major [K/32][N/16][64 lanes][8] order, so its staging
+//   is a straight contiguous copy and its ds_read_b128 is conflict-free
+//   with no transpose anywhere.
+// - 3-deep LDS buffer ring with counted `s_waitcnt vmcnt(N)` and raw
+//   s_barriers: up to 2 tiles of DMA stay in flight across each barrier
+//   (a plain __syncthreads() would drain the DMA queue to vmcnt(0)).
+// - split-K over grid.y writes private fp32 slabs (no atomics; every
+//   block fully writes its stripe, empty slices write zeros) reduced by
+//   gemm_skinny.hip's reduce kernel; both launches hipGraph-capturable.
+//
+// Per-wave geometry: wave w owns output rows [32w, 32w+32) — MF=2
+// m-fragments x NF n-fragments of v_mfma_f32_16x16x32_bf16 per 32-deep
+// k-step, BK=64 (2 k-steps per staged tile).
+//
+// Replaces the library GEMM on the decode hot path for the projection
+// shapes (models/llama.py builds the fragment-major twins); prefill and
+// lm_head remain on the TunableOp-tuned library.
+
+#include "common.h"
+
+#define GM_BK 64  // k-depth of one staged tile (2 MFMA k-steps)
+
+typedef __attribute__((__vector_size__(8 * sizeof(short)))) short bf16x8;
+typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4;
+
+__device__ __forceinline__ f32x4 gm_mfma(bf16x8 a, bf16x8 b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+
+// async 16-B-per-lane global->LDS copy. The LDS destination is
+// wave-uniform base + lane*16 (hardware rule); the global source is
+// per-lane. `ldsoff` must be wave-uniform — computed from wave id, which
+// is uniform in fact but not provably so to the compiler (T20), hence the
+// readfirstlane.
+__device__ __forceinline__ void gm_glds16(const void* g, char* smem, int ldsoff) {
+    ldsoff = __builtin_amdgcn_readfirstlane(ldsoff);
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)g,
+        (__attribute__((address_space(3))) void*)(smem + ldsoff), 16, 0, 0);
+}
+
+template <int N>
+__device__ __forceinline__ void gm_vmwait() {
+    asm volatile("s_waitcnt vmcnt(%0)" ::"n"(N) : "memory");
+}
+
+// compiler-only memory fence: keeps LDS reads/DMA issues from being
+// scheduled across a raw s_barrier (the builtin is convergent but not a
+// compiler memory barrier)
+__device__ __forceinline__ void gm_cfence() { asm volatile("" ::: "memory"); }
+
+// MW = waves per block (BM = 32*MW rows), NF = 16-col n-fragments (BN =
+// 16*NF). LDS ring: NBUF=3 buffers of [BM][BK] X + [2*NF fragments] W.
+template <int MW, int NF, bool SPLITK>
+__launch_bounds__(MW * WAVE_SIZE)
+__global__ void gemm_m256_kernel(
+    bf16* __restrict__ y,        // [M, N] (!SPLITK)
+    float* __restrict__ yw,      // [nsk, M, N] fp32 slabs (SPLITK)
+    const bf16* __restrict__ x,  // [M, K]
+    const bf16* __restrict__ w,  // fragment-major [K/32][N/16][64][8]
+    int M, int N, int K, int nsk) {
+    constexpr int BM = MW * 32;
+    constexpr int BN = NF * 16;
+    constexpr int NBUF = 3;
+    constexpr int XB = BM * GM_BK * 2;  // X tile bytes (row stride 128 B)
+    constexpr int WB = GM_BK * BN * 2;  // W tile bytes (2*NF 1-KiB frags)
+    constexpr int BUFB = XB + WB;
+    constexpr int XG_W = 4;             // X glds per wave (4 x 8 rows = 32)
+    constexpr int WG_W = (2 * NF + MW - 1) / MW;  // W glds per wave
+    constexpr int GPW = XG_W + WG_W;    // glds per wave per tile
+    // every wave must issue exactly GPW glds per tile or the counted
+    // vmcnt bookkeeping breaks
+    static_assert((2 * NF) % MW == 0, "W frag split uneven across waves");
+
+    __shared__ __attribute__((aligned(16))) char smem[NBUF * BUFB];
+
+    const int lane = threadIdx.x & (WAVE_SIZE - 1);
+    const int wave = threadIdx.x >> 6;
+    const int n0 = blockIdx.x * BN;
+    const int n16 = N / 16;
+
+    const int ktiles = K / GM_BK;
+    const int kt_per = SPLITK ? (ktiles + nsk - 1) / nsk : ktiles;
+    const int kt0 = SPLITK ? blockIdx.y * kt_per : 0;
+    const int ntiles = min(ktiles - kt0, kt_per) > 0 ? min(ktiles - kt0, kt_per) : 0;
+
+    // ---- staging addresses (loop-invariant parts) ----
+    // X: glds i covers LDS rows [32*wave + 8i, +8); lane l -> row
+    // base+l/8, 16-B chunk l%8, source chunk (l%8) ^ (row & 7) (the read
+    // XOR's inverse — same involution; row&7 is invariant under +8i).
+    // Rows are 128-B aligned in global (K % 64 == 0), so the in-row
+    // permutation stays within one cache line. Per-glds row clamp keeps
+    // tail blocks (M < BM) in bounds; clamped rows are skipped by the
+    // epilogue.
+    const int xrow = wave * 32 + (lane >> 3);  // + 8*i per glds
+    const int xchunk = (lane & 7) ^ (xrow & 7);
+    const bf16* xsrc[XG_W];
+#pragma unroll
+    for (int i = 0; i < XG_W; ++i) {
+        const int r = xrow + i * 8;
+        xsrc[i] = x + (size_t)(r < M ? r : 0) * K + xchunk * 8;
+    }
+    const int xdst0 = wave * 32 * 128;  // wave-uniform LDS base
+
+    // W: fragment fi = wave + j*MW (j < WG_W), global fragment index
+    // (2*kt + fi/NF)*n16 + n0/16 + fi%NF; source is 1 KiB contiguous.
+    int wfi[WG_W];
+#pragma unroll
+    for (int j = 0; j < WG_W; ++j) wfi[j] = wave + j * MW;
+
+    f32x4 acc[2][NF];
+#pragma unroll
+    for (int f = 0; f < 2; ++f)
+#pragma unroll
+        for (int n = 0; n < NF; ++n) acc[f][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+#define GM_STAGE(T)                                                            \
+    do {                                                                       \
+        const int kt__ = kt0 + (T);                                            \
+        char* buf__ = smem + ((T) % NBUF) * BUFB;                              \
+        _Pragma("unroll") for (int i = 0; i < XG_W; ++i)                       \
+            gm_glds16(xsrc[i] + (size_t)(kt__)*GM_BK, buf__,                   \
+                      xdst0 + i * 8 * 128);                                    \
+        _Pragma("unroll") for (int j = 0; j < WG_W; ++j) {                     \
+            const size_t gfi__ = (size_t)(2 * kt__ + wfi[j] / NF) * n16 +      \
+                                 n0 / 16 + wfi[j] % NF;                        \
+            gm_glds16(w + gfi__ * 512 + lane * 8, buf__, XB + wfi[j] * 1024);  \
+        }                                                                      \
+    } while (0)
+
+    // ds_read addresses (byte offsets within a buffer)
+    const int arow0 = wave * 32 + (lane & 15);          // f=0 row, f=1 adds 16
+    const int alk = lane >> 4;                          // k-chunk 0..3
+
+    if (ntiles > 0) {
+        GM_STAGE(0);
+        if (ntiles > 1) GM_STAGE(1);
+        for (int t = 0; t < ntiles; ++t) {
+            if (t + NBUF - 1 < ntiles) GM_STAGE(t + NBUF - 1);
+            // wait for tile t's DMA: allow the glds of the tiles beyond t
+            // to stay in flight across the barrier
+            const int ahead = min(ntiles - 1 - t, NBUF - 1);
+            if (ahead >= 2) gm_vmwait<2 * GPW>();
+            else if (ahead == 1) gm_vmwait<GPW>();
+            else gm_vmwait<0>();
+            __builtin_amdgcn_s_barrier();
+
+            const char* buf = smem + (t % NBUF) * BUFB;
+#pragma unroll
+            for (int ks = 0; ks < 2; ++ks) {
+                bf16x8 a[2], b[NF];
+#pragma unroll
+                for (int f = 0; f < 2; ++f) {
+                    const int row = arow0 + f * 16;
+                    const int chunk = (ks * 4 + alk) ^ (row & 7);
+                    a[f] = *(const __attribute__((address_space(3))) bf16x8*)(
+                        (const __attribute__((address_space(3))) char*)buf +
+                        row * 128 + chunk * 16);
+                }
+#pragma unroll
+                for (int n = 0; n < NF; ++n)
+                    b[n] = *(const __attribute__((address_space(3))) bf16x8*)(
+                        (const __attribute__((address_space(3))) char*)buf +
+                        XB + (ks * NF + n) * 1024 + lane * 16);
+#pragma unroll
+                for (int n = 0; n < NF; ++n)
+#pragma unroll
+                    for (int f = 0; f < 2; ++f)
+                        acc[f][n] = gm_mfma(a[f], b[n], acc[f][n]);
+            }
+            // all waves done reading buf t%NBUF before iter t+1 refills it
+            gm_cfence();
+            __builtin_amdgcn_s_barrier();
+            gm_cfence();
+        }
+    }
+#undef GM_STAGE
+
+    // ---- epilogue: C fragment row = 4*(lane>>4) + r, col = lane&15 ----
+    const int m_base = wave * 32;
+    if (SPLITK) {
+        float* slab = yw + (size_t)blockIdx.y * M * N;
+#pragma unroll
+        for (int f = 0; f < 2; ++f)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int row = m_base + f * 16 + (lane >> 4) * 4 + r;
+                if (row >= M) continue;
+#pragma unroll
+                for (int n = 0; n < NF; ++n)
+                    slab[(size_t)row * N + n0 + n * 16 + (lane & 15)] =
+                        acc[f][n][r];
+            }
+    } else {
+#pragma unroll
+        for (int f = 0; f < 2; ++f)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int row = m_base + f * 16 + (lane >> 4) * 4 + r;
+                if (row >= M) continue;
+#pragma unroll
+                for (int n = 0; n < NF; ++n)
+                    y[(size_t)row * N + n0 + n * 16 + (lane & 15)] =
+                        f2bf(acc[f][n][r]);
+            }
+    }
+}
+
+extern "C" hipError_t launch_gemm_reduce(void*, const float*, int64_t, int,
+                                         hipStream_t);  // gemm_skinny.hip
+
+// nf: 4 (BN=64) or 8 (BN=128). M <= 256; N % (16*nf) == 0; K % 64 == 0.
+extern "C" hipError_t launch_gemm_m256(
+    void* y, float* workspace, const void* x, const void* w, int M, int N,
+    int K, int nsk, int nf, hipStream_t stream) {
+    if (M <= 0 || M > 256) return hipErrorInvalidValue;
+    if (nf != 4 && nf != 8) return hipErrorInvalidValue;
+    if ((N % (16 * nf)) != 0 || (K % GM_BK) != 0) return hipErrorInvalidValue;
+    if (nsk < 1 || (nsk > 1 && workspace == nullptr)) return hipErrorInvalidValue;
+    int mw = 1;
+    while (mw * 32 < M) mw *= 2;  // 1,2,4,8
+    const int tiles = N / (16 * nf);
+    dim3 grid(tiles, nsk);
+    dim3 block(mw * WAVE_SIZE);
+#define GM_L2(MWV, NFV, SPLIT)                                                 \
+    gemm_m256_kernel<MWV, NFV, SPLIT><<<grid, block, 0, stream>>>(             \
+        (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K, nsk)
+#define GM_L1(MWV, NFV)                                                        \
+    do {                                                                       \
+        if (nsk > 1) GM_L2(MWV, NFV, true);                                    \
+        else GM_L2(MWV, NFV, false);                                           \
+    } while (0)
+#define GM_L0(MWV)                                                             \
+    do {                                                                       \
+        if (nf == 4) GM_L1(MWV, 4);                                            \
+        else GM_L1(MWV, 8);                                                    \
+    } while (0)
+    switch (mw) {
+        case 1: GM_L0(1); break;
+        case 2: GM_L0(2); break;
+        case 4: GM_L0(4); break;
+        default: GM_L0(8); break;
+    }
+#undef GM_L0
+#undef GM_L1
+#undef GM_L2
+    HIP_CHECK_LAST();
+    if (nsk > 1) {
+        const int64_t mn = (int64_t)M * N;  // N % 64 == 0 -> mn % 4 == 0
+        return launch_gemm_reduce(y, workspace, mn, nsk, stream);
+    }
+    return hipSuccess;
+}

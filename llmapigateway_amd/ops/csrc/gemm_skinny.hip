@@ -252,6 +252,17 @@ __global__ void gemm_skinny_reduce_kernel(
     *reinterpret_cast<uint2*>(y + i4) = packed;
 }
 
+// standalone split-K slab reduction (shared with gemm_m256.hip)
+extern "C" hipError_t launch_gemm_reduce(
+    void* y, const float* yw, int64_t mn, int nsk, hipStream_t stream) {
+    const int64_t thr = mn / 4;  // mn % 4 == 0 (N % 64 == 0)
+    const int tpb = 256;
+    gemm_skinny_reduce_kernel<<<dim3((thr + tpb - 1) / tpb), dim3(tpb), 0,
+                                stream>>>((bf16*)y, yw, mn, nsk);
+    HIP_CHECK_LAST();
+    return hipSuccess;
+}
+
 extern "C" hipError_t launch_gemm_skinny(
     void* y, float* workspace, const void* x, const void* w, int M, int N,
     int K, int nsk, int swz, hipStream_t stream) {

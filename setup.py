@@ -24,6 +24,7 @@ sources = [
     os.path.join(CSRC, "attention_prefill.hip"),
     os.path.join(CSRC, "sampling.hip"),
     os.path.join(CSRC, "gemm_skinny.hip"),
+    os.path.join(CSRC, "gemm_m256.hip"),
 ]
 
 setup(

@@ -1,5 +1,13 @@
-"""Time ops.linear (custom skinny-M streaming GEMM) vs F.linear (tuned
-library) for the decode projection shapes. Run on the GPU box."""
+"""Time the custom decode GEMMs vs F.linear (tuned library) for the decode
+projection shapes, cold-L3 (2 GB weight rotation). Run on the GPU box.
+
+Arms:
+  m256    — macro-tile LDS-staged kernel (csrc/gemm_m256.hip, fragment twin)
+  library — TunableOp-tuned hipBLASLt/rocBLAS
+  skinny  — round-1 streaming kernel (only with --skinny; historical)
+
+Usage: python tools/gemm_probe.py [M N K] [--nf 4|8] [--nsk N] [--skinny]
+"""
 
 import os
 import sys
@@ -26,17 +34,17 @@ SHAPES = [
     (64, 6144, 4096),
     (64, 4096, 14336),
     (64, 28672, 4096),
+    (16, 6144, 4096),
     (1, 4096, 14336),
 ]
 
 
 def bench(fn, iters=50):
-    # rotate weight copies so W is cold in cache. CAVEAT: the 256 MB
-    # rotation cap leaves >100 MB weights resident in the 256 MB Infinity
-    # Cache (L3) — such rows measure the L3-warm regime, NOT the engine's
-    # L3-cold weight cycle. Cross-check any dispatch decision against an
-    # in-engine bench (this bit us once: an L3-warm "parity" result
-    # regressed batch-256 decode 2x in situ).
+    # rotate weight copies so W is cold in cache. CAVEAT: the rotation cap
+    # must exceed the 256 MB Infinity Cache (L3) or rows measure the
+    # L3-warm regime, NOT the engine's L3-cold weight cycle (this bit us
+    # once: an L3-warm "parity" result regressed batch-256 decode 2x in
+    # situ).
     torch.cuda.synchronize()
     t0 = time.monotonic()
     for i in range(iters):
@@ -49,28 +57,43 @@ def main():
     assert torch.cuda.is_available()
     dev = "cuda:0"
     global SHAPES
-    if len(sys.argv) == 4:  # probe a single shape (for rocprofv3 --pmc runs)
-        SHAPES = [tuple(int(a) for a in sys.argv[1:4])]
-    print(f"{'M':>4} {'N':>7} {'K':>6} {'custom us':>10} {'swz us':>9} {'library us':>11} {'lib/swz':>8}")
+    args = [a for a in sys.argv[1:] if not a.startswith("--")]
+    flags = [a for a in sys.argv[1:] if a.startswith("--")]
+    nf_force = nsk_force = None
+    with_skinny = "--skinny" in flags
+    for f in flags:
+        if f.startswith("--nf"):
+            nf_force = int(f.split("=")[1]) if "=" in f else None
+        if f.startswith("--nsk"):
+            nsk_force = int(f.split("=")[1]) if "=" in f else None
+    if len(args) == 3:  # probe a single shape (for rocprofv3 --pmc runs)
+        SHAPES = [tuple(int(a) for a in args)]
+    hdr = f"{'M':>4} {'N':>7} {'K':>6} {'m256 us':>9} {'library us':>11} {'lib/m256':>9} {'TB/s':>6}"
+    if with_skinny:
+        hdr += f" {'skinny us':>10}"
+    print(hdr)
     for (M, N, K) in SHAPES:
         x = (torch.randn(M, K, device=dev) * 0.5).bfloat16()
-        ncopies = max(1, min(16, (2048 << 20) // (N * K * 2)))  # rotate >2 GB: defeat the 256 MB L3
+        ncopies = max(1, min(16, (2048 << 20) // (N * K * 2)))  # rotate >2 GB
         ws = [(torch.randn(N, K, device=dev) * 0.02).bfloat16() for _ in range(ncopies)]
-        # correctness spot check
         ref = x.float() @ ws[0].float().T
-        got = ops.linear(x, ws[0]).float()
+        wf = [ops.swizzle_weight_frag(w) for w in ws]
+        got = ops.gemm_m256(x, wf[0], nf=nf_force, nsk=nsk_force).float()
         err = (got - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
-        assert err < 0.02, f"skinny GEMM wrong for {(M,N,K)}: rel {err}"
-        wz = [ops.swizzle_weight(w) for w in ws]
-        got_s = ops.linear(x, ws[0], wz[0]).float()
-        err_s = (got_s - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
-        assert err_s < 0.02, f"swizzled GEMM wrong for {(M,N,K)}: rel {err_s}"
+        assert err < 0.02, f"gemm_m256 wrong for {(M,N,K)}: rel {err}"
         for _ in range(5):
-            ops.linear(x, ws[0]); F.linear(x, ws[0]); ops.linear(x, ws[0], wz[0])
-        t_c = bench(lambda i: ops.linear(x, ws[i % ncopies]))
-        t_s = bench(lambda i: ops.linear(x, ws[i % ncopies], wz[i % ncopies]))
+            ops.gemm_m256(x, wf[0], nf=nf_force, nsk=nsk_force); F.linear(x, ws[0])
+        t_m = bench(lambda i: ops.gemm_m256(x, wf[i % ncopies], nf=nf_force, nsk=nsk_force))
         t_l = bench(lambda i: F.linear(x, ws[i % ncopies]))
-        print(f"{M:>4} {N:>7} {K:>6} {t_c:>10.1f} {t_s:>9.1f} {t_l:>11.1f} {t_l / t_s:>8.2f}")
+        tbs = N * K * 2 / t_m / 1e6  # effective W-stream rate
+        line = f"{M:>4} {N:>7} {K:>6} {t_m:>9.1f} {t_l:>11.1f} {t_l / t_m:>9.2f} {tbs:>6.2f}"
+        if with_skinny:
+            got_s = ops.linear(x, ws[0]).float()
+            err_s = (got_s - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
+            assert err_s < 0.02, f"skinny GEMM wrong for {(M,N,K)}: rel {err_s}"
+            t_s = bench(lambda i: ops.linear(x, ws[i % ncopies]))
+            line += f" {t_s:>10.1f}"
+        print(line)
 
 
 if __name__ == "__main__":
