@@ -333,9 +333,13 @@ def _m256_nsk(N: int, K: int, nf: int) -> int:
 _SKINNY_SWZ_MAX_M = int(os.environ.get("SKINNY_GEMM_SWZ_MAX_M", "256"))
 
 
+# 0 = glds-staged, 1 = register-staged T14 (see gemm_m256.hip header)
+_M256_VARIANT = int(os.environ.get("LLMAPI_M256_VARIANT", "1"))
+
+
 def gemm_m256(
     x: torch.Tensor, w_frag: torch.Tensor, nf: Optional[int] = None,
-    nsk: Optional[int] = None,
+    nsk: Optional[int] = None, variant: Optional[int] = None,
 ) -> torch.Tensor:
     """y = x @ w.T with w pre-swizzled fragment-major (swizzle_weight_frag).
     The macro-tile LDS-staged decode GEMM (csrc/gemm_m256.hip); M <= 256."""
@@ -347,9 +351,11 @@ def gemm_m256(
         nf = 4
     if nsk is None:
         nsk = _m256_nsk(N, K, nf)
+    if variant is None:
+        variant = _M256_VARIANT
     y = torch.empty((M, N), dtype=torch.bfloat16, device=x.device)
     ws = _skinny_scratch(x.device, nsk * M * N) if nsk > 1 else None
-    _native().gemm_m256(y, x, w_frag, ws, nsk, nf)
+    _native().gemm_m256(y, x, w_frag, ws, nsk, nf, variant)
     return y
 
 

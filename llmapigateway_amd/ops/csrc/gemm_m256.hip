@@ -222,13 +222,188 @@ __global__ void gemm_m256_kernel(
     }
 }
 
+// ---------------------------------------------------------------------------
+// Register-staged variant (T14: global->reg loads issued one tile ahead,
+// ds_write AFTER the barrier, plain k-loop, 2 LDS buffers).
+//
+// Why it exists: the all-glds variant above is capped by the LDS-DMA
+// landing cadence (~25-40 GB/s per CU; MI355X_MICROARCH.md "ldsdma-fill"),
+// which binds exactly when one block must stream a lot of bytes — the
+// M=256 big-N shapes measured ~33 GB/s/CU (profiles/r02_gemm_m256_probe).
+// Ordinary vector loads ride the full per-CU load path (~122 GB/s), so
+// staging through registers + ds_write lifts the ceiling ~3x there. The
+// ds_write applies the same X bank-conflict XOR at write time (no source
+// permutation needed); W's fragment-major twin writes lane-linear.
+// Both ds_write_b128 patterns are conflict-free in the 8-lane
+// contiguous store groups.
+// ---------------------------------------------------------------------------
+
+template <int MW, int NF, bool SPLITK>
+__launch_bounds__(MW * WAVE_SIZE)
+__global__ void gemm_m256r_kernel(
+    bf16* __restrict__ y, float* __restrict__ yw,
+    const bf16* __restrict__ x,  // [M, K]
+    const bf16* __restrict__ w,  // fragment-major [K/32][N/16][64][8]
+    int M, int N, int K, int nsk) {
+    constexpr int BM = MW * 32;
+    constexpr int BN = NF * 16;
+    constexpr int XB = BM * GM_BK * 2;
+    constexpr int WB = GM_BK * BN * 2;
+    constexpr int BUFB = XB + WB;
+    constexpr int XG_W = 4;                       // 1-KiB x units per wave
+    constexpr int WG_W = (2 * NF + MW - 1) / MW;  // 1-KiB w units per wave
+    static_assert((2 * NF) % MW == 0, "W frag split uneven across waves");
+
+    __shared__ __attribute__((aligned(16))) char smem[2 * BUFB];
+
+    const int lane = threadIdx.x & (WAVE_SIZE - 1);
+    const int wave = threadIdx.x >> 6;
+    const int n0 = blockIdx.x * BN;
+    const int n16 = N / 16;
+
+    const int ktiles = K / GM_BK;
+    const int kt_per = SPLITK ? (ktiles + nsk - 1) / nsk : ktiles;
+    const int kt0 = SPLITK ? blockIdx.y * kt_per : 0;
+    const int ntiles = min(ktiles - kt0, kt_per) > 0 ? min(ktiles - kt0, kt_per) : 0;
+
+    // X staging: unit i = rows [32w+8i, +8); lane -> row base+l/8, chunk
+    // l%8. Loads are linear (full 128-B lines); the XOR goes on the
+    // ds_write address. Per-unit row clamp for M < BM tails.
+    const int xrow = wave * 32 + (lane >> 3);
+    const bf16* xsrc[XG_W];
+#pragma unroll
+    for (int i = 0; i < XG_W; ++i) {
+        const int r = xrow + i * 8;
+        xsrc[i] = x + (size_t)(r < M ? r : 0) * K + (lane & 7) * 8;
+    }
+    // ds_write byte offset (swizzled) per unit: row*128 + (c ^ row&7)*16
+    const int xwoff = (xrow & (BM - 1)) * 128 + (((lane & 7) ^ (xrow & 7)) << 4);
+
+    int wfi[WG_W];
+#pragma unroll
+    for (int j = 0; j < WG_W; ++j) wfi[j] = wave + j * MW;
+
+    f32x4 acc[2][NF];
+#pragma unroll
+    for (int f = 0; f < 2; ++f)
+#pragma unroll
+        for (int n = 0; n < NF; ++n) acc[f][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+    bf16x8 xr[XG_W], wr[WG_W];  // one in-flight staging tile
+
+#define GMR_LOAD(T)                                                            \
+    do {                                                                       \
+        const int kt__ = kt0 + (T);                                            \
+        _Pragma("unroll") for (int i = 0; i < XG_W; ++i) xr[i] =               \
+            *(const __attribute__((address_space(1))) bf16x8*)(                \
+                xsrc[i] + (size_t)(kt__)*GM_BK);                               \
+        _Pragma("unroll") for (int j = 0; j < WG_W; ++j) {                     \
+            const size_t gfi__ = (size_t)(2 * kt__ + wfi[j] / NF) * n16 +      \
+                                 n0 / 16 + wfi[j] % NF;                        \
+            wr[j] = *(const __attribute__((address_space(1))) bf16x8*)(        \
+                w + gfi__ * 512 + lane * 8);                                   \
+        }                                                                      \
+    } while (0)
+#define GMR_WRITE(T)                                                           \
+    do {                                                                       \
+        char* buf__ = smem + ((T)&1) * BUFB;                                   \
+        _Pragma("unroll") for (int i = 0; i < XG_W; ++i)                       \
+            *(__attribute__((address_space(3))) bf16x8*)(                      \
+                (__attribute__((address_space(3))) char*)buf__ + xwoff +       \
+                i * 8 * 128) = xr[i];                                          \
+        _Pragma("unroll") for (int j = 0; j < WG_W; ++j)                       \
+            *(__attribute__((address_space(3))) bf16x8*)(                      \
+                (__attribute__((address_space(3))) char*)buf__ + XB +          \
+                wfi[j] * 1024 + lane * 16) = wr[j];                            \
+    } while (0)
+
+    const int arow0 = wave * 32 + (lane & 15);
+    const int alk = lane >> 4;
+
+    if (ntiles > 0) {
+        GMR_LOAD(0);
+        GMR_WRITE(0);
+        if (ntiles > 1) GMR_LOAD(1);  // in flight during compute(0)
+        gm_cfence();
+        __builtin_amdgcn_s_barrier();
+        gm_cfence();
+        for (int t = 0; t < ntiles; ++t) {
+            if (t + 1 < ntiles) {
+                // regs hold tile t+1 (issued last iter): write after the
+                // barrier, then immediately re-issue for tile t+2 so the
+                // loads overlap compute(t)
+                GMR_WRITE(t + 1);
+                if (t + 2 < ntiles) GMR_LOAD(t + 2);
+            }
+            const char* buf = smem + (t & 1) * BUFB;
+#pragma unroll
+            for (int ks = 0; ks < 2; ++ks) {
+                bf16x8 a[2], b[NF];
+#pragma unroll
+                for (int f = 0; f < 2; ++f) {
+                    const int row = arow0 + f * 16;
+                    const int chunk = (ks * 4 + alk) ^ (row & 7);
+                    a[f] = *(const __attribute__((address_space(3))) bf16x8*)(
+                        (const __attribute__((address_space(3))) char*)buf +
+                        (row & (BM - 1)) * 128 + chunk * 16);
+                }
+#pragma unroll
+                for (int n = 0; n < NF; ++n)
+                    b[n] = *(const __attribute__((address_space(3))) bf16x8*)(
+                        (const __attribute__((address_space(3))) char*)buf +
+                        XB + (ks * NF + n) * 1024 + lane * 16);
+#pragma unroll
+                for (int n = 0; n < NF; ++n)
+#pragma unroll
+                    for (int f = 0; f < 2; ++f)
+                        acc[f][n] = gm_mfma(a[f], b[n], acc[f][n]);
+            }
+            gm_cfence();
+            __builtin_amdgcn_s_barrier();
+            gm_cfence();
+        }
+    }
+#undef GMR_LOAD
+#undef GMR_WRITE
+
+    const int m_base = wave * 32;
+    if (SPLITK) {
+        float* slab = yw + (size_t)blockIdx.y * M * N;
+#pragma unroll
+        for (int f = 0; f < 2; ++f)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int row = m_base + f * 16 + (lane >> 4) * 4 + r;
+                if (row >= M) continue;
+#pragma unroll
+                for (int n = 0; n < NF; ++n)
+                    slab[(size_t)row * N + n0 + n * 16 + (lane & 15)] =
+                        acc[f][n][r];
+            }
+    } else {
+#pragma unroll
+        for (int f = 0; f < 2; ++f)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int row = m_base + f * 16 + (lane >> 4) * 4 + r;
+                if (row >= M) continue;
+#pragma unroll
+                for (int n = 0; n < NF; ++n)
+                    y[(size_t)row * N + n0 + n * 16 + (lane & 15)] =
+                        f2bf(acc[f][n][r]);
+            }
+    }
+}
+
 extern "C" hipError_t launch_gemm_reduce(void*, const float*, int64_t, int,
                                          hipStream_t);  // gemm_skinny.hip
 
 // nf: 4 (BN=64) or 8 (BN=128). M <= 256; N % (16*nf) == 0; K % 64 == 0.
+// variant: 0 = glds-staged (DMA-cadence bound, best at small per-block
+// streams), 1 = register-staged T14 (full load-path rate).
 extern "C" hipError_t launch_gemm_m256(
     void* y, float* workspace, const void* x, const void* w, int M, int N,
-    int K, int nsk, int nf, hipStream_t stream) {
+    int K, int nsk, int nf, int variant, hipStream_t stream) {
     if (M <= 0 || M > 256) return hipErrorInvalidValue;
     if (nf != 4 && nf != 8) return hipErrorInvalidValue;
     if ((N % (16 * nf)) != 0 || (K % GM_BK) != 0) return hipErrorInvalidValue;
@@ -239,8 +414,16 @@ extern "C" hipError_t launch_gemm_m256(
     dim3 grid(tiles, nsk);
     dim3 block(mw * WAVE_SIZE);
 #define GM_L2(MWV, NFV, SPLIT)                                                 \
-    gemm_m256_kernel<MWV, NFV, SPLIT><<<grid, block, 0, stream>>>(             \
-        (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K, nsk)
+    do {                                                                       \
+        if (variant == 1)                                                      \
+            gemm_m256r_kernel<MWV, NFV, SPLIT><<<grid, block, 0, stream>>>(    \
+                (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,  \
+                nsk);                                                          \
+        else                                                                   \
+            gemm_m256_kernel<MWV, NFV, SPLIT><<<grid, block, 0, stream>>>(     \
+                (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,  \
+                nsk);                                                          \
+    } while (0)
 #define GM_L1(MWV, NFV)                                                        \
     do {                                                                       \
         if (nsk > 1) GM_L2(MWV, NFV, true);                                    \

@@ -291,11 +291,12 @@ def test_gemm_m256(M, N, K, nf):
     x = (torch.randn(M, K, device=DEV) * 0.5).bfloat16()
     w = (torch.randn(N, K, device=DEV) * 0.02).bfloat16()
     wf = ops.swizzle_weight_frag(w)
-    got = ops.gemm_m256(x, wf, nf=nf)
     ref = x.float() @ w.float().T
-    err = (got.float() - ref).abs().max().item()
-    scale = ref.abs().max().item() + 1e-3
-    assert err / scale < 0.02, f"max_err={err} scale={scale}"
+    for variant in (0, 1):  # glds-staged and register-staged forms
+        got = ops.gemm_m256(x, wf, nf=nf, variant=variant)
+        err = (got.float() - ref).abs().max().item()
+        scale = ref.abs().max().item() + 1e-3
+        assert err / scale < 0.02, f"v{variant} max_err={err} scale={scale}"
 
 
 def test_gemm_m256_via_linear_twin_dispatch():
