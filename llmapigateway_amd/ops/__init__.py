@@ -334,28 +334,41 @@ _SKINNY_SWZ_MAX_M = int(os.environ.get("SKINNY_GEMM_SWZ_MAX_M", "256"))
 
 
 # 0 = glds-staged, 1 = register-staged T14 (see gemm_m256.hip header)
-_M256_VARIANT = int(os.environ.get("LLMAPI_M256_VARIANT", "1"))
+_M256_VARIANT = int(os.environ.get("LLMAPI_M256_VARIANT", "0"))
 
 
 def gemm_m256(
     x: torch.Tensor, w_frag: torch.Tensor, nf: Optional[int] = None,
     nsk: Optional[int] = None, variant: Optional[int] = None,
+    pipe: Optional[int] = None,
 ) -> torch.Tensor:
     """y = x @ w.T with w pre-swizzled fragment-major (swizzle_weight_frag).
-    The macro-tile LDS-staged decode GEMM (csrc/gemm_m256.hip); M <= 256."""
+    The macro-tile LDS-staged decode GEMM (csrc/gemm_m256.hip); M <= 256.
+    pipe selects the DMA ring geometry for the glds variant:
+    0=(BK64,N3) 1=(BK64,N4,nf4) 2=(BK32,N4,nf8) 3=(BK32,N6,nf8)."""
     M, K = x.shape
     N = w_frag.shape[1] * 16
     if nf is None:
-        nf = 8 if N // 64 >= 448 else 4  # wide tiles only where blocks abound
+        if N % 128 == 0 and M > 64:
+            nf = 8  # deep BK32 ring needs BN=128
+        else:
+            nf = 8 if N // 64 >= 448 else 4
     if N % (16 * nf) != 0:
         nf = 4
     if nsk is None:
         nsk = _m256_nsk(N, K, nf)
     if variant is None:
         variant = _M256_VARIANT
+    if pipe is None:
+        pipe = 3 if (variant == 0 and nf == 8 and K % 32 == 0) else 0
+    if variant == 0:
+        if pipe in (2, 3) and nf != 8:
+            pipe = 0
+        if pipe == 1 and nf != 4:
+            pipe = 0
     y = torch.empty((M, N), dtype=torch.bfloat16, device=x.device)
     ws = _skinny_scratch(x.device, nsk * M * N) if nsk > 1 else None
-    _native().gemm_m256(y, x, w_frag, ws, nsk, nf, variant)
+    _native().gemm_m256(y, x, w_frag, ws, nsk, nf, variant, pipe)
     return y
 
 

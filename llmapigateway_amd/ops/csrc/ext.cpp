@@ -22,7 +22,7 @@ hipError_t launch_attention_decode(void*, const void*, const void*, const void*,
 hipError_t launch_attention_prefill(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, float, int, int, int, int64_t, int64_t, int64_t, const void*, const void*, const int*, const int*, int, int, hipStream_t);
 hipError_t launch_sample(int64_t*, const float*, const float*, const float*, float*, int*, int, int, hipStream_t);
 hipError_t launch_gemm_skinny(void*, float*, const void*, const void*, int, int, int, int, int, hipStream_t);
-hipError_t launch_gemm_m256(void*, float*, const void*, const void*, int, int, int, int, int, int, hipStream_t);
+hipError_t launch_gemm_m256(void*, float*, const void*, const void*, int, int, int, int, int, int, int, hipStream_t);
 }
 
 namespace {
@@ -266,7 +266,7 @@ void gemm_skinny(torch::Tensor y, torch::Tensor x, torch::Tensor w,
 
 void gemm_m256(torch::Tensor y, torch::Tensor x, torch::Tensor w_frag,
                c10::optional<torch::Tensor> workspace, int64_t nsk,
-               int64_t nf, int64_t variant) {
+               int64_t nf, int64_t variant, int64_t pipe) {
     check_bf16(x, "x");
     check_bf16(w_frag, "w_frag");
     check_bf16(y, "y");
@@ -288,7 +288,7 @@ void gemm_m256(torch::Tensor y, torch::Tensor x, torch::Tensor w_frag,
     }
     CHECK_HIP(launch_gemm_m256(y.data_ptr(), ws, x.data_ptr(),
                                w_frag.data_ptr(), M, N, K, (int)nsk, (int)nf,
-                               (int)variant, current_stream()));
+                               (int)variant, (int)pipe, current_stream()));
 }
 
 }  // namespace

@@ -65,9 +65,21 @@ __device__ __forceinline__ void gm_vmwait() {
 // compiler memory barrier)
 __device__ __forceinline__ void gm_cfence() { asm volatile("" ::: "memory"); }
 
+// counted wait for "ahead" tiles of GPW glds each still in flight
+// (compile-time unrolled: vmcnt immediates must be constants)
+template <int GPW, int NBUF, int A = 0>
+__device__ __forceinline__ void gm_wait_ahead(int ahead) {
+    if constexpr (A < NBUF) {
+        if (ahead == A) gm_vmwait<A * GPW>();
+        else gm_wait_ahead<GPW, NBUF, A + 1>(ahead);
+    }
+}
+
 // MW = waves per block (BM = 32*MW rows), NF = 16-col n-fragments (BN =
-// 16*NF). LDS ring: NBUF=3 buffers of [BM][BK] X + [2*NF fragments] W.
-template <int MW, int NF, bool SPLITK>
+// 16*NF), BK = staged k-depth per tile (32 or 64), NBUF = LDS ring depth
+// (NBUF-1 tiles of DMA in flight across the barriers — the knob that
+// covers HBM latency on the W stream).
+template <int MW, int NF, int BK, int NBUF, bool SPLITK>
 __launch_bounds__(MW * WAVE_SIZE)
 __global__ void gemm_m256_kernel(
     bf16* __restrict__ y,        // [M, N] (!SPLITK)
@@ -77,16 +89,18 @@ __global__ void gemm_m256_kernel(
     int M, int N, int K, int nsk) {
     constexpr int BM = MW * 32;
     constexpr int BN = NF * 16;
-    constexpr int NBUF = 3;
-    constexpr int XB = BM * GM_BK * 2;  // X tile bytes (row stride 128 B)
-    constexpr int WB = GM_BK * BN * 2;  // W tile bytes (2*NF 1-KiB frags)
+    constexpr int KS = BK / 32;          // MFMA k-steps per tile
+    constexpr int CPR = BK / 8;          // 16-B chunks per X row
+    constexpr int XB = BM * BK * 2;      // X tile bytes (row stride BK*2)
+    constexpr int WB = BK * BN * 2;      // W tile bytes (KS*NF 1-KiB frags)
     constexpr int BUFB = XB + WB;
-    constexpr int XG_W = 4;             // X glds per wave (4 x 8 rows = 32)
-    constexpr int WG_W = (2 * NF + MW - 1) / MW;  // W glds per wave
-    constexpr int GPW = XG_W + WG_W;    // glds per wave per tile
+    constexpr int XG_W = BK / 16;        // X glds per wave
+    constexpr int WG_W = KS * NF / MW;   // W glds per wave
+    constexpr int GPW = XG_W + WG_W;     // glds per wave per tile
     // every wave must issue exactly GPW glds per tile or the counted
     // vmcnt bookkeeping breaks
-    static_assert((2 * NF) % MW == 0, "W frag split uneven across waves");
+    static_assert((KS * NF) % MW == 0, "W frag split uneven across waves");
+    static_assert(NBUF * BUFB <= 160 * 1024, "LDS ring exceeds 160 KiB");
 
     __shared__ __attribute__((aligned(16))) char smem[NBUF * BUFB];
 
@@ -95,31 +109,31 @@ __global__ void gemm_m256_kernel(
     const int n0 = blockIdx.x * BN;
     const int n16 = N / 16;
 
-    const int ktiles = K / GM_BK;
+    const int ktiles = K / BK;
     const int kt_per = SPLITK ? (ktiles + nsk - 1) / nsk : ktiles;
     const int kt0 = SPLITK ? blockIdx.y * kt_per : 0;
     const int ntiles = min(ktiles - kt0, kt_per) > 0 ? min(ktiles - kt0, kt_per) : 0;
 
     // ---- staging addresses (loop-invariant parts) ----
-    // X: glds i covers LDS rows [32*wave + 8i, +8); lane l -> row
-    // base+l/8, 16-B chunk l%8, source chunk (l%8) ^ (row & 7) (the read
-    // XOR's inverse — same involution; row&7 is invariant under +8i).
-    // Rows are 128-B aligned in global (K % 64 == 0), so the in-row
-    // permutation stays within one cache line. Per-glds row clamp keeps
-    // tail blocks (M < BM) in bounds; clamped rows are skipped by the
-    // epilogue.
-    const int xrow = wave * 32 + (lane >> 3);  // + 8*i per glds
-    const int xchunk = (lane & 7) ^ (xrow & 7);
+    // X: glds i covers LDS rows [32*wave + i*(512/BK), +512/BK); lane l ->
+    // row base + l/CPR, 16-B chunk l%CPR, source chunk (l%CPR) ^ (row &
+    // (CPR-1)) (the read XOR's inverse — same involution; row&(CPR-1) is
+    // invariant under the unit stride). Rows are BK*2-aligned in global
+    // (K % 64 == 0), so the permutation stays within one/two cache lines.
+    // Per-glds row clamp keeps tail blocks (M < BM) in bounds; clamped
+    // rows are skipped by the epilogue.
+    const int xrow = wave * 32 + (lane / CPR);  // + (512/BK)*i per glds
+    const int xchunk = (lane % CPR) ^ (xrow & (CPR - 1));
     const bf16* xsrc[XG_W];
 #pragma unroll
     for (int i = 0; i < XG_W; ++i) {
-        const int r = xrow + i * 8;
+        const int r = xrow + i * (512 / BK);
         xsrc[i] = x + (size_t)(r < M ? r : 0) * K + xchunk * 8;
     }
-    const int xdst0 = wave * 32 * 128;  // wave-uniform LDS base
+    const int xdst0 = wave * 32 * (BK * 2);  // wave-uniform LDS base
 
     // W: fragment fi = wave + j*MW (j < WG_W), global fragment index
-    // (2*kt + fi/NF)*n16 + n0/16 + fi%NF; source is 1 KiB contiguous.
+    // (KS*kt + fi/NF)*n16 + n0/16 + fi%NF; source is 1 KiB contiguous.
     int wfi[WG_W];
 #pragma unroll
     for (int j = 0; j < WG_W; ++j) wfi[j] = wave + j * MW;
@@ -135,10 +149,10 @@ __global__ void gemm_m256_kernel(
         const int kt__ = kt0 + (T);                                            \
         char* buf__ = smem + ((T) % NBUF) * BUFB;                              \
         _Pragma("unroll") for (int i = 0; i < XG_W; ++i)                       \
-            gm_glds16(xsrc[i] + (size_t)(kt__)*GM_BK, buf__,                   \
-                      xdst0 + i * 8 * 128);                                    \
+            gm_glds16(xsrc[i] + (size_t)(kt__)*BK, buf__,                      \
+                      xdst0 + i * 1024);                                       \
         _Pragma("unroll") for (int j = 0; j < WG_W; ++j) {                     \
-            const size_t gfi__ = (size_t)(2 * kt__ + wfi[j] / NF) * n16 +      \
+            const size_t gfi__ = (size_t)(KS * kt__ + wfi[j] / NF) * n16 +     \
                                  n0 / 16 + wfi[j] % NF;                        \
             gm_glds16(w + gfi__ * 512 + lane * 8, buf__, XB + wfi[j] * 1024);  \
         }                                                                      \
@@ -149,29 +163,27 @@ __global__ void gemm_m256_kernel(
     const int alk = lane >> 4;                          // k-chunk 0..3
 
     if (ntiles > 0) {
-        GM_STAGE(0);
-        if (ntiles > 1) GM_STAGE(1);
+#pragma unroll
+        for (int p = 0; p < NBUF - 1; ++p)
+            if (p < ntiles) GM_STAGE(p);
         for (int t = 0; t < ntiles; ++t) {
             if (t + NBUF - 1 < ntiles) GM_STAGE(t + NBUF - 1);
             // wait for tile t's DMA: allow the glds of the tiles beyond t
             // to stay in flight across the barrier
-            const int ahead = min(ntiles - 1 - t, NBUF - 1);
-            if (ahead >= 2) gm_vmwait<2 * GPW>();
-            else if (ahead == 1) gm_vmwait<GPW>();
-            else gm_vmwait<0>();
+            gm_wait_ahead<GPW, NBUF>(min(ntiles - 1 - t, NBUF - 1));
             __builtin_amdgcn_s_barrier();
 
             const char* buf = smem + (t % NBUF) * BUFB;
 #pragma unroll
-            for (int ks = 0; ks < 2; ++ks) {
+            for (int ks = 0; ks < KS; ++ks) {
                 bf16x8 a[2], b[NF];
 #pragma unroll
                 for (int f = 0; f < 2; ++f) {
                     const int row = arow0 + f * 16;
-                    const int chunk = (ks * 4 + alk) ^ (row & 7);
+                    const int chunk = (ks * 4 + alk) ^ (row & (CPR - 1));
                     a[f] = *(const __attribute__((address_space(3))) bf16x8*)(
                         (const __attribute__((address_space(3))) char*)buf +
-                        row * 128 + chunk * 16);
+                        row * (BK * 2) + chunk * 16);
                 }
 #pragma unroll
                 for (int n = 0; n < NF; ++n)
@@ -399,15 +411,22 @@ extern "C" hipError_t launch_gemm_reduce(void*, const float*, int64_t, int,
                                          hipStream_t);  // gemm_skinny.hip
 
 // nf: 4 (BN=64) or 8 (BN=128). M <= 256; N % (16*nf) == 0; K % 64 == 0.
-// variant: 0 = glds-staged (DMA-cadence bound, best at small per-block
-// streams), 1 = register-staged T14 (full load-path rate).
+// variant: 0 = glds-staged (DMA pipeline, depth per `pipe`),
+//          1 = register-staged T14 (2 buffers, loads one tile ahead).
+// pipe (variant 0 only): 0=(BK64,NBUF3) 1=(BK64,NBUF4,nf4)
+//          2=(BK32,NBUF4,nf8) 3=(BK32,NBUF6,nf8) — deeper rings keep more
+//          HBM latency covered on the W stream.
 extern "C" hipError_t launch_gemm_m256(
     void* y, float* workspace, const void* x, const void* w, int M, int N,
-    int K, int nsk, int nf, int variant, hipStream_t stream) {
+    int K, int nsk, int nf, int variant, int pipe, hipStream_t stream) {
     if (M <= 0 || M > 256) return hipErrorInvalidValue;
     if (nf != 4 && nf != 8) return hipErrorInvalidValue;
     if ((N % (16 * nf)) != 0 || (K % GM_BK) != 0) return hipErrorInvalidValue;
     if (nsk < 1 || (nsk > 1 && workspace == nullptr)) return hipErrorInvalidValue;
+    if (variant == 0) {
+        if (pipe == 1 && nf != 4) return hipErrorInvalidValue;
+        if ((pipe == 2 || pipe == 3) && nf != 8) return hipErrorInvalidValue;
+    }
     int mw = 1;
     while (mw * 32 < M) mw *= 2;  // 1,2,4,8
     const int tiles = N / (16 * nf);
@@ -419,8 +438,24 @@ extern "C" hipError_t launch_gemm_m256(
             gemm_m256r_kernel<MWV, NFV, SPLIT><<<grid, block, 0, stream>>>(    \
                 (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,  \
                 nsk);                                                          \
+        else if (pipe == 1 && NFV == 4)                                        \
+            gemm_m256_kernel<MWV, 4, 64, 4, SPLIT><<<grid, block, 0,           \
+                                                     stream>>>(               \
+                (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,  \
+                nsk);                                                          \
+        else if (pipe == 2 && NFV == 8)                                        \
+            gemm_m256_kernel<MWV, 8, 32, 4, SPLIT><<<grid, block, 0,           \
+                                                     stream>>>(               \
+                (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,  \
+                nsk);                                                          \
+        else if (pipe == 3 && NFV == 8)                                        \
+            gemm_m256_kernel<MWV, 8, 32, 6, SPLIT><<<grid, block, 0,           \
+                                                     stream>>>(               \
+                (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,  \
+                nsk);                                                          \
         else                                                                   \
-            gemm_m256_kernel<MWV, NFV, SPLIT><<<grid, block, 0, stream>>>(     \
+            gemm_m256_kernel<MWV, NFV, 64, 3, SPLIT><<<grid, block, 0,         \
+                                                       stream>>>(             \
                 (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,  \
                 nsk);                                                          \
     } while (0)

@@ -68,8 +68,8 @@ def main():
             nsk_force = int(f.split("=")[1]) if "=" in f else None
     if len(args) == 3:  # probe a single shape (for rocprofv3 --pmc runs)
         SHAPES = [tuple(int(a) for a in args)]
-    hdr = (f"{'M':>4} {'N':>7} {'K':>6} {'glds us':>9} {'reg us':>8} "
-           f"{'library us':>11} {'lib/reg':>8} {'TB/s':>6}")
+    hdr = (f"{'M':>4} {'N':>7} {'K':>6} {'p0 us':>8} {'p3 us':>8} {'reg us':>8} "
+           f"{'library us':>11} {'lib/best':>9} {'TB/s':>6}")
     if with_skinny:
         hdr += f" {'skinny us':>10}"
     print(hdr)
@@ -79,19 +79,28 @@ def main():
         ws = [(torch.randn(N, K, device=dev) * 0.02).bfloat16() for _ in range(ncopies)]
         ref = x.float() @ ws[0].float().T
         wf = [ops.swizzle_weight_frag(w) for w in ws]
-        for v in (0, 1):
-            got = ops.gemm_m256(x, wf[0], nf=nf_force, nsk=nsk_force, variant=v).float()
+        checks = [dict(variant=0, pipe=0), dict(variant=1)]
+        if N % 128 == 0:
+            checks += [dict(variant=0, nf=8, pipe=2), dict(variant=0, nf=8, pipe=3)]
+        for kw in checks:
+            kw.setdefault("nf", nf_force)
+            got = ops.gemm_m256(x, wf[0], nsk=nsk_force, **kw).float()
             err = (got - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
-            assert err < 0.02, f"gemm_m256 v{v} wrong for {(M,N,K)}: rel {err}"
+            assert err < 0.02, f"gemm_m256 {kw} wrong for {(M,N,K)}: rel {err}"
         for _ in range(5):
             ops.gemm_m256(x, wf[0], nf=nf_force, nsk=nsk_force); F.linear(x, ws[0])
-        t_g = bench(lambda i: ops.gemm_m256(x, wf[i % ncopies], nf=nf_force, nsk=nsk_force, variant=0))
+        t_g = bench(lambda i: ops.gemm_m256(x, wf[i % ncopies], nf=nf_force, nsk=nsk_force, variant=0, pipe=0))
+        if N % 128 == 0:
+            t_p = bench(lambda i: ops.gemm_m256(x, wf[i % ncopies], nf=8, nsk=nsk_force, variant=0, pipe=3))
+        else:
+            t_p = float("nan")
         t_r = bench(lambda i: ops.gemm_m256(x, wf[i % ncopies], nf=nf_force, nsk=nsk_force, variant=1))
         t_l = bench(lambda i: F.linear(x, ws[i % ncopies]))
-        t_m = min(t_g, t_r)
+        cand = [t for t in (t_g, t_p, t_r) if t == t]
+        t_m = min(cand)
         tbs = N * K * 2 / t_m / 1e6  # effective W-stream rate
-        line = (f"{M:>4} {N:>7} {K:>6} {t_g:>9.1f} {t_r:>8.1f} {t_l:>11.1f} "
-                f"{t_l / t_r:>8.2f} {tbs:>6.2f}")
+        line = (f"{M:>4} {N:>7} {K:>6} {t_g:>8.1f} {t_p:>8.1f} {t_r:>8.1f} {t_l:>11.1f} "
+                f"{t_l / t_m:>9.2f} {tbs:>6.2f}")
         if with_skinny:
             got_s = ops.linear(x, ws[0]).float()
             err_s = (got_s - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
