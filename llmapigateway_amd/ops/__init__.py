@@ -364,7 +364,7 @@ def gemm_m256(
     if variant == 0:
         if pipe in (2, 3) and nf != 8:
             pipe = 0
-        if pipe == 1 and nf != 4:
+        if pipe in (1, 4, 5) and nf != 4:
             pipe = 0
     y = torch.empty((M, N), dtype=torch.bfloat16, device=x.device)
     ws = _skinny_scratch(x.device, nsk * M * N) if nsk > 1 else None

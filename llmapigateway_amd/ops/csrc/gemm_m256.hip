@@ -95,11 +95,14 @@ __global__ void gemm_m256_kernel(
     constexpr int WB = BK * BN * 2;      // W tile bytes (KS*NF 1-KiB frags)
     constexpr int BUFB = XB + WB;
     constexpr int XG_W = BK / 16;        // X glds per wave
-    constexpr int WG_W = KS * NF / MW;   // W glds per wave
-    constexpr int GPW = XG_W + WG_W;     // glds per wave per tile
-    // every wave must issue exactly GPW glds per tile or the counted
-    // vmcnt bookkeeping breaks
-    static_assert((KS * NF) % MW == 0, "W frag split uneven across waves");
+    constexpr int WU = KS * NF;          // W 1-KiB units per tile
+    // W units round-robin over waves; when MW does not divide WU the low
+    // waves carry one extra unit and the counted vmcnt waits use each
+    // wave's own glds count (the wait is a per-wave counter).
+    constexpr int WG_HI = (WU + MW - 1) / MW;  // units on waves < WU%MW (or all)
+    constexpr int WG_LO = WU / MW;
+    constexpr int GPW_HI = XG_W + WG_HI;
+    constexpr int GPW_LO = XG_W + WG_LO;
     static_assert(NBUF * BUFB <= 160 * 1024, "LDS ring exceeds 160 KiB");
 
     __shared__ __attribute__((aligned(16))) char smem[NBUF * BUFB];
@@ -132,11 +135,13 @@ __global__ void gemm_m256_kernel(
     }
     const int xdst0 = wave * 32 * (BK * 2);  // wave-uniform LDS base
 
-    // W: fragment fi = wave + j*MW (j < WG_W), global fragment index
-    // (KS*kt + fi/NF)*n16 + n0/16 + fi%NF; source is 1 KiB contiguous.
-    int wfi[WG_W];
+    // W: fragment fi = wave + j*MW (j < WG_HI, fi < WU), global fragment
+    // index (KS*kt + fi/NF)*n16 + n0/16 + fi%NF; source is 1 KiB
+    // contiguous.
+    int wfi[WG_HI];
 #pragma unroll
-    for (int j = 0; j < WG_W; ++j) wfi[j] = wave + j * MW;
+    for (int j = 0; j < WG_HI; ++j) wfi[j] = wave + j * MW;
+    const bool w_extra = wave < (WU % MW == 0 ? MW : WU % MW);
 
     f32x4 acc[2][NF];
 #pragma unroll
@@ -151,10 +156,13 @@ __global__ void gemm_m256_kernel(
         _Pragma("unroll") for (int i = 0; i < XG_W; ++i)                       \
             gm_glds16(xsrc[i] + (size_t)(kt__)*BK, buf__,                      \
                       xdst0 + i * 1024);                                       \
-        _Pragma("unroll") for (int j = 0; j < WG_W; ++j) {                     \
-            const size_t gfi__ = (size_t)(KS * kt__ + wfi[j] / NF) * n16 +     \
-                                 n0 / 16 + wfi[j] % NF;                        \
-            gm_glds16(w + gfi__ * 512 + lane * 8, buf__, XB + wfi[j] * 1024);  \
+        _Pragma("unroll") for (int j = 0; j < WG_HI; ++j) {                    \
+            if (wfi[j] < WU) {                                                 \
+                const size_t gfi__ = (size_t)(KS * kt__ + wfi[j] / NF) * n16 + \
+                                     n0 / 16 + wfi[j] % NF;                    \
+                gm_glds16(w + gfi__ * 512 + lane * 8, buf__,                   \
+                          XB + wfi[j] * 1024);                                 \
+            }                                                                  \
         }                                                                      \
     } while (0)
 
@@ -169,8 +177,10 @@ __global__ void gemm_m256_kernel(
         for (int t = 0; t < ntiles; ++t) {
             if (t + NBUF - 1 < ntiles) GM_STAGE(t + NBUF - 1);
             // wait for tile t's DMA: allow the glds of the tiles beyond t
-            // to stay in flight across the barrier
-            gm_wait_ahead<GPW, NBUF>(min(ntiles - 1 - t, NBUF - 1));
+            // to stay in flight across the barrier (per-wave counts)
+            const int ahead = min(ntiles - 1 - t, NBUF - 1);
+            if (w_extra) gm_wait_ahead<GPW_HI, NBUF>(ahead);
+            else gm_wait_ahead<GPW_LO, NBUF>(ahead);
             __builtin_amdgcn_s_barrier();
 
             const char* buf = smem + (t % NBUF) * BUFB;
@@ -415,7 +425,9 @@ extern "C" hipError_t launch_gemm_reduce(void*, const float*, int64_t, int,
 //          1 = register-staged T14 (2 buffers, loads one tile ahead).
 // pipe (variant 0 only): 0=(BK64,NBUF3) 1=(BK64,NBUF4,nf4)
 //          2=(BK32,NBUF4,nf8) 3=(BK32,NBUF6,nf8) — deeper rings keep more
-//          HBM latency covered on the W stream.
+//          HBM latency covered on the W stream; 4=(BK64,NBUF2,nf4) and
+//          5=(BK32,NBUF3,nf4) fit 2 blocks/CU so a second block fills the
+//          barrier-lockstep wait gaps.
 extern "C" hipError_t launch_gemm_m256(
     void* y, float* workspace, const void* x, const void* w, int M, int N,
     int K, int nsk, int nf, int variant, int pipe, hipStream_t stream) {
@@ -424,7 +436,8 @@ extern "C" hipError_t launch_gemm_m256(
     if ((N % (16 * nf)) != 0 || (K % GM_BK) != 0) return hipErrorInvalidValue;
     if (nsk < 1 || (nsk > 1 && workspace == nullptr)) return hipErrorInvalidValue;
     if (variant == 0) {
-        if (pipe == 1 && nf != 4) return hipErrorInvalidValue;
+        if ((pipe == 1 || pipe == 4 || pipe == 5) && nf != 4)
+            return hipErrorInvalidValue;
         if ((pipe == 2 || pipe == 3) && nf != 8) return hipErrorInvalidValue;
     }
     int mw = 1;
@@ -450,6 +463,16 @@ extern "C" hipError_t launch_gemm_m256(
                 nsk);                                                          \
         else if (pipe == 3 && NFV == 8)                                        \
             gemm_m256_kernel<MWV, 8, 32, 6, SPLIT><<<grid, block, 0,           \
+                                                     stream>>>(               \
+                (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,  \
+                nsk);                                                          \
+        else if (pipe == 4 && NFV == 4)                                        \
+            gemm_m256_kernel<MWV, 4, 64, 2, SPLIT><<<grid, block, 0,           \
+                                                     stream>>>(               \
+                (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,  \
+                nsk);                                                          \
+        else if (pipe == 5 && NFV == 4)                                        \
+            gemm_m256_kernel<MWV, 4, 32, 3, SPLIT><<<grid, block, 0,           \
                                                      stream>>>(               \
                 (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,  \
                 nsk);                                                          \
