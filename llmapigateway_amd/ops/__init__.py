@@ -314,6 +314,26 @@ def swizzle_weight_frag(w: torch.Tensor) -> torch.Tensor:
     )
 
 
+def _m256_config(M: int, N: int, K: int) -> Optional[dict]:
+    """Measured dispatch table for the decode projections (cold-L3 sweep,
+    profiles/r02_gemm_m256_sweep.md). Returns gemm_m256 kwargs, or None
+    when the tuned library wins the shape (gate_up N=28672, lm_head)."""
+    if M > 256:
+        return None
+    if N >= 16384:
+        # W-stream-heavy wide shapes: custom best (v1 nf8) ties the
+        # library at 0.96x — not worth the dispatch
+        return None
+    if K > 8192 and N % 128 == 0:
+        return {"nf": 8, "nsk": 8, "variant": 0, "pipe": 0}  # down 1.34x
+    if N <= 8192:
+        # o-proj class: BK64/NBUF2 2-blocks/CU + split-K to ~256 blocks
+        tiles = N // 64
+        nsk = max(1, min(-(-256 // tiles), (K // 64) // 2, 8))
+        return {"nf": 4, "nsk": nsk, "variant": 0, "pipe": 4}  # o 1.30x
+    return {"nf": 4, "variant": 0, "pipe": 0}
+
+
 def _m256_nsk(N: int, K: int, nf: int) -> int:
     """Split-K factor for gemm_m256: fill the 256 CUs (one 8-wave block
     per CU at the 120 KB LDS ring) when the column-tile count alone
@@ -396,10 +416,14 @@ def linear(
         if (
             w_swz is not None
             and w_swz.dim() == 4
-            and M <= 256
+            and 8 < M <= 256
             and K % 64 == 0
         ):
-            return gemm_m256(x, w_swz)
+            cfg = _m256_config(M, N, K)
+            if cfg is not None:
+                return gemm_m256(x, w_swz, **cfg)
+            # fall through to library (measured faster for this shape)
+            return torch.nn.functional.linear(x, w)
         if w_swz is not None and w_swz.dim() == 3 and M <= _SKINNY_SWZ_MAX_M:
             # legacy k-major twin -> streaming skinny kernel
             y = torch.empty((M, N), dtype=torch.bfloat16, device=x.device)
