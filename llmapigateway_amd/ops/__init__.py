@@ -320,18 +320,17 @@ def _m256_config(M: int, N: int, K: int) -> Optional[dict]:
     when the tuned library wins the shape (gate_up N=28672, lm_head)."""
     if M > 256:
         return None
-    if N >= 16384:
-        # W-stream-heavy wide shapes: custom best (v1 nf8) ties the
-        # library at 0.96x — not worth the dispatch
+    if N > 4096:
+        # library wins the wide shapes: gate_up N=28672 (custom best
+        # 73.3 vs 70.3 us) and qkv N=6144 (36.4 vs ~28 us — hipBLASLt's
+        # MT112x256 kernel is strong exactly there)
         return None
     if K > 8192 and N % 128 == 0:
         return {"nf": 8, "nsk": 8, "variant": 0, "pipe": 0}  # down 1.34x
-    if N <= 8192:
-        # o-proj class: BK64/NBUF2 2-blocks/CU + split-K to ~256 blocks
-        tiles = N // 64
-        nsk = max(1, min(-(-256 // tiles), (K // 64) // 2, 8))
-        return {"nf": 4, "nsk": nsk, "variant": 0, "pipe": 4}  # o 1.30x
-    return {"nf": 4, "variant": 0, "pipe": 0}
+    # o-proj class (N<=4096, K<=8192): BK64/NBUF2 2-blocks/CU + split-K
+    tiles = N // 64
+    nsk = max(1, min(-(-256 // tiles), (K // 64) // 2, 8))
+    return {"nf": 4, "nsk": nsk, "variant": 0, "pipe": 4}  # o 1.30x
 
 
 def _m256_nsk(N: int, K: int, nf: int) -> int:
