@@ -1,12 +1,20 @@
-"""Serving benchmark: req/sec + p50 TTFT for Llama-3-8B bf16 (BASELINE.json).
+"""Serving benchmark: req/sec + p50 TTFT at fixed QPS for Llama-3-8B bf16
+(BASELINE.json headline metric).
 
 One rank per GPU (weak scaling: each rank runs an identical engine replica —
-the gateway's rotation mode, BASELINE configs[3]). A *step* is one wave of
-``--batch`` synthetic requests (random-init weights, random token prompts of
---prompt-len, greedy decode of --gen-tokens) driven to completion through
-the continuous-batching engine. Timed region: K steps bracketed by a
-barrier + torch.cuda.synchronize on both sides; value is the whole-job
-req/sec aggregated over all ranks (max rank time).
+the gateway's rotation mode, BASELINE configs[3]). Default mode is the
+OPEN-LOOP fixed-QPS benchmark the baseline names: each rank schedules
+``--steps x --batch`` synthetic requests (random-init weights, random token
+prompts) as a Poisson arrival process at ``--qps`` requests/s (``auto`` =
+1.1x a capacity estimate from the warmup waves, so the engine is saturated
+and the value measures capacity) and drives them through the
+continuous-batching engine as they arrive; TTFT is measured from each
+request's SCHEDULED arrival (queueing included). A *step* is one
+batch-sized cohort of arrivals; the timed region is the whole continuous
+run bracketed by a barrier + torch.cuda.synchronize on both sides; value
+is the whole-job achieved req/s aggregated over all ranks (max rank
+time). ``--mode wave`` keeps the round-1 closed-loop wave benchmark for
+comparability.
 
 Launch (the driver does this for N>1):
   python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
@@ -18,6 +26,7 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import random
 import statistics
 import sys
 import time
@@ -39,6 +48,16 @@ def parse_args():
     p.add_argument("--prompt-len", type=int, default=128)
     p.add_argument("--gen-tokens", type=int, default=64)
     p.add_argument("--kv-blocks", type=int, default=None)
+    p.add_argument(
+        "--mode", choices=["qps", "wave"], default="qps",
+        help="qps = open-loop Poisson arrivals at --qps (the BASELINE "
+        "metric); wave = round-1 closed-loop waves",
+    )
+    p.add_argument(
+        "--qps", default="auto",
+        help="offered request rate per GPU for --mode qps; 'auto' = 1.1x "
+        "capacity estimated from the warmup waves",
+    )
     return p.parse_args()
 
 
@@ -66,6 +85,48 @@ def run_wave(engine: LLMEngine, prompts, gen_tokens: int):
     assert all(r.state == "finished" for r in reqs)
     assert all(len(r.out_ids) == gen_tokens for r in reqs)
     return ttfts
+
+
+def run_open_loop(engine: LLMEngine, prompts, gen_tokens: int, qps: float, seed: int):
+    """Drive len(prompts) requests as a Poisson process at `qps` req/s.
+
+    Returns (ttfts_ms, elapsed_s). TTFT is measured from each request's
+    scheduled arrival time; requests arriving while the engine is busy
+    queue exactly as gateway traffic would."""
+    rng = random.Random(seed)
+    t = 0.0
+    arrivals = []
+    for _ in prompts:
+        arrivals.append(t)
+        t += rng.expovariate(qps)
+    reqs = [
+        EngineRequest(p, SamplingParams(max_tokens=gen_tokens, ignore_eos=True))
+        for p in prompts
+    ]
+    t0 = time.monotonic()
+    nxt = 0
+    n = len(reqs)
+    while True:
+        now = time.monotonic() - t0
+        while nxt < n and arrivals[nxt] <= now:
+            engine.add_request(reqs[nxt])
+            nxt += 1
+        if engine.has_work():
+            engine.step()
+        elif nxt < n:
+            time.sleep(min(0.002, max(0.0, arrivals[nxt] - (time.monotonic() - t0))))
+        else:
+            break
+    if engine.device.type == "cuda":
+        torch.cuda.synchronize(engine.device)
+    elapsed = time.monotonic() - t0
+    ttfts = [
+        (r.first_token_time - t0 - a) * 1000.0
+        for r, a in zip(reqs, arrivals)
+        if r.first_token_time
+    ]
+    assert all(r.state == "finished" for r in reqs)
+    return ttfts, elapsed
 
 
 def main():
@@ -106,18 +167,46 @@ def main():
         if have_cuda:
             torch.cuda.synchronize(engine.device)
 
-    # warmup
+    # warmup (closed-loop waves; also the capacity estimate for --qps auto)
+    wave_rates = []
     for w in range(args.warmup):
+        tw = time.monotonic()
         run_wave(engine, make_prompts(args.batch, args.prompt_len, vocab, 1000 + w), args.gen_tokens)
+        wave_rates.append(args.batch / (time.monotonic() - tw))
+
+    qps = None
+    if args.mode == "qps":
+        if args.qps == "auto":
+            cap = max(wave_rates) if wave_rates else 50.0
+            qps = cap * 1.1  # saturate: achieved req/s measures capacity
+        else:
+            qps = float(args.qps)
+        # every rank must offer the same load: agree on rank 0's value
+        if dist is not None:
+            tq = torch.tensor([qps], dtype=torch.float64)
+            if have_cuda:
+                tq = tq.to(engine.device)
+            dist.broadcast(tq, src=0)
+            qps = float(tq.item())
 
     barrier_sync()
     t_start = time.monotonic()
     ttfts = []
-    for s in range(args.steps):
-        ttfts += run_wave(
-            engine, make_prompts(args.batch, args.prompt_len, vocab, 2000 + s + rank * 7919),
-            args.gen_tokens,
+    if args.mode == "qps":
+        prompts = []
+        for s in range(args.steps):
+            prompts += make_prompts(
+                args.batch, args.prompt_len, vocab, 2000 + s + rank * 7919
+            )
+        ttfts, _ = run_open_loop(
+            engine, prompts, args.gen_tokens, qps, seed=97 + rank
         )
+    else:
+        for s in range(args.steps):
+            ttfts += run_wave(
+                engine, make_prompts(args.batch, args.prompt_len, vocab, 2000 + s + rank * 7919),
+                args.gen_tokens,
+            )
     barrier_sync()
     elapsed = time.monotonic() - t_start
 
@@ -132,8 +221,9 @@ def main():
     if rank == 0:
         total_reqs = args.batch * args.steps * world_size
         req_per_sec = total_reqs / elapsed
+        ttfts.sort()
         result = {
-            "metric": "req/sec",
+            "metric": "req/sec @ fixed QPS" if args.mode == "qps" else "req/sec",
             "value": round(req_per_sec, 3),
             "unit": "req/s",
             "n_gpus": world_size,
@@ -152,7 +242,10 @@ def main():
                 "prompt_len": args.prompt_len,
                 "gen_tokens": args.gen_tokens,
                 "parallelism": f"dp{world_size}",
+                "mode": "open-loop-poisson" if args.mode == "qps" else "closed-loop-waves",
+                "qps_offered_per_gpu": round(qps, 1) if qps else None,
                 "p50_ttft_ms": round(statistics.median(ttfts), 2) if ttfts else None,
+                "p95_ttft_ms": round(ttfts[int(len(ttfts) * 0.95)], 2) if ttfts else None,
                 "tokens_per_sec": round(
                     total_reqs * (args.prompt_len + args.gen_tokens) / elapsed, 1
                 ),
