@@ -694,28 +694,28 @@ class LLMEngine:
         return self._sample_dev(logits, reqs).tolist()
 
     def _apply_topk_topp(self, logits: torch.Tensor, reqs: List[EngineRequest]) -> torch.Tensor:
-        needs = [
-            i
-            for i, r in enumerate(reqs)
-            if r.params.temperature > 0 and (r.params.top_k > 0 or r.params.top_p < 1.0)
-        ]
-        if not needs:
+        """In-place top-k/top-p filtering over the whole batch: the GPU
+        path is the histogram-threshold kernel (ops/csrc/sampling.hip —
+        no full-vocab sort, no per-row host loop); CPU uses the batched
+        torch reference. Rows that need no filtering are untouched."""
+        any_filter = False
+        topp = []
+        topk = []
+        for r in reqs:
+            active = r.params.temperature > 0 and (
+                r.params.top_k > 0 or r.params.top_p < 1.0
+            )
+            any_filter = any_filter or active
+            topp.append(r.params.top_p if active else 1.0)
+            topk.append(r.params.top_k if active else 0)
+        if not any_filter:
             return logits
-        sub = logits[needs].float()
-        sorted_logits, sorted_idx = torch.sort(sub, descending=True, dim=-1)
-        mask = torch.zeros_like(sub, dtype=torch.bool)
-        for row, i in enumerate(needs):
-            p = reqs[i].params
-            if p.top_k > 0:
-                mask[row].scatter_(0, sorted_idx[row, p.top_k :], True)
-            if p.top_p < 1.0:
-                probs = torch.softmax(sorted_logits[row], dim=-1)
-                cdf = probs.cumsum(dim=-1)
-                cut = torch.searchsorted(cdf, torch.tensor(p.top_p, device=cdf.device)) + 1
-                mask[row].scatter_(0, sorted_idx[row, cut:], True)
-        sub.masked_fill_(mask, float("-inf"))
-        logits[needs] = sub.to(logits.dtype)
-        return logits
+        dev = logits.device
+        return ops.topk_topp_filter(
+            logits,
+            torch.tensor(topp, dtype=torch.float32, device=dev),
+            torch.tensor(topk, dtype=torch.int32, device=dev),
+        )
 
     # ---- delivery / lifecycle (call with lock held) ----
     def _deliver(self, reqs: List[EngineRequest], tokens: List[int]) -> None:

@@ -439,6 +439,19 @@ def linear(
     return torch.nn.functional.linear(x, w)
 
 
+def topk_topp_filter(
+    logits: torch.Tensor, topp: torch.Tensor, topk: torch.Tensor
+) -> torch.Tensor:
+    """Mask logits outside the per-row top-k / top-p set to -inf, in
+    place, and return logits. GPU: histogram-threshold kernel
+    (csrc/sampling.hip, no sort, no host loop); CPU: batched torch sort
+    (reference semantics)."""
+    if logits.is_cuda:
+        _native().topk_topp_filter(logits, topp, topk)
+        return logits
+    return reference.topk_topp_filter(logits, topp, topk)
+
+
 def sample(
     logits: torch.Tensor,
     temperature: torch.Tensor,
@@ -467,5 +480,6 @@ __all__ = [
     "attention_prefill",
     "attention_decode",
     "sample",
+    "topk_topp_filter",
     "reference",
 ]

@@ -21,6 +21,7 @@ hipError_t launch_rope_kv(void*, void*, const void*, void*, void*, const int64_t
 hipError_t launch_attention_decode(void*, const void*, const void*, const void*, const int*, const int*, float, int, int, int, int, int, int, int64_t, float*, float*, int, hipStream_t);
 hipError_t launch_attention_prefill(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, float, int, int, int, int64_t, int64_t, int64_t, const void*, const void*, const int*, const int*, int, int, hipStream_t);
 hipError_t launch_sample(int64_t*, const float*, const float*, const float*, float*, int*, int, int, hipStream_t);
+hipError_t launch_topk_topp_filter(float*, const float*, const int*, int, int, hipStream_t);
 hipError_t launch_gemm_skinny(void*, float*, const void*, const void*, int, int, int, int, int, hipStream_t);
 hipError_t launch_gemm_m256(void*, float*, const void*, const void*, int, int, int, int, int, int, int, hipStream_t);
 }
@@ -234,6 +235,19 @@ void sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temperature,
                             B, logits.size(1), current_stream()));
 }
 
+void topk_topp_filter(torch::Tensor logits, torch::Tensor topp, torch::Tensor topk) {
+    TORCH_CHECK(logits.is_cuda() && logits.scalar_type() == torch::kFloat32 &&
+                logits.is_contiguous() && logits.dim() == 2);
+    TORCH_CHECK(topp.scalar_type() == torch::kFloat32 && topp.is_contiguous());
+    TORCH_CHECK(topk.scalar_type() == torch::kInt32 && topk.is_contiguous());
+    const int B = logits.size(0), V = logits.size(1);
+    TORCH_CHECK(topp.numel() == B && topk.numel() == B);
+    CHECK_HIP(launch_topk_topp_filter(logits.data_ptr<float>(),
+                                      topp.data_ptr<float>(),
+                                      topk.data_ptr<int>(), B, V,
+                                      current_stream()));
+}
+
 void gemm_skinny(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                  c10::optional<torch::Tensor> workspace, int64_t nsk,
                  bool swizzled) {
@@ -303,6 +317,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("attention_decode", &attention_decode, "paged decode attention");
     m.def("attention_prefill", &attention_prefill, "varlen causal prefill attention");
     m.def("sample", &sample, "greedy / gumbel-max sampling");
+    m.def("topk_topp_filter", &topk_topp_filter,
+          "in-place top-k/top-p logit filtering (histogram threshold)");
     m.def("gemm_skinny", &gemm_skinny,
           "skinny-M weight-streaming GEMM y = x @ w.T (decode projections)");
     m.def("gemm_m256", &gemm_m256,

@@ -200,3 +200,31 @@ def sample(
     gumbel = -torch.log(inner)
     scored = torch.where(greedy, lf, lf / temps.clamp_min(1e-6) + gumbel)
     return scored.argmax(dim=-1)
+
+
+def topk_topp_filter(
+    logits: torch.Tensor, topp: torch.Tensor, topk: torch.Tensor
+) -> torch.Tensor:
+    """Batched top-k / top-p filtering: mask entries outside the kept set
+    to -inf, in place, and return logits. Semantics: top-p keeps the
+    smallest prefix of the descending-sorted row whose softmax mass
+    reaches p (the crossing token included); top-k keeps rank < k; both
+    given -> intersection. Rows with topk==0 and topp>=1 are untouched."""
+    B, V = logits.shape
+    active = (topk > 0) | (topp < 1.0)
+    if not bool(active.any()):
+        return logits
+    lf = logits.float()
+    sorted_logits, sorted_idx = torch.sort(lf, descending=True, dim=-1)
+    pos = torch.arange(V, device=logits.device).unsqueeze(0)  # [1, V]
+    k = torch.where(topk > 0, topk, torch.full_like(topk, V)).unsqueeze(1)
+    mask_sorted = pos >= k  # beyond top-k rank
+    probs = torch.softmax(sorted_logits, dim=-1)
+    cum = probs.cumsum(dim=-1)
+    # keep through the first position where cumulative >= p
+    crossed = (cum - probs) >= topp.unsqueeze(1)
+    mask_sorted |= crossed
+    mask_sorted &= active.unsqueeze(1)
+    mask = torch.zeros_like(mask_sorted).scatter_(1, sorted_idx, mask_sorted)
+    logits.masked_fill_(mask, float("-inf"))
+    return logits

@@ -441,3 +441,53 @@ def test_gemm_skinny_swizzled_matches_plain(M, N, K):
     got = ops.linear(x, w, wz).float()
     err = (got - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
     assert err < 0.02, f"swizzled rel err {err}"
+
+
+# ---------- top-k / top-p filter (sampling.hip histogram kernel) ----------
+
+@pytest.mark.parametrize(
+    "B,V,pk",
+    [
+        (8, 128256, [(0.9, 0), (1.0, 10), (0.5, 50), (1.0, 0),
+                     (0.95, 100), (0.1, 0), (0.9, 5), (0.3, 1)]),
+        (3, 4096, [(0.8, 0), (1.0, 1), (0.99, 2000)]),
+    ],
+)
+def test_topk_topp_filter_matches_reference(B, V, pk):
+    torch.manual_seed(B * V)
+    logits = torch.randn(B, V, device=DEV) * 3.0
+    topp = torch.tensor([p for p, _ in pk], dtype=torch.float32, device=DEV)
+    topk = torch.tensor([k for _, k in pk], dtype=torch.int32, device=DEV)
+    got = ops.topk_topp_filter(logits.clone(), topp, topk)
+    ref = reference.topk_topp_filter(logits.cpu().clone(), topp.cpu(), topk.cpu())
+    got_kept = (got > float("-inf")).cpu()
+    ref_kept = ref > float("-inf")
+    # the kernel's cut has 3.8e-6-logit granularity: identical kept sets
+    # on continuous random logits (ties at that scale are measure-zero)
+    assert torch.equal(got_kept, ref_kept), (
+        f"kept mismatch: {got_kept.sum(1).tolist()} vs {ref_kept.sum(1).tolist()}"
+    )
+    # surviving entries keep their exact values
+    assert torch.equal(got.cpu()[got_kept], logits.cpu()[got_kept])
+
+
+def test_topk_topp_filter_untouched_rows():
+    V = 8192
+    logits = torch.randn(2, V, device=DEV)
+    orig = logits.clone()
+    topp = torch.tensor([1.0, 0.5], device=DEV)
+    topk = torch.tensor([0, 4], dtype=torch.int32, device=DEV)
+    got = ops.topk_topp_filter(logits, topp, topk)
+    assert torch.equal(got[0], orig[0])  # disabled row untouched
+    assert (got[1] > float("-inf")).sum().item() <= 4
+
+
+def test_topk_topp_filter_argmax_survives():
+    # extreme p: the argmax token must always remain sampleable
+    logits = torch.randn(4, 128256, device=DEV)
+    topp = torch.full((4,), 1e-6, device=DEV)
+    topk = torch.zeros(4, dtype=torch.int32, device=DEV)
+    am = logits.argmax(dim=-1)
+    got = ops.topk_topp_filter(logits.clone(), topp, topk)
+    for b in range(4):
+        assert got[b, am[b]] > float("-inf")
