@@ -48,6 +48,9 @@ def parse_args():
     p.add_argument("--prompt-len", type=int, default=128)
     p.add_argument("--gen-tokens", type=int, default=64)
     p.add_argument("--kv-blocks", type=int, default=None)
+    p.add_argument("--temperature", type=float, default=0.0)
+    p.add_argument("--top-p", type=float, default=1.0)
+    p.add_argument("--top-k", type=int, default=0)
     p.add_argument(
         "--mode", choices=["qps", "wave"], default="qps",
         help="qps = open-loop Poisson arrivals at --qps (the BASELINE "
@@ -67,10 +70,10 @@ def make_prompts(batch: int, prompt_len: int, vocab: int, seed: int):
     return [t.tolist() for t in toks]
 
 
-def run_wave(engine: LLMEngine, prompts, gen_tokens: int):
+def run_wave(engine: LLMEngine, prompts, gen_tokens: int, sp: dict = {}):
     t0 = time.monotonic()
     reqs = [
-        EngineRequest(p, SamplingParams(max_tokens=gen_tokens, ignore_eos=True))
+        EngineRequest(p, SamplingParams(max_tokens=gen_tokens, ignore_eos=True, **sp))
         for p in prompts
     ]
     for r in reqs:
@@ -87,7 +90,8 @@ def run_wave(engine: LLMEngine, prompts, gen_tokens: int):
     return ttfts
 
 
-def run_open_loop(engine: LLMEngine, prompts, gen_tokens: int, qps: float, seed: int):
+def run_open_loop(engine: LLMEngine, prompts, gen_tokens: int, qps: float, seed: int,
+                  sp: dict = {}):
     """Drive len(prompts) requests as a Poisson process at `qps` req/s.
 
     Returns (ttfts_ms, elapsed_s). TTFT is measured from each request's
@@ -100,7 +104,7 @@ def run_open_loop(engine: LLMEngine, prompts, gen_tokens: int, qps: float, seed:
         arrivals.append(t)
         t += rng.expovariate(qps)
     reqs = [
-        EngineRequest(p, SamplingParams(max_tokens=gen_tokens, ignore_eos=True))
+        EngineRequest(p, SamplingParams(max_tokens=gen_tokens, ignore_eos=True, **sp))
         for p in prompts
     ]
     t0 = time.monotonic()
@@ -167,11 +171,15 @@ def main():
         if have_cuda:
             torch.cuda.synchronize(engine.device)
 
+    sp = {}
+    if args.temperature > 0:
+        sp = {"temperature": args.temperature, "top_p": args.top_p, "top_k": args.top_k}
+
     # warmup (closed-loop waves; also the capacity estimate for --qps auto)
     wave_rates = []
     for w in range(args.warmup):
         tw = time.monotonic()
-        run_wave(engine, make_prompts(args.batch, args.prompt_len, vocab, 1000 + w), args.gen_tokens)
+        run_wave(engine, make_prompts(args.batch, args.prompt_len, vocab, 1000 + w), args.gen_tokens, sp)
         wave_rates.append(args.batch / (time.monotonic() - tw))
 
     qps = None
@@ -199,13 +207,13 @@ def main():
                 args.batch, args.prompt_len, vocab, 2000 + s + rank * 7919
             )
         ttfts, _ = run_open_loop(
-            engine, prompts, args.gen_tokens, qps, seed=97 + rank
+            engine, prompts, args.gen_tokens, qps, seed=97 + rank, sp=sp
         )
     else:
         for s in range(args.steps):
             ttfts += run_wave(
                 engine, make_prompts(args.batch, args.prompt_len, vocab, 2000 + s + rank * 7919),
-                args.gen_tokens,
+                args.gen_tokens, sp,
             )
     barrier_sync()
     elapsed = time.monotonic() - t_start
@@ -243,6 +251,7 @@ def main():
                 "gen_tokens": args.gen_tokens,
                 "parallelism": f"dp{world_size}",
                 "mode": "open-loop-poisson" if args.mode == "qps" else "closed-loop-waves",
+                "sampling": sp or "greedy",
                 "qps_offered_per_gpu": round(qps, 1) if qps else None,
                 "p50_ttft_ms": round(statistics.median(ttfts), 2) if ttfts else None,
                 "p95_ttft_ms": round(ttfts[int(len(ttfts) * 0.95)], 2) if ttfts else None,
