@@ -103,6 +103,7 @@ class EngineRequest:
         self.num_cached = 0  # prompt tokens served from the prefix cache
         self.prefill_pos = 0  # prompt tokens already written to the KV cache
         self.created = time.monotonic()
+        self.enqueued: float = self.created
         self.prefill_start_time: Optional[float] = None
         self.first_token_time: Optional[float] = None
         self.finished_time: Optional[float] = None
@@ -147,6 +148,8 @@ class LLMEngine:
         prefix_caching: bool = False,
         prefill_budget: int = 8192,
         tokenizer: Optional[object] = None,
+        admit_min_batch: int = 16,
+        admit_max_wait: float = 0.004,
     ):
         full_config = get_model_config(model) if isinstance(model, str) else model
         self.full_config = full_config
@@ -180,6 +183,8 @@ class LLMEngine:
             prefix_caching=prefix_caching,
         )
         self.prefix_caching = prefix_caching
+        self.admit_min_batch = max(1, admit_min_batch)
+        self.admit_max_wait = admit_max_wait
 
         # hipGraph-captured decode (GPU only; TP group ops are capturable
         # with RCCL but kept off by default under TP until validated)
@@ -249,6 +254,7 @@ class LLMEngine:
                 f"Prompt of {len(req.prompt_ids)} tokens exceeds max_model_len={self.max_model_len}"
             )
         with self._work:
+            req.enqueued = time.monotonic()
             self.waiting.append(req)
             self.stats["requests"] += 1
             self._work.notify_all()
@@ -274,7 +280,20 @@ class LLMEngine:
     # ---- scheduling ----
     def _admit(self) -> List[EngineRequest]:
         """Move waiting requests into the prefilling set (blocks for the
-        whole prompt are allocated up front; KV fills chunk by chunk)."""
+        whole prompt are allocated up front; KV fills chunk by chunk).
+
+        Admission batching: while decodes are running, hold a trickle of
+        arrivals back (up to admit_min_batch or admit_max_wait, whichever
+        first) so open-loop traffic produces a few large prefill steps
+        instead of many one-prompt steps that each stall the whole decode
+        batch. An idle engine always admits immediately."""
+        if (
+            self.waiting
+            and (self.running or self.prefilling)
+            and len(self.waiting) < self.admit_min_batch
+            and (time.monotonic() - self.waiting[0].enqueued) < self.admit_max_wait
+        ):
+            return []
         admitted: List[EngineRequest] = []
         while (
             self.waiting
@@ -384,16 +403,14 @@ class LLMEngine:
                 work.append((req, req.prefill_pos, req.prefill_pos + chunk,
                              req.prefill_pos + chunk == L))
                 budget -= chunk
-            # decode rows ride along ONLY while some prompt is being
-            # chunked (a whole-prompt prefill is one short step; adding
-            # decode rows to it desyncs batch-uniform workloads and costs
-            # the hipGraph path for no stall-avoidance benefit)
-            chunking = any(
-                (start > req.num_cached) or (not final)
-                for (req, start, end, final) in work
-            )
+            # decode rows ride along whenever any are running: a prefill
+            # step without them stalls the whole decode batch for its
+            # duration, which dominates under continuous (open-loop)
+            # arrivals. Batch-uniform wave workloads are unaffected: their
+            # prefills happen while nothing is decoding, so the
+            # pure-decode hipGraph path still serves every decode step.
             dec_reqs: List[EngineRequest] = []
-            if chunking:
+            if True:
                 i = 0
                 while i < len(self.running):
                     req = self.running[i]

@@ -211,3 +211,38 @@ def test_seeded_sampling_reproducible_across_batches():
         return reqs[0].out_ids
 
     assert run(0) == run(3)
+
+
+def test_admission_batching_holds_trickle_then_flushes():
+    """While decodes run, a single new arrival waits (up to admit_max_wait)
+    so open-loop traffic prefills in batches; the hold must flush by time
+    and never deadlock an idle engine."""
+    import time as _time
+
+    eng = LLMEngine(
+        model="tiny-llama", device="cpu", dtype=torch.float32, num_blocks=64,
+        admit_min_batch=4, admit_max_wait=0.05,
+    )
+    # a long-running request keeps the engine busy
+    bg = EngineRequest([1, 2, 3, 4], SamplingParams(max_tokens=64, ignore_eos=True))
+    eng.add_request(bg)
+    eng.step()  # prefill, bg now running
+    late = EngineRequest([5, 6, 7], SamplingParams(max_tokens=4, ignore_eos=True))
+    eng.add_request(late)
+    eng.step()
+    assert late.state == "waiting", "trickle arrival must be held back"
+    _time.sleep(0.06)
+    eng.step()  # hold expired -> admitted
+    assert late.state != "waiting"
+    while late.state in ("waiting", "running"):
+        eng.step()
+    assert late.state == "finished" and len(late.out_ids) == 4
+
+
+def test_admission_batch_admits_when_idle():
+    eng = LLMEngine(
+        model="tiny-llama", device="cpu", dtype=torch.float32, num_blocks=64,
+        admit_min_batch=8, admit_max_wait=10.0,
+    )
+    r = eng.generate([1, 2, 3], SamplingParams(max_tokens=3, ignore_eos=True))
+    assert r.state == "finished"  # idle engine never waits for a batch
