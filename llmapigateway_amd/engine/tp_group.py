@@ -85,10 +85,21 @@ def _worker_main(
         num_blocks=num_blocks if num_blocks is not None else (None if have_cuda else 256),
         max_model_len=max_model_len,
         seed=seed,  # same seed on every rank: identical sampling decisions
-        use_hipgraph=False,
+        # hipGraph decode under TP: RCCL collectives are capturable on
+        # ROCm; a capture failure falls back to eager (engine.step's
+        # graph_runner try/except). LLMAPI_NO_TP_HIPGRAPH=1 opts out.
+        use_hipgraph=(
+            None if have_cuda and not os.environ.get("LLMAPI_NO_TP_HIPGRAPH")
+            else False
+        ),
         tp_group=dist.group.WORLD,
         tp_rank=rank,
         tp_size=world,
+        # lockstep determinism: admission must be a pure function of the
+        # broadcast command stream — the wall-clock admission hold-back
+        # would diverge across ranks
+        admit_min_batch=1,
+        admit_max_wait=0.0,
     )
     reqs: Dict[str, EngineRequest] = {}
 
@@ -115,27 +126,59 @@ def _worker_main(
                 )
             )
 
+    # Lockstep command channel: rank 0 drains the gateway queue and the
+    # group agrees on each step's command batch through ONE fixed-size
+    # int32 broadcast (the pickled blob follows only when non-empty) —
+    # the round-1 per-step broadcast_object_list cost 2 object
+    # collectives + pickle on EVERY decode step. Checked only every
+    # CMD_STRIDE busy steps (deterministic across ranks: the step counter
+    # advances in lockstep), so steady-state decode pays one tiny
+    # broadcast per stride.
+    import pickle
+
+    CMD_STRIDE = 4
+    flag = torch.zeros(1, dtype=torch.int32, device=device)
+    since_check = CMD_STRIDE  # first loop always checks
+    t_bcast = 0.0  # busy-path time inside the command collective
+    t_step = 0.0
     stop = False
     while not stop:
         cmds: List[tuple] = []
-        if rank == 0:
-            busy = engine.has_work()
-            try:
-                cmds.append(inbox.get(timeout=0.0 if busy else 0.05))
-            except queue_mod.Empty:
-                pass
-            while True:
+        # `since_check` advances only on lockstep busy steps and resets at
+        # every collective, so all ranks agree on which iterations check
+        # (has_work() is identical across ranks by the lockstep invariant)
+        if since_check >= CMD_STRIDE or not engine.has_work():
+            blob = b""
+            if rank == 0:
+                raw: List[tuple] = []
+                busy = engine.has_work()
                 try:
-                    cmds.append(inbox.get_nowait())
+                    raw.append(inbox.get(timeout=0.0 if busy else 0.05))
                 except queue_mod.Empty:
-                    break
-            if not busy and not cmds:
-                continue  # idle: don't broadcast, workers wait in broadcast
-        # lockstep point: every step is bracketed by one (tiny) object
-        # broadcast so all ranks admit/abort at the same step boundary
-        box = [cmds]
-        dist.broadcast_object_list(box, src=0)
-        cmds = box[0]
+                    pass
+                while True:
+                    try:
+                        raw.append(inbox.get_nowait())
+                    except queue_mod.Empty:
+                        break
+                if not busy and not raw:
+                    continue  # idle: workers are parked in the broadcast
+                blob = pickle.dumps(raw) if raw else b""
+                flag[0] = len(blob)
+            tb0 = time.monotonic() if engine.has_work() else None
+            dist.broadcast(flag, src=0)
+            since_check = 0
+            n = int(flag.item())
+            if tb0 is not None:
+                t_bcast += time.monotonic() - tb0
+            if n:
+                buf = torch.empty(n, dtype=torch.uint8, device=device)
+                if rank == 0:
+                    buf.copy_(torch.frombuffer(bytearray(blob), dtype=torch.uint8))
+                dist.broadcast(buf, src=0)
+                cmds = pickle.loads(bytes(buf.cpu().numpy().tobytes()))
+        else:
+            since_check += 1
         for cmd in cmds:
             kind = cmd[0]
             if kind == "add":
@@ -169,12 +212,17 @@ def _worker_main(
         if stop:
             break
         if engine.has_work():
+            ts0 = time.monotonic()
             try:
                 engine.step()
             except Exception:
                 # step() already failed + reported the affected requests
                 logger.exception("tp rank %d step error", rank)
+            t_step += time.monotonic() - ts0
 
+    if rank == 0:
+        # lockstep overhead report (world-4 CPU test asserts bcast < 5%)
+        outbox.put(("lockstep_stats", None, {"bcast_s": t_bcast, "step_s": t_step}))
     dist.destroy_process_group()
 
 
@@ -217,6 +265,7 @@ class TPEngineClient:
         for p in self.procs:
             p.start()
         self._reqs: Dict[str, object] = {}
+        self.lockstep_stats: Optional[dict] = None
         self._lock = threading.Lock()
         self._dead: Optional[str] = None
         self.stats: Dict[str, float] = {"requests": 0, "finished": 0, "failed": 0}
@@ -268,6 +317,9 @@ class TPEngineClient:
                 continue
             except (EOFError, OSError):
                 return
+            if kind == "lockstep_stats":
+                self.lockstep_stats = value
+                continue
             with self._lock:
                 req = self._reqs.get(rid)
             if req is None:

@@ -111,6 +111,19 @@ MODEL_PRESETS = {
         rope_theta=10000.0,
         max_positions=512,
     ),
+    # tiny shape divisible for world-4 TP lockstep tests
+    "tiny-llama-tp4": ModelConfig(
+        name="tiny-llama-tp4",
+        hidden_size=256,
+        intermediate_size=512,
+        num_layers=2,
+        num_heads=8,
+        num_kv_heads=4,
+        vocab_size=512,
+        head_dim=32,
+        rope_theta=10000.0,
+        max_positions=512,
+    ),
     # small-but-real shape for 1-GPU kernel shakedown
     "llama-1b": ModelConfig(
         name="llama-1b",
