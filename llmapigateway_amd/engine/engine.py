@@ -406,11 +406,15 @@ class LLMEngine:
             # decode rows ride along whenever any are running: a prefill
             # step without them stalls the whole decode batch for its
             # duration, which dominates under continuous (open-loop)
-            # arrivals. Batch-uniform wave workloads are unaffected: their
-            # prefills happen while nothing is decoding, so the
-            # pure-decode hipGraph path still serves every decode step.
+            # arrivals. LLMAPI_MIXED_RIDE=chunking restores the round-1
+            # gate (ride only while a prompt is mid-chunk) for A/B.
+            ride = os.environ.get("LLMAPI_MIXED_RIDE", "always")
+            chunking = ride == "always" or any(
+                (start > req.num_cached) or (not final)
+                for (req, start, end, final) in work
+            )
             dec_reqs: List[EngineRequest] = []
-            if True:
+            if chunking:
                 i = 0
                 while i < len(self.running):
                     req = self.running[i]
