@@ -148,7 +148,7 @@ class LLMEngine:
         prefix_caching: bool = False,
         prefill_budget: int = 8192,
         tokenizer: Optional[object] = None,
-        admit_min_batch: int = 16,
+        admit_min_batch: Optional[int] = None,
         admit_max_wait: float = 0.004,
     ):
         full_config = get_model_config(model) if isinstance(model, str) else model
@@ -183,6 +183,8 @@ class LLMEngine:
             prefix_caching=prefix_caching,
         )
         self.prefix_caching = prefix_caching
+        if admit_min_batch is None:
+            admit_min_batch = int(os.environ.get("LLMAPI_ADMIT_MIN", "16"))
         self.admit_min_batch = max(1, admit_min_batch)
         self.admit_max_wait = admit_max_wait
 
@@ -408,7 +410,7 @@ class LLMEngine:
             # duration, which dominates under continuous (open-loop)
             # arrivals. LLMAPI_MIXED_RIDE=chunking restores the round-1
             # gate (ride only while a prompt is mid-chunk) for A/B.
-            ride = os.environ.get("LLMAPI_MIXED_RIDE", "always")
+            ride = os.environ.get("LLMAPI_MIXED_RIDE", "chunking")
             chunking = ride == "always" or any(
                 (start > req.num_cached) or (not final)
                 for (req, start, end, final) in work
