@@ -56,6 +56,7 @@ def parse_args():
         help="qps = open-loop Poisson arrivals at --qps (the BASELINE "
         "metric); wave = round-1 closed-loop waves",
     )
+    p.add_argument("--debug-stats", action="store_true")
     p.add_argument(
         "--qps", default="auto",
         help="offered request rate per GPU for --mode qps; 'auto' = 1.1x "
@@ -261,6 +262,8 @@ def main():
             },
         }
         print(json.dumps(result))
+        if args.debug_stats:
+            print("ENGINE_STATS", json.dumps(engine.stats), file=sys.stderr)
 
     if dist is not None:
         dist.destroy_process_group()

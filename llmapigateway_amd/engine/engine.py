@@ -184,7 +184,7 @@ class LLMEngine:
         )
         self.prefix_caching = prefix_caching
         if admit_min_batch is None:
-            admit_min_batch = int(os.environ.get("LLMAPI_ADMIT_MIN", "16"))
+            admit_min_batch = int(os.environ.get("LLMAPI_ADMIT_MIN", "1"))
         self.admit_min_batch = max(1, admit_min_batch)
         self.admit_max_wait = admit_max_wait
 
@@ -247,6 +247,14 @@ class LLMEngine:
             "prefill_tokens": 0,
             "decode_tokens": 0,
             "steps": 0,
+            # step-kind anatomy (serving-regime diagnosis)
+            "prefill_steps": 0,
+            "mixed_steps": 0,
+            "decode_steps": 0,
+            "graph_steps": 0,
+            "flushes": 0,
+            "prefill_s": 0.0,
+            "decode_s": 0.0,
         }
 
     # ---- request intake ----
@@ -348,6 +356,7 @@ class LLMEngine:
             return
         reqs, n, event = self._pending
         self._pending = None
+        self.stats["flushes"] += 1
         event.synchronize()
         tokens = self._pend_pinned[:n].tolist()
         with self._lock:
@@ -362,14 +371,18 @@ class LLMEngine:
         try:
             if has_prefill:
                 self._flush_pending()  # running set is about to change
+                t0 = time.monotonic()
                 produced = self._prefill_step()
+                self.stats["prefill_s"] += time.monotonic() - t0
             else:
                 with self._lock:
                     idle = not self.running
                 if idle:
                     self._flush_pending()  # deliver the final in-flight step
                     return 0
+                t0 = time.monotonic()
                 produced = self._decode_step()
+                self.stats["decode_s"] += time.monotonic() - t0
         except Exception as e:
             logger.exception("engine step failed")
             with self._lock:
@@ -500,6 +513,7 @@ class LLMEngine:
             dec_context_lens=dec_ctx_t,
             logits_indices=torch.from_numpy(logits_idx).to(device),
         )
+        self.stats["mixed_steps" if nd else "prefill_steps"] += 1
         logits = self.model.forward(batch, self.kv.k_caches, self.kv.v_caches)
         finals = [w[0] for w in work if w[3]]
         sample_reqs = reqs + dec_reqs
@@ -599,10 +613,12 @@ class LLMEngine:
                 dtype=np.int64, count=n,
             )
 
+        self.stats["decode_steps"] += 1
         logits = None
         if self.graph_runner is not None:
             try:
                 logits = self.graph_runner.run(last_tokens, pos, slots, tables_np, ctx)
+                self.stats["graph_steps"] += 1
             except Exception:
                 logger.exception("hipGraph decode failed; falling back to eager")
                 self.graph_runner = None
