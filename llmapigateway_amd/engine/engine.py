@@ -149,7 +149,7 @@ class LLMEngine:
         prefill_budget: int = 8192,
         tokenizer: Optional[object] = None,
         admit_min_batch: Optional[int] = None,
-        admit_max_wait: float = 0.004,
+        admit_max_wait: Optional[float] = None,
     ):
         full_config = get_model_config(model) if isinstance(model, str) else model
         self.full_config = full_config
@@ -186,6 +186,8 @@ class LLMEngine:
         if admit_min_batch is None:
             admit_min_batch = int(os.environ.get("LLMAPI_ADMIT_MIN", "1"))
         self.admit_min_batch = max(1, admit_min_batch)
+        if admit_max_wait is None:
+            admit_max_wait = float(os.environ.get("LLMAPI_ADMIT_WAIT", "0.004"))
         self.admit_max_wait = admit_max_wait
 
         # hipGraph-captured decode (GPU only; TP group ops are capturable
