@@ -221,9 +221,9 @@ def test_admission_batching_holds_trickle_then_flushes():
 
     eng = LLMEngine(
         model="tiny-llama", device="cpu", dtype=torch.float32, num_blocks=64,
-        admit_min_batch=4, admit_max_wait=0.05,
+        max_batch_size=2, admit_min_batch=4, admit_max_wait=0.05,
     )
-    # a long-running request keeps the engine busy
+    # a long-running request keeps the engine >= half-loaded
     bg = EngineRequest([1, 2, 3, 4], SamplingParams(max_tokens=64, ignore_eos=True))
     eng.add_request(bg)
     eng.step()  # prefill, bg now running
