@@ -162,9 +162,10 @@ class LLMEngine:
         )
         self.max_batch_size = max_batch_size
         if tokenizer is None:
-            from .tokenizer import ByteTokenizer
+            from .tokenizer import get_tokenizer
 
-            tokenizer = ByteTokenizer()
+            # real BPE for production presets, byte stand-in for tiny ones
+            tokenizer = get_tokenizer("auto", full_config.vocab_size)
         self.tokenizer = tokenizer  # stop-string decode + EOS detection
 
         if self.device.type == "cuda":

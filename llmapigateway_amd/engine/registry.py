@@ -266,12 +266,13 @@ class EngineRegistry:
                 f"(queue depth {depth} >= {self.settings.engine_max_queue})"
             )
 
+        tokenizer = getattr(engine, "tokenizer", None) or self.tokenizer
         messages = payload.get("messages")
         if messages:
-            prompt_text = self.tokenizer.render_chat(messages)
+            prompt_text = tokenizer.render_chat(messages)
         else:
             prompt_text = str(payload.get("prompt", ""))
-        prompt_ids = self.tokenizer.encode(prompt_text)
+        prompt_ids = tokenizer.encode(prompt_text)
         if len(prompt_ids) >= engine.max_model_len:
             return None, (
                 f"Prompt of {len(prompt_ids)} tokens exceeds model context "
@@ -334,7 +335,8 @@ class EngineRegistry:
 
         if not is_streaming:
             return await self._collect_nonstream(
-                req, queue, kind, value, completion_id, created, model_name, provider_name
+                req, engine, queue, kind, value, completion_id, created,
+                model_name, provider_name,
             )
         return (
             self._stream_response(
@@ -400,7 +402,7 @@ class EngineRegistry:
         model_name = payload.get("model", "unknown")
         completion_id = f"chatcmpl-{uuid.uuid4().hex[:24]}"
         created = int(time.time())
-        tokenizer = self.tokenizer
+        tokenizer = getattr(engine, "tokenizer", None) or self.tokenizer
 
         try:
             first = await asyncio.wait_for(shared.get(), timeout=300.0)
@@ -502,7 +504,8 @@ class EngineRegistry:
         )
 
     async def _collect_nonstream(
-        self, req, queue, kind, value, completion_id, created, model_name, provider_name
+        self, req, engine, queue, kind, value, completion_id, created,
+        model_name, provider_name,
     ):
         tokens: List[int] = []
         finished: Optional[EngineRequest] = None
@@ -515,7 +518,7 @@ class EngineRegistry:
             kind, value = await queue.get()
         if finished is not None and finished.state == "failed":
             return None, finished.error or "engine failure"
-        text = self.tokenizer.decode(tokens)
+        text = (getattr(engine, "tokenizer", None) or self.tokenizer).decode(tokens)
         for s in req.params.stop:  # OpenAI semantics: stop string excluded
             idx = text.find(s)
             if idx >= 0:
@@ -548,7 +551,7 @@ class EngineRegistry:
     def _stream_response(
         self, req, engine, queue, first_kind, first_value, completion_id, created, model_name
     ) -> StreamingResponse:
-        tokenizer = self.tokenizer
+        tokenizer = getattr(engine, "tokenizer", None) or self.tokenizer
 
         def chunk(delta: Dict[str, Any], finish: Optional[str] = None, usage=None) -> bytes:
             obj: Dict[str, Any] = {

@@ -242,8 +242,10 @@ class TPEngineClient:
         start_timeout: float = 600.0,
     ):
         from ..models.configs import get_model_config
+        from .tokenizer import get_tokenizer
 
         self.full_config = get_model_config(model) if isinstance(model, str) else model
+        self.tokenizer = get_tokenizer("auto", self.full_config.vocab_size)
         self.max_model_len = max_model_len or self.full_config.max_positions
         self.tp = tp
         ctx = mp.get_context("spawn")
