@@ -1,21 +1,27 @@
 # LLM API Gateway (MI355X) — deployment image.
-# Parity: the reference's multi-stage python-slim Dockerfile (+ the ROCm
-# runtime needed for local engines). Build on a host with ROCm 7.x and the
-# gfx950 extension prebuilt in-tree (setup.py build_ext --inplace), or use
-# a rocm/pytorch base to build inside the image.
-FROM rocm/pytorch:latest AS base
+# Multi-stage (parity with the reference's python-slim multi-stage build,
+# /root/reference/Dockerfile:1-93, on a ROCm base for the local engines):
+# the builder compiles the gfx950 HIP extension in-tree (hipcc
+# cross-compiles without a GPU); the runtime stage carries only the app,
+# the built extension and the runtime deps.
+
+FROM rocm/pytorch:latest AS builder
+
+WORKDIR /build
+COPY requirements.txt setup.py ./
+COPY llmapigateway_amd/ llmapigateway_amd/
+RUN PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+FROM rocm/pytorch:latest AS runtime
 
 WORKDIR /app
-
 COPY requirements.txt .
 RUN pip install --no-cache-dir -r requirements.txt
 
-COPY llmapigateway_amd/ llmapigateway_amd/
+# app + the prebuilt extension from the builder stage
+COPY --from=builder /build/llmapigateway_amd/ llmapigateway_amd/
 COPY static/ static/
-COPY setup.py main.py ./
-
-# compile the gfx950 HIP extension in-tree (cross-compiles without a GPU)
-RUN PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+COPY main.py ./
 
 COPY docker/entrypoint.sh /entrypoint.sh
 COPY docker/healthcheck.py /healthcheck.py

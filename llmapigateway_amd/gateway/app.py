@@ -118,18 +118,24 @@ def create_app(
         m.refresh_engine_gauges(registry)
         if registry is not None:
             try:
+                # counters must be monotonic: advance by the delta since
+                # the last scrape, tracked on our side (no reliance on
+                # prometheus-client private internals)
+                last = getattr(app.state, "_metrics_last", None)
+                if last is None:
+                    last = app.state._metrics_last = {}
                 gen = sum(r.get("decode_tokens", 0) for r in registry.stats())
                 pre = sum(r.get("prefill_tokens", 0) for r in registry.stats())
                 pc = sum(r.get("prefix_cached_tokens", 0) for r in registry.stats())
-                # counters must be monotonic: advance by the delta since last scrape
-                for counter, total in (
-                    (m.generated_tokens_total, gen),
-                    (m.prefill_tokens_total, pre),
-                    (m.prefix_cached_tokens_total, pc),
+                for name, counter, total in (
+                    ("gen", m.generated_tokens_total, gen),
+                    ("pre", m.prefill_tokens_total, pre),
+                    ("pc", m.prefix_cached_tokens_total, pc),
                 ):
-                    delta = total - counter._value.get()
+                    delta = total - last.get(name, 0)
                     if delta > 0:
                         counter.inc(delta)
+                        last[name] = total
             except Exception:  # pragma: no cover
                 pass
         return Response(content=m.render(), media_type=CONTENT_TYPE_LATEST)

@@ -95,12 +95,44 @@ class RequestLoggingMiddleware:
         scope.setdefault("state", {})["request_id"] = request_id
         start = time.perf_counter()
         method, path = scope.get("method", ""), scope.get("path", "")
+
+        # POST chat payload summary with messages/tools excluded (the
+        # reference logs the same shape: request_logging.py:49-61). The
+        # body is buffered and replayed through a wrapped `receive`.
+        summary = None
+        if method == "POST" and path.endswith("/chat/completions"):
+            body = b""
+            more = True
+            while more:
+                message = await receive()
+                body += message.get("body", b"")
+                more = message.get("more_body", False)
+            try:
+                payload = json.loads(body or b"{}")
+                if isinstance(payload, dict):
+                    summary = {
+                        k: v for k, v in payload.items()
+                        if k not in ("messages", "tools")
+                    }
+            except ValueError:
+                summary = "<unparseable>"
+            replayed = {"done": False}
+            orig_receive = receive
+
+            async def receive():  # noqa: F811 — replay the buffered body
+                if not replayed["done"]:
+                    replayed["done"] = True
+                    return {"type": "http.request", "body": body, "more_body": False}
+                # afterwards delegate (disconnect detection during SSE)
+                return await orig_receive()
+
         logger.info(
-            "rid=%s --> %s %s headers=%s",
+            "rid=%s --> %s %s headers=%s%s",
             request_id,
             method,
             path,
             _mask_headers(scope.get("headers", [])),
+            f" payload={summary}" if summary is not None else "",
         )
 
         status_holder = {"status": 0}

@@ -310,8 +310,10 @@ __global__ void attention_prefill_kernel(
     }
 }
 
-// Wait: accO[dsub][r] must use the PV C-layout row = lk*4 + r, col = lrow —
-// which matches the S layout used above (both are 16x16x32 C fragments).
+// accO[dsub][r] uses the PV C-fragment layout row = lk*4 + r, col = lrow,
+// identical to the S layout above: both are mfma_f32_16x16x32 C fragments,
+// and the C/D lane map on gfx950 is dtype- and operand-independent
+// (cdna_hip_programming.md §3), so no relayout is needed between S and O.
 
 extern "C" hipError_t launch_attention_prefill(
     void* out, const void* q, const void* k, const void* v,

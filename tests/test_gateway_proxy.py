@@ -414,3 +414,46 @@ def test_models_degrades_when_upstream_unreachable(tmp_path):
         assert r.status_code == 200
         ids = [m["id"] for m in r.json()["data"]]
         assert ids and all(i.startswith("gw/") for i in ids)
+
+
+def test_request_log_includes_payload_summary(caplog):
+    """POST chat payloads are logged with messages/tools excluded
+    (reference request_logging.py:49-61 parity)."""
+    import logging as _logging
+
+    from llmapigateway_amd.gateway.middleware import RequestLoggingMiddleware
+
+    captured = {}
+
+    async def app(scope, receive, send):
+        msg = await receive()
+        captured["body"] = msg.get("body")
+        await send({"type": "http.response.start", "status": 200, "headers": []})
+        await send({"type": "http.response.body", "body": b"{}"})
+
+    mw = RequestLoggingMiddleware(app)
+    body = (
+        b'{"model": "m", "temperature": 0.5, '
+        b'"messages": [{"role": "user", "content": "secret"}], "tools": []}'
+    )
+    sent = []
+
+    async def receive():
+        return {"type": "http.request", "body": body, "more_body": False}
+
+    async def send(m):
+        sent.append(m)
+
+    scope = {
+        "type": "http",
+        "method": "POST",
+        "path": "/v1/chat/completions",
+        "headers": [],
+    }
+    import asyncio
+
+    with caplog.at_level(_logging.INFO):
+        asyncio.get_event_loop().run_until_complete(mw(scope, receive, send))
+    assert captured["body"] == body  # replayed intact to the app
+    line = next(r.message for r in caplog.records if "payload=" in r.message)
+    assert "temperature" in line and "secret" not in line and "messages" not in line
