@@ -453,7 +453,7 @@ def test_request_log_includes_payload_summary(caplog):
     import asyncio
 
     with caplog.at_level(_logging.INFO):
-        asyncio.get_event_loop().run_until_complete(mw(scope, receive, send))
+        asyncio.run(mw(scope, receive, send))
     assert captured["body"] == body  # replayed intact to the app
     line = next(r.message for r in caplog.records if "payload=" in r.message)
     assert "temperature" in line and "secret" not in line and "messages" not in line
