@@ -1,0 +1,160 @@
+"""Memory-safety guard tests for every gfx950 kernel: each output (and
+scratch) tensor is carved out of a larger canary-filled buffer; after the
+kernel runs, the guard bands on both sides must be untouched. This is the
+hipMemcheck-class check SURVEY §5 asks for, built from first principles
+(no ASAN runtime ships for amdgpu in this image); tools/sanitize_gpu.sh
+runs this file plus the full GPU suite under AMD_SERIALIZE_KERNEL=3 so
+any stray fault is attributed to its launching kernel.
+"""
+
+import pytest
+import torch
+
+from llmapigateway_amd import ops
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+PAD = 1024  # guard elements on each side
+CANARY = 0x5A
+
+
+def guarded(shape, dtype):
+    """Return (view, check) where view is a tensor of `shape` carved from
+    a canary-guarded buffer and check() asserts the guards are intact."""
+    numel = 1
+    for s in shape:
+        numel *= s
+    buf = torch.empty(numel + 2 * PAD, dtype=dtype, device=DEV)
+    raw = buf.view(torch.uint8)
+    raw.fill_(CANARY)
+    view = buf[PAD : PAD + numel].view(shape)
+    esz = buf.element_size()
+
+    def check(name=""):
+        torch.cuda.synchronize()
+        head = buf[:PAD].view(torch.uint8)
+        tail = buf[PAD + numel :].view(torch.uint8)
+        assert bool((head == CANARY).all()), f"{name}: guard before tensor clobbered"
+        assert bool((tail == CANARY).all()), f"{name}: guard after tensor clobbered"
+
+    _ = esz
+    return view, check
+
+
+def test_rmsnorm_guards():
+    x = torch.randn(33, 4096, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(4096, dtype=torch.bfloat16, device=DEV)
+    out, check = guarded((33, 4096), torch.bfloat16)
+    ops._native().rmsnorm(out, x, w, 1e-5)
+    check("rmsnorm out")
+
+
+def test_swiglu_guards():
+    x = torch.randn(17, 2 * 14336, dtype=torch.bfloat16, device=DEV)
+    out, check = guarded((17, 14336), torch.bfloat16)
+    ops._native().swiglu(out, x)
+    check("swiglu out")
+
+
+def test_rope_kv_write_guards():
+    T, Hq, Hkv, D, BS, NB = 50, 8, 2, 128, 16, 8
+    qkv = torch.randn(T, (Hq + 2 * Hkv) * D, dtype=torch.bfloat16, device=DEV)
+    q = qkv[:, : Hq * D].view(T, Hq, D)
+    k = qkv[:, Hq * D : (Hq + Hkv) * D].view(T, Hkv, D)
+    v = qkv[:, (Hq + Hkv) * D :].view(T, Hkv, D)
+    kc, kcheck = guarded((NB, Hkv, BS, D), torch.bfloat16)
+    vc, vcheck = guarded((NB, Hkv, BS, D), torch.bfloat16)
+    pos = torch.arange(T, dtype=torch.int64, device=DEV)
+    # include skip slots (-1) and the LAST slot (boundary write)
+    slots = torch.randperm(NB * BS, device=DEV)[:T]
+    slots[0] = -1
+    slots[1] = NB * BS - 1
+    cs = ops.build_rope_cache(64, D, 10000.0, device=DEV)
+    ops.rope_and_kv_write(q, k, v, kc, vc, pos, cs, slots)
+    kcheck("k_cache")
+    vcheck("v_cache")
+
+
+def test_attention_prefill_guards():
+    Hq, Hkv, D, L = 8, 2, 128, 200
+    q = torch.randn(L, Hq, D, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(L, Hkv, D, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(L, Hkv, D, dtype=torch.bfloat16, device=DEV)
+    cu = torch.tensor([0, L], dtype=torch.int32, device=DEV)
+    out, check = guarded((L, Hq, D), torch.bfloat16)
+    tile_seq, tile_off = ops.build_prefill_tiles([L], DEV)
+    ops._native().attention_prefill(
+        out, q, k, v, cu, tile_seq, tile_off, float(D) ** -0.5,
+        None, None, None, None,
+    )
+    check("prefill out")
+
+
+def test_attention_decode_guards():
+    B, Hq, Hkv, BS, D, NB = 8, 32, 8, 64, 128, 40
+    kc = torch.randn(NB, Hkv, BS, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn_like(kc)
+    maxb = 8
+    bt = torch.randperm(NB)[: B * maxb].view(B, maxb).int().to(DEV)
+    q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=DEV)
+    ctx = torch.tensor([1, 64, 65, 100, 128, 250, 300, 512], dtype=torch.int32, device=DEV)
+    out, check = guarded((B, Hq, D), torch.bfloat16)
+    # exercise the split-context path (scratch guards too)
+    nsplit = 4
+    pa, pacheck = guarded((B * Hq * nsplit * D,), torch.float32)
+    pm, pmcheck = guarded((B * Hq * nsplit * 2,), torch.float32)
+    ops._native().attention_decode(
+        out, q, kc, vc, bt, ctx, float(D) ** -0.5, pa, pm, nsplit
+    )
+    check("decode out")
+    pacheck("decode part_acc")
+    pmcheck("decode part_ml")
+
+
+@pytest.mark.parametrize("M,N,K,nf,nsk", [(256, 4096, 4096, 4, 4), (33, 4096, 14336, 8, 8)])
+def test_gemm_m256_guards(M, N, K, nf, nsk):
+    x = (torch.randn(M, K, device=DEV) * 0.5).bfloat16()
+    w = (torch.randn(N, K, device=DEV) * 0.02).bfloat16()
+    wf = ops.swizzle_weight_frag(w)
+    y, ycheck = guarded((M, N), torch.bfloat16)
+    ws, wscheck = guarded((nsk * M * N,), torch.float32)
+    for variant in (0, 1):
+        ops._native().gemm_m256(y, x, wf, ws, nsk, nf, variant, 0 if nf == 4 else 0)
+        ycheck(f"gemm_m256 y v{variant}")
+        wscheck(f"gemm_m256 slab v{variant}")
+    ref = x.float() @ w.float().T
+    err = (y.float() - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
+    assert err < 0.02
+
+
+def test_gemm_skinny_guards():
+    M, N, K = 16, 4096, 4096
+    x = (torch.randn(M, K, device=DEV) * 0.5).bfloat16()
+    w = (torch.randn(N, K, device=DEV) * 0.02).bfloat16()
+    y, ycheck = guarded((M, N), torch.bfloat16)
+    nsk = 4
+    ws, wscheck = guarded((nsk * M * N,), torch.float32)
+    ops._native().gemm_skinny(y, x, w, ws, nsk, False)
+    ycheck("gemm_skinny y")
+    wscheck("gemm_skinny slab")
+
+
+def test_sampling_guards():
+    B, V = 64, 128256
+    logits = torch.randn(B, V, device=DEV)
+    temps = torch.rand(B, device=DEV)
+    noise = torch.rand(B, V, device=DEV)
+    out, ocheck = guarded((B,), torch.int64)
+    ops._native().sample(out, logits, temps, noise)
+    ocheck("sample out")
+
+
+def test_topk_topp_filter_guards():
+    B, V = 16, 128256
+    logits, check = guarded((B, V), torch.float32)
+    logits.copy_(torch.randn(B, V, device=DEV))
+    topp = torch.full((B,), 0.9, device=DEV)
+    topk = torch.full((B,), 40, dtype=torch.int32, device=DEV)
+    ops._native().topk_topp_filter(logits, topp, topk)
+    check("filter logits")
