@@ -92,7 +92,7 @@ def test_attention_prefill_guards():
 
 
 def test_attention_decode_guards():
-    B, Hq, Hkv, BS, D, NB = 8, 32, 8, 64, 128, 40
+    B, Hq, Hkv, BS, D, NB = 8, 32, 8, 64, 128, 70
     kc = torch.randn(NB, Hkv, BS, D, dtype=torch.bfloat16, device=DEV)
     vc = torch.randn_like(kc)
     maxb = 8
