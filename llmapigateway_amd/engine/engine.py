@@ -237,6 +237,12 @@ class LLMEngine:
         self.batch_notifier: Optional[Callable[[list], None]] = None
 
         self._gen = torch.Generator(device=self.device).manual_seed(seed ^ 0x5EED)
+        # block tables of requests aborted while a forward may be in
+        # flight: freed at the next step() top (stream-ordered after the
+        # in-flight KV writes) instead of immediately — an immediate free
+        # could hand the blocks to a newly admitted request while the old
+        # forward still writes into them
+        self._deferred_free: List[List[int]] = []
         self._lock = threading.Lock()
         self._work = threading.Condition(self._lock)
         self.waiting: deque[EngineRequest] = deque()
@@ -277,7 +283,11 @@ class LLMEngine:
         with self._work:
             if req in self.waiting:
                 self.waiting.remove(req)
-            self._finish(req, "aborted")
+                self._finish(req, "aborted")
+            else:
+                # running/prefilling: a forward may be mid-flight on the
+                # engine thread — defer the KV free to the next step top
+                self._finish(req, "aborted", defer_free=True)
 
     def has_work(self) -> bool:
         with self._lock:
@@ -372,6 +382,12 @@ class LLMEngine:
     def step(self) -> int:
         """Run one engine iteration. Returns number of tokens produced."""
         with self._lock:
+            if self._deferred_free:
+                # previous step's forward has been issued; aborted
+                # requests' blocks are now safe to reuse
+                for bt in self._deferred_free:
+                    self.kv.manager.free(bt)
+                self._deferred_free.clear()
             self.prefilling.extend(self._admit())
             has_prefill = bool(self.prefilling)
         try:
@@ -521,14 +537,21 @@ class LLMEngine:
         )
         self.stats["mixed_steps" if nd else "prefill_steps"] += 1
         logits = self.model.forward(batch, self.kv.k_caches, self.kv.v_caches)
-        finals = [w[0] for w in work if w[3]]
         sample_reqs = reqs + dec_reqs
         tokens = self._sample(logits, sample_reqs)  # non-final prefill rows discarded
-        final_tokens = [t for t, w in zip(tokens, work) if w[3]]
         dec_tokens = tokens[len(reqs):]
         self.stats["prefill_tokens"] += Tp
         self.stats["decode_tokens"] += nd
         with self._lock:
+            # requests aborted while the forward was in flight are gone
+            # from self.prefilling and must not be revived (their blocks
+            # sit in _deferred_free — never register them as prefixes)
+            live = [
+                (req, s, e, final)
+                for (req, s, e, final) in work
+                if req.state == "running"
+            ]
+            finals = [req for (req, s, e, final) in live if final]
             if self.prefix_caching:
                 # KV for this chunk is now written (stream-ordered before
                 # any later forward): make completed full prompt blocks
@@ -536,10 +559,14 @@ class LLMEngine:
                 # threads mutate the same refcount tables via kv.manager.
                 for req in finals:
                     self.kv.manager.register_prefix(req.prompt_ids, req.block_table)
-            for req, s, e, final in work:
+            for req, s, e, final in live:
                 req.prefill_pos = e
                 if final:
                     self.prefilling.remove(req)
+            final_tokens = [
+                t for t, (req, _, _, final) in zip(tokens, work)
+                if final and req.state == "running"
+            ]
             if nd:
                 self._deliver(dec_reqs, dec_tokens)
             self.running.extend(finals)
@@ -810,7 +837,7 @@ class LLMEngine:
             return "length"
         return None
 
-    def _finish(self, req: EngineRequest, reason: str) -> None:
+    def _finish(self, req: EngineRequest, reason: str, defer_free: bool = False) -> None:
         if req.state in ("finished", "failed"):
             return
         if req in self.running:  # e.g. abort of a running request
@@ -822,7 +849,10 @@ class LLMEngine:
         req.finished_time = time.monotonic()
         self.stats["finished" if req.state == "finished" else "failed"] += 1
         if req.block_table:
-            self.kv.manager.free(req.block_table)
+            if defer_free:
+                self._deferred_free.append(req.block_table)
+            else:
+                self.kv.manager.free(req.block_table)
             req.block_table = []
         self._free_slot(req)
         if req.on_finish is not None:

@@ -16,7 +16,7 @@ import torch
 from llmapigateway_amd.engine import EngineRequest, LLMEngine, SamplingParams
 
 
-@pytest.mark.parametrize("seed", [1234, 777, 4242])
+@pytest.mark.parametrize("seed", [1234, 777, 4242, 31337, 99, 2718, 16180, 555])
 def test_engine_soak_randomized_workload(seed):
     rng = random.Random(seed)
     engine = LLMEngine(
@@ -44,13 +44,24 @@ def test_engine_soak_randomized_workload(seed):
     driver = threading.Thread(target=loop, daemon=True)
     driver.start()
 
-    shared_prefix = [rng.randrange(3, 400) for _ in range(32)]
+    # several shared prefixes: re-hits AND evictions of cached chains
+    prefixes = [
+        [rng.randrange(3, 400) for _ in range(rng.choice([16, 32, 48]))]
+        for _ in range(3)
+    ]
+    # greedy determinism probes: identical prompts issued at different
+    # times must produce identical outputs whatever the scheduler did in
+    # between (catches KV corruption e.g. the r1 prefix-alias bug class)
+    probe_prompt = [rng.randrange(3, 400) for _ in range(24)]
+    probes = []
     reqs = []
     aborted = []
     try:
-        for i in range(40):
-            if rng.random() < 0.4:  # prefix-cache candidates
-                prompt = shared_prefix + [rng.randrange(3, 400) for _ in range(rng.randrange(1, 40))]
+        for i in range(90):
+            if rng.random() < 0.1:
+                prompt = list(probe_prompt)
+            elif rng.random() < 0.4:  # prefix-cache candidates
+                prompt = rng.choice(prefixes) + [rng.randrange(3, 400) for _ in range(rng.randrange(1, 40))]
             else:
                 prompt = [rng.randrange(3, 400) for _ in range(rng.randrange(4, 120))]
             params = SamplingParams(
@@ -60,9 +71,13 @@ def test_engine_soak_randomized_workload(seed):
                 top_p=rng.choice([1.0, 0.9]),
                 presence_penalty=rng.choice([0.0, 0.0, 0.5]),
             )
+            if prompt == probe_prompt:
+                params = SamplingParams(max_tokens=8, ignore_eos=True)
             req = EngineRequest(prompt, params)
             engine.add_request(req)
             reqs.append(req)
+            if prompt == probe_prompt:
+                probes.append(req)
             if rng.random() < 0.15:
                 victim = rng.choice(reqs)
                 engine.abort_request(victim)
@@ -84,6 +99,11 @@ def test_engine_soak_randomized_workload(seed):
         assert r.state in ("finished", "failed"), r
         if r.state == "finished" and r.finish_reason == "length" and r not in aborted:
             assert len(r.out_ids) == r.params.max_tokens, r
+    done_probes = [p for p in probes if p.state == "finished" and p not in aborted]
+    if len(done_probes) > 1:
+        first = done_probes[0].out_ids
+        for p in done_probes[1:]:
+            assert p.out_ids == first, "greedy determinism broken (KV corruption?)"
     # no KV leak: everything freed or retained as evictable cached blocks
     mgr = engine.kv.manager
     assert mgr.num_free_blocks + len(mgr._evictable) == 48
