@@ -31,6 +31,15 @@ __device__ __forceinline__ f32x4 mfma16(bf16x8 a, bf16x8 b, f32x4 c) {
     return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
 }
 
+// XOR column swizzles (16-B chunk index ^ row) applied identically at
+// write and read: the row-major tiles otherwise put every fragment
+// read's 16-lane group on one bank slot — k_lds rows are 256 B (one full
+// bank row -> up to 16-way conflict, the Guideline-4 D=128 case) and
+// vt/p rows 128 B (4-way). Measured before the fix: 1.27M
+// SQ_LDS_BANK_CONFLICT cycles per dispatch (profiles/r02_pmc_bench_kernels.csv).
+__device__ __forceinline__ int swz16(int chunk, int row) { return chunk ^ (row & 15); }
+__device__ __forceinline__ int swz8(int chunk, int row) { return chunk ^ (row & 7); }
+
 __launch_bounds__(PF_WAVES* WAVE_SIZE)
 __global__ void attention_prefill_kernel(
     bf16* __restrict__ out,             // [T, Hq, D]
@@ -124,10 +133,11 @@ __global__ void attention_prefill_kernel(
                 kraw = *reinterpret_cast<const uint4*>(k_cache + base + d0);
                 vraw = *reinterpret_cast<const uint4*>(v_cache + base + d0);
             }
-            *reinterpret_cast<uint4*>(&k_lds[kk][d0]) = kraw;
+            *reinterpret_cast<uint4*>(&k_lds[kk][swz16(d0 / 8, kk) * 8]) = kraw;
             const bf16* v8 = reinterpret_cast<const bf16*>(&vraw);
 #pragma unroll
-            for (int j = 0; j < 8; ++j) vt_lds[d0 + j][kk] = v8[j];
+            for (int j = 0; j < 8; ++j)
+                vt_lds[d0 + j][swz8(kk >> 3, d0 + j) * 8 + (kk & 7)] = v8[j];
         }
         __syncthreads();
 
@@ -138,7 +148,7 @@ __global__ void attention_prefill_kernel(
 #pragma unroll
             for (int kc = 0; kc < 4; ++kc) {
                 const bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
-                    &k_lds[sub * 16 + lrow][kc * 32 + lk * 8]);
+                    &k_lds[sub * 16 + lrow][swz16(kc * 4 + lk, sub * 16 + lrow) * 8]);
                 s[sub] = mfma16(q_frag[kc], bfrag, s[sub]);
             }
         }
@@ -183,16 +193,17 @@ __global__ void attention_prefill_kernel(
         for (int sub = 0; sub < 4; ++sub)
 #pragma unroll
             for (int r = 0; r < 4; ++r)
-                p_lds[wave][lk * 4 + r][sub * 16 + lrow] = f2bf(p_val[sub][r]);
+                p_lds[wave][lk * 4 + r][swz8(sub * 2 + (lrow >> 3), lk * 4 + r) * 8 +
+                                        (lrow & 7)] = f2bf(p_val[sub][r]);
 
 #pragma unroll
         for (int dsub = 0; dsub < 8; ++dsub) {
 #pragma unroll
             for (int kc = 0; kc < 2; ++kc) {
                 const bf16x8 pa = *reinterpret_cast<const bf16x8*>(
-                    &p_lds[wave][lrow][kc * 32 + lk * 8]);
+                    &p_lds[wave][lrow][swz8(kc * 4 + lk, lrow) * 8]);
                 const bf16x8 vb = *reinterpret_cast<const bf16x8*>(
-                    &vt_lds[dsub * 16 + lrow][kc * 32 + lk * 8]);
+                    &vt_lds[dsub * 16 + lrow][swz8(kc * 4 + lk, dsub * 16 + lrow) * 8]);
                 accO[dsub] = mfma16(pa, vb, accO[dsub]);
             }
         }
@@ -219,10 +230,11 @@ __global__ void attention_prefill_kernel(
                 vraw = *reinterpret_cast<const uint4*>(
                     v + t * v_stride + (size_t)kvh * PF_D + d0);
             }
-            *reinterpret_cast<uint4*>(&k_lds[kk][d0]) = kraw;
+            *reinterpret_cast<uint4*>(&k_lds[kk][swz16(d0 / 8, kk) * 8]) = kraw;
             const bf16* v8 = reinterpret_cast<const bf16*>(&vraw);
 #pragma unroll
-            for (int j = 0; j < 8; ++j) vt_lds[d0 + j][kk] = v8[j];
+            for (int j = 0; j < 8; ++j)
+                vt_lds[d0 + j][swz8(kk >> 3, d0 + j) * 8 + (kk & 7)] = v8[j];
         }
         __syncthreads();
 
@@ -234,7 +246,7 @@ __global__ void attention_prefill_kernel(
 #pragma unroll
             for (int kc = 0; kc < 4; ++kc) {
                 const bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
-                    &k_lds[sub * 16 + lrow][kc * 32 + lk * 8]);
+                    &k_lds[sub * 16 + lrow][swz16(kc * 4 + lk, sub * 16 + lrow) * 8]);
                 s[sub] = mfma16(q_frag[kc], bfrag, s[sub]);
             }
         }
@@ -279,7 +291,8 @@ __global__ void attention_prefill_kernel(
         for (int sub = 0; sub < 4; ++sub)
 #pragma unroll
             for (int r = 0; r < 4; ++r)
-                p_lds[wave][lk * 4 + r][sub * 16 + lrow] = f2bf(p_val[sub][r]);
+                p_lds[wave][lk * 4 + r][swz8(sub * 2 + (lrow >> 3), lk * 4 + r) * 8 +
+                                        (lrow & 7)] = f2bf(p_val[sub][r]);
         // wave-synchronous LDS: the same wave reads it next, no barrier
 
         // ---- O += P V  (8 dim-subtiles, 2 key-chunks of 32) ----
@@ -288,9 +301,9 @@ __global__ void attention_prefill_kernel(
 #pragma unroll
             for (int kc = 0; kc < 2; ++kc) {
                 const bf16x8 pa = *reinterpret_cast<const bf16x8*>(
-                    &p_lds[wave][lrow][kc * 32 + lk * 8]);
+                    &p_lds[wave][lrow][swz8(kc * 4 + lk, lrow) * 8]);
                 const bf16x8 vb = *reinterpret_cast<const bf16x8*>(
-                    &vt_lds[dsub * 16 + lrow][kc * 32 + lk * 8]);
+                    &vt_lds[dsub * 16 + lrow][swz8(kc * 4 + lk, dsub * 16 + lrow) * 8]);
                 accO[dsub] = mfma16(pa, vb, accO[dsub]);
             }
         }
