@@ -6,13 +6,14 @@
   "use strict";
 
   const root = document.documentElement;
+  const themeSel = document.getElementById("themeSelector");
   function applyTheme(t) {
     root.dataset.theme = t;
     localStorage.setItem("gw-theme", t);
+    if (themeSel.value !== t) themeSel.value = t;
   }
   applyTheme(localStorage.getItem("gw-theme") || "dark");
-  document.getElementById("themeToggle").onclick = () =>
-    applyTheme(root.dataset.theme === "dark" ? "light" : "dark");
+  themeSel.addEventListener("change", () => applyTheme(themeSel.value));
 
   document.querySelectorAll(".tab").forEach((btn) => {
     btn.onclick = () => {
