@@ -42,6 +42,19 @@ logger = logging.getLogger(__name__)
 _req_counter = itertools.count()
 
 
+class _DropRow:
+    """Placeholder for sampled-but-not-delivered rows in logprob handling."""
+
+    class params:  # noqa: N801 — mimics SamplingParams statics
+        logprobs = False
+        top_logprobs = 0
+
+    state = "discarded"
+
+
+_DROP = _DropRow()
+
+
 @dataclass
 class SamplingParams:
     temperature: float = 0.0
@@ -54,6 +67,8 @@ class SamplingParams:
     seed: Optional[int] = None
     stop: List[str] = field(default_factory=list)
     ignore_eos: bool = False
+    logprobs: bool = False
+    top_logprobs: int = 0
 
     @classmethod
     def from_payload(cls, payload: dict, default_max_tokens: int = 256) -> "SamplingParams":
@@ -77,6 +92,8 @@ class SamplingParams:
             ),
             seed=payload.get("seed"),
             stop=[s for s in stop if isinstance(s, str)],
+            logprobs=bool(payload.get("logprobs")),
+            top_logprobs=max(0, min(20, int(payload.get("top_logprobs") or 0))),
         )
 
 
@@ -92,6 +109,8 @@ class EngineRequest:
         self.prompt_ids = list(prompt_ids)
         self.params = params
         self.out_ids: List[int] = []
+        # per emitted token, when params.logprobs: (logprob, [(id, lp), ...])
+        self.out_logprobs: List[tuple] = []
         self.text = ""  # incrementally decoded output (stop-string detection)
         self.block_table: List[int] = []
         self.state = "waiting"  # waiting | running | finished | failed
@@ -216,12 +235,24 @@ class LLMEngine:
         # next step device-side, host delivery runs one step late. Flushed
         # at every boundary: prefill, preemption, length cap, idle.
         self.async_sampling = self.device.type == "cuda"
-        self._pending: Optional[tuple] = None  # (reqs, n, event)
+        self._pending: Optional[tuple] = None  # (reqs, n, event, want_lp)
         self._pend_tokens_dev: Optional[torch.Tensor] = None
         self._pend_pinned: Optional[torch.Tensor] = None
+        self._lp_pinned = None
+        self._topv_pinned = None
+        self._topi_pinned = None
         if self.async_sampling:
             self._pend_pinned = torch.empty(
                 max_batch_size, dtype=torch.long, pin_memory=True
+            )
+            self._lp_pinned = torch.empty(
+                max_batch_size, dtype=torch.float32, pin_memory=True
+            )
+            self._topv_pinned = torch.empty(
+                max_batch_size, 20, dtype=torch.float32, pin_memory=True
+            )
+            self._topi_pinned = torch.empty(
+                max_batch_size, 20, dtype=torch.long, pin_memory=True
             )
 
         # host-side persistent block tables: one stable row per admitted
@@ -370,12 +401,19 @@ class LLMEngine:
     def _flush_pending(self) -> None:
         if self._pending is None:
             return
-        reqs, n, event = self._pending
+        reqs, n, event, want_lp = self._pending
         self._pending = None
         self.stats["flushes"] += 1
         event.synchronize()
         tokens = self._pend_pinned[:n].tolist()
         with self._lock:
+            if want_lp:
+                self._attach_logprobs(
+                    reqs,
+                    self._lp_pinned[:n].tolist(),
+                    self._topv_pinned[:n].tolist(),
+                    self._topi_pinned[:n].tolist(),
+                )
             self._deliver(reqs, tokens)
 
     # ---- the step ----
@@ -538,7 +576,8 @@ class LLMEngine:
         self.stats["mixed_steps" if nd else "prefill_steps"] += 1
         logits = self.model.forward(batch, self.kv.k_caches, self.kv.v_caches)
         sample_reqs = reqs + dec_reqs
-        tokens = self._sample(logits, sample_reqs)  # non-final prefill rows discarded
+        deliver_mask = [w[3] for w in work] + [True] * nd
+        tokens = self._sample(logits, sample_reqs, deliver_mask)
         dec_tokens = tokens[len(reqs):]
         self.stats["prefill_tokens"] += Tp
         self.stats["decode_tokens"] += nd
@@ -675,19 +714,28 @@ class LLMEngine:
         tokens_dev = self._sample_dev(
             logits, reqs, noise_pos=[len(r.out_ids) + inflight for r in reqs]
         )
+        lp = self._logprobs_dev(logits, tokens_dev, reqs)
         self.stats["decode_tokens"] += n
         if self.async_sampling:
             # previous pending was either flushed or belongs to these same
             # reqs and is already delivered-by-flush above when needed
             self._flush_pending()
             self._pend_pinned[:n].copy_(tokens_dev, non_blocking=True)
+            if lp is not None:
+                chosen, tv, ti = lp
+                k = tv.shape[1]
+                self._lp_pinned[:n].copy_(chosen, non_blocking=True)
+                self._topv_pinned[:n, :k].copy_(tv, non_blocking=True)
+                self._topi_pinned[:n, :k].copy_(ti, non_blocking=True)
             ev = torch.cuda.Event()
             ev.record()
-            self._pending = (reqs, n, ev)
+            self._pending = (reqs, n, ev, lp is not None)
             self._pend_tokens_dev = tokens_dev
             return n
         tokens = tokens_dev.tolist()
         with self._lock:
+            if lp is not None:
+                self._attach_logprobs(reqs, *(t.tolist() for t in lp))
             self._deliver(reqs, tokens)
         return n
 
@@ -762,8 +810,52 @@ class LLMEngine:
             noise = None
         return ops.sample(filtered, temps, noise)
 
-    def _sample(self, logits: torch.Tensor, reqs: List[EngineRequest]) -> List[int]:
-        return self._sample_dev(logits, reqs).tolist()
+    def _sample(
+        self,
+        logits: torch.Tensor,
+        reqs: List[EngineRequest],
+        deliver_mask: Optional[List[bool]] = None,
+    ) -> List[int]:
+        """Sample + attach logprobs for the rows whose token will actually
+        be delivered (non-final chunked-prefill rows sample a discarded
+        token and must not accumulate logprobs)."""
+        tokens = self._sample_dev(logits, reqs)
+        lp_reqs = (
+            reqs if deliver_mask is None
+            else [r if m else _DROP for r, m in zip(reqs, deliver_mask)]
+        )
+        lp = self._logprobs_dev(logits, tokens, lp_reqs)
+        toks = tokens.tolist()
+        if lp is not None:
+            self._attach_logprobs(lp_reqs, *(t.tolist() for t in lp))
+        return toks
+
+    def _logprobs_dev(self, logits, tokens, reqs):
+        """(chosen_lp [B], top_v [B,K], top_i [B,K]) device tensors, or
+        None when no request in the batch asked for logprobs. Raw-model
+        log-softmax (pre-temperature), OpenAI semantics."""
+        if not any(r.params.logprobs for r in reqs):
+            return None
+        lf = logits.float()
+        lse = torch.logsumexp(lf, dim=-1)
+        chosen = lf.gather(1, tokens.unsqueeze(1)).squeeze(1) - lse
+        k = max([r.params.top_logprobs for r in reqs] + [0])
+        if k > 0:
+            tv, ti = torch.topk(lf, k, dim=-1)
+            tv = tv - lse.unsqueeze(1)
+        else:
+            B = lf.shape[0]
+            tv = torch.empty(B, 0, device=lf.device)
+            ti = torch.empty(B, 0, dtype=torch.long, device=lf.device)
+        return chosen, tv, ti
+
+    def _attach_logprobs(self, reqs, chosen, top_v, top_i) -> None:
+        for i, r in enumerate(reqs):
+            if not r.params.logprobs or r.state != "running":
+                continue
+            k = r.params.top_logprobs
+            tops = list(zip(top_i[i][:k], top_v[i][:k])) if k else []
+            r.out_logprobs.append((chosen[i], tops))
 
     def _apply_topk_topp(self, logits: torch.Tensor, reqs: List[EngineRequest]) -> torch.Tensor:
         """In-place top-k/top-p filtering over the whole batch: the GPU

@@ -90,6 +90,28 @@ class _FailureState:
         return False
 
 
+def _format_logprobs(tokenizer, ids, lps) -> Dict[str, Any]:
+    """OpenAI chat `logprobs` object from delivered token ids and the
+    engine's per-token (logprob, [(id, lp), ...]) records."""
+    content = []
+    for tok_id, (lp, tops) in zip(ids, lps):
+        t = tokenizer.decode([int(tok_id)])
+        content.append({
+            "token": t,
+            "logprob": float(lp),
+            "bytes": list(t.encode("utf-8")),
+            "top_logprobs": [
+                {
+                    "token": tokenizer.decode([int(i)]),
+                    "logprob": float(v),
+                    "bytes": list(tokenizer.decode([int(i)]).encode("utf-8")),
+                }
+                for i, v in tops
+            ],
+        })
+    return {"content": content}
+
+
 class EngineRegistry:
     def __init__(self, settings: Optional[Settings] = None):
         self.settings = settings or Settings()
@@ -540,6 +562,13 @@ class EngineRegistry:
                     {
                         "index": 0,
                         "message": {"role": "assistant", "content": text},
+                        "logprobs": _format_logprobs(
+                            getattr(engine, "tokenizer", None) or self.tokenizer,
+                            req.out_ids[: len(req.out_logprobs)],
+                            req.out_logprobs,
+                        )
+                        if req.params.logprobs and getattr(req, "out_logprobs", None)
+                        else None,
                         "finish_reason": req.finish_reason or "stop",
                     }
                 ],
@@ -553,13 +582,15 @@ class EngineRegistry:
     ) -> StreamingResponse:
         tokenizer = getattr(engine, "tokenizer", None) or self.tokenizer
 
-        def chunk(delta: Dict[str, Any], finish: Optional[str] = None, usage=None) -> bytes:
+        def chunk(delta: Dict[str, Any], finish: Optional[str] = None, usage=None,
+                  logprobs=None) -> bytes:
             obj: Dict[str, Any] = {
                 "id": completion_id,
                 "object": "chat.completion.chunk",
                 "created": created,
                 "model": model_name,
-                "choices": [{"index": 0, "delta": delta, "finish_reason": finish}],
+                "choices": [{"index": 0, "delta": delta, "finish_reason": finish,
+                             "logprobs": logprobs}],
             }
             if usage is not None:
                 obj["usage"] = usage
@@ -567,6 +598,17 @@ class EngineRegistry:
 
         stops = req.params.stop
         holdback = max((len(s) for s in stops), default=1) - 1 if stops else 0
+
+        want_lp = bool(req.params.logprobs)
+        lp_pos = {"i": 0}
+
+        def chunk_logprobs(toks):
+            if not want_lp:
+                return None
+            i0 = lp_pos["i"]
+            lps = req.out_logprobs[i0 : i0 + len(toks)]
+            lp_pos["i"] = i0 + len(toks)
+            return _format_logprobs(tokenizer, toks[: len(lps)], lps)
 
         async def gen():
             pending = ""
@@ -592,6 +634,7 @@ class EngineRegistry:
                                 nxt = (k2, v2)
                                 break
                         text = tokenizer.decode(toks)
+                        lp_obj = chunk_logprobs(toks)
                         if stops and not stopped:
                             # hold back enough text to cleanly cut a stop
                             # string before it reaches the client
@@ -612,7 +655,7 @@ class EngineRegistry:
                                 yield chunk({"content": emit})
                         elif not stopped:
                             if text:
-                                yield chunk({"content": text})
+                                yield chunk({"content": text}, logprobs=lp_obj)
                         if nxt is not None:
                             kind, value = nxt
                             continue

@@ -108,7 +108,10 @@ def _worker_main(
 
     def on_token(req: EngineRequest, tok: int) -> None:
         if rank == 0:
-            outbox.put(("token", req.id, tok))
+            if req.params.logprobs and req.out_logprobs:
+                outbox.put(("tokenlp", req.id, (tok, req.out_logprobs[-1])))
+            else:
+                outbox.put(("token", req.id, tok))
 
     def on_finish(req: EngineRequest) -> None:
         reqs.pop(req.id, None)
@@ -326,7 +329,12 @@ class TPEngineClient:
                 req = self._reqs.get(rid)
             if req is None:
                 continue
-            if kind == "token":
+            if kind in ("token", "tokenlp"):
+                if kind == "tokenlp":
+                    value, lp = value
+                    if not hasattr(req, "out_logprobs"):
+                        req.out_logprobs = []
+                    req.out_logprobs.append(lp)
                 req.out_ids.append(int(value))
                 if req.first_token_time is None:
                     req.first_token_time = time.monotonic()
@@ -408,4 +416,6 @@ def _params_dict(params) -> dict:
         "seed": params.seed,
         "stop": list(params.stop),
         "ignore_eos": params.ignore_eos,
+        "logprobs": params.logprobs,
+        "top_logprobs": params.top_logprobs,
     }

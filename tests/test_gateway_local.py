@@ -358,3 +358,40 @@ def test_providers_reload_prunes_removed_engine(tmp_repo):
         assert client.post("/v1/config/providers", content=only_other.encode()).status_code == 200
         assert len(reg._engines) == 0  # old engine pruned
         assert not handle.thread.is_alive()
+
+
+def test_logprobs_nonstream(client):
+    r = chat(
+        client, "local/simple", max_tokens=5,
+        logprobs=True, top_logprobs=3,
+    )
+    assert r.status_code == 200
+    choice = r.json()["choices"][0]
+    lp = choice["logprobs"]
+    assert lp is not None and len(lp["content"]) == 5
+    for e in lp["content"]:
+        assert e["logprob"] <= 0.0
+        assert len(e["top_logprobs"]) == 3
+        # top-1 alternative is at least as likely as the chosen token
+        assert e["top_logprobs"][0]["logprob"] >= e["logprob"] - 1e-4
+        assert isinstance(e["bytes"], list)
+
+
+def test_logprobs_stream(client):
+    import json as _json
+
+    r = chat(
+        client, "local/simple", stream=True, max_tokens=4,
+        logprobs=True, top_logprobs=2,
+    )
+    assert r.status_code == 200
+    entries = []
+    for line in r.text.splitlines():
+        if not line.startswith("data: ") or line == "data: [DONE]":
+            continue
+        obj = _json.loads(line[6:])
+        lp = obj["choices"][0].get("logprobs")
+        if lp:
+            entries += lp["content"]
+    assert len(entries) == 4
+    assert all(len(e["top_logprobs"]) == 2 for e in entries)
