@@ -417,12 +417,218 @@ __global__ void gemm_m256r_kernel(
     }
 }
 
+// ---------------------------------------------------------------------------
+// Producer/consumer variant (v2): wave-specialized for the W-stream-heavy
+// shapes where both ring variants above hit the lockstep-straggler wall
+// (profiles/r02_gemm_m256_sweep.md).
+//
+// Why specialization: the hardware has ONE in-order vmcnt counter per
+// wave, so any single wave that both stages X (short-latency, barrier-
+// synced) and prefetches W (HBM-latency) has its W pipeline drained by
+// every X wait. Splitting the roles gives each wave a private counter:
+//
+// - waves 6,7 (loaders): stage the shared X tile ([256][64] bf16, XOR-
+//   swizzled) into a 3-slot LDS ring via ordinary loads + ds_write_b128
+//   (T14: issue one tile ahead, write after the barrier). Their waits
+//   touch only X loads.
+// - waves 0-5 (consumers): each owns a 16-column strip (block BN = 96).
+//   W fragments stream straight to VGPRs from the fragment-major twin in
+//   an 8-deep register ring (compile-time indices via 8x-unrolled k-loop)
+//   — ~8 KiB per wave in HBM flight, never drained: plain register loads
+//   survive __syncthreads (no LDS-DMA in this kernel), and each wave's
+//   counted waits see only its own stream.
+// - one __syncthreads per 64-deep k-tile; per k-step a consumer does
+//   16 ds_read_b128 (conflict-free, XOR layout) + 16 MFMA into a 16x1
+//   fragment accumulator (64 VGPRs).
+// ---------------------------------------------------------------------------
+
+#define PC_WAVES 8
+#define PC_CONS 6               // consumer waves -> BN = 96 columns
+#define PC_BN (PC_CONS * 16)
+#define PC_BK 64                // k-depth per staged X tile (2 k-steps)
+#define PC_NSLOT 3              // X LDS ring slots
+#define PC_WD 8                 // W register-ring depth in 32-deep k-steps
+
+template <bool SPLITK>
+__launch_bounds__(PC_WAVES * WAVE_SIZE)
+__global__ void gemm_m256pc_kernel(
+    bf16* __restrict__ y,        // [M, N] (!SPLITK)
+    float* __restrict__ yw,      // [nsk, M, N] fp32 slabs (SPLITK)
+    const bf16* __restrict__ x,  // [M, K]
+    const bf16* __restrict__ w,  // fragment-major [K/32][N/16][64][8]
+    int M, int N, int K, int nsk) {
+    constexpr int XB = 256 * PC_BK * 2;  // one X slot: 32 KiB
+    __shared__ __attribute__((aligned(16))) char smem[PC_NSLOT * XB];
+
+    const int lane = threadIdx.x & (WAVE_SIZE - 1);
+    const int wave = threadIdx.x >> 6;
+    const int n0 = blockIdx.x * PC_BN;
+    const int n16 = N / 16;
+
+    const int ktiles = K / PC_BK;
+    const int kt_per = SPLITK ? (ktiles + nsk - 1) / nsk : ktiles;
+    const int kt0 = SPLITK ? blockIdx.y * kt_per : 0;
+    const int ntiles = min(ktiles - kt0, kt_per) > 0 ? min(ktiles - kt0, kt_per) : 0;
+    const int nk = ntiles * 2;  // 32-deep k-steps in this slice
+
+    if (wave >= PC_CONS) {
+        // ---- loader: rows [128*(wave-6), +128), 16 KiB per tile ----
+        // unit i covers 8 rows x 64 cols; lane -> row base + l/8, chunk
+        // l%8; the XOR swizzle is applied at the ds_write address.
+        const int lw = wave - PC_CONS;  // 0 or 1
+        const int xrow = lw * 128 + (lane >> 3);  // + 8*i per unit
+        const int xchunk = lane & 7;
+        const bf16* xsrc[16];
+#pragma unroll
+        for (int i = 0; i < 16; ++i) {
+            const int r = xrow + i * 8;
+            xsrc[i] = x + (size_t)(r < M ? r : 0) * K + xchunk * 8;
+        }
+        const int xwoff = (xrow)*128 + ((xchunk ^ (xrow & 7)) << 4);
+        bf16x8 xr[16];
+#define PC_LOAD(T)                                                             \
+    do {                                                                       \
+        const size_t kof__ = (size_t)(kt0 + (T)) * PC_BK;                      \
+        _Pragma("unroll") for (int i = 0; i < 16; ++i) xr[i] =                 \
+            *(const __attribute__((address_space(1))) bf16x8*)(xsrc[i] +       \
+                                                               kof__);         \
+    } while (0)
+#define PC_WRITE(T)                                                            \
+    do {                                                                       \
+        char* buf__ = smem + ((T) % PC_NSLOT) * XB;                            \
+        _Pragma("unroll") for (int i = 0; i < 16; ++i)                         \
+            *(__attribute__((address_space(3))) bf16x8*)(                      \
+                (__attribute__((address_space(3))) char*)buf__ + xwoff +       \
+                i * 8 * 128) = xr[i];                                          \
+    } while (0)
+        if (ntiles > 0) {
+            PC_LOAD(0);
+            PC_WRITE(0);
+            if (ntiles > 1) PC_LOAD(1);
+            __syncthreads();
+            for (int t = 0; t < ntiles; ++t) {
+                if (t + 1 < ntiles) {
+                    PC_WRITE(t + 1);
+                    if (t + 2 < ntiles) PC_LOAD(t + 2);
+                }
+                __syncthreads();
+            }
+        }
+#undef PC_LOAD
+#undef PC_WRITE
+        return;  // loaders take no part in the epilogue
+    }
+
+    // ---- consumer wave: columns [n0 + wave*16, +16) ----
+    // W fragment stream: one 1-KiB lane-linear fragment per 32-deep
+    // k-step, clamped to the last valid fragment for tail/padding reads
+    // (results discarded).
+    const int nfrag = min(n0 / 16 + wave, n16 - 1);
+    const bf16* wbase = w + (size_t)lane * 8;
+    auto wsrc = [&](int ks) {  // ks = k-step index within this slice
+        const int g = kt0 * 2 + min(ks, nk > 0 ? nk - 1 : 0);
+        return (const __attribute__((address_space(1))) bf16x8*)(
+            wbase + ((size_t)g * n16 + nfrag) * 512);
+    };
+
+    f32x4 acc[16];
+#pragma unroll
+    for (int m = 0; m < 16; ++m) acc[m] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+    const int arow_l = lane & 15;  // A-frag row within an m-tile
+    const int alk = lane >> 4;
+
+    if (ntiles > 0) {
+        bf16x8 wring[PC_WD];
+#pragma unroll
+        for (int d = 0; d < PC_WD; ++d) wring[d] = *wsrc(d);
+        __syncthreads();  // X slot 0 staged
+
+        // consume k-steps in pairs (one X tile); ring indices static via
+        // the %PC_WD structure of the 4-tile unrolled main loop
+        int t = 0;
+        for (; t + 4 <= ntiles && 2 * (t + 4) <= nk; t += 4) {
+#pragma unroll
+            for (int tt = 0; tt < 4; ++tt) {
+                const char* buf = smem + ((t + tt) % PC_NSLOT) * XB;
+#pragma unroll
+                for (int ks = 0; ks < 2; ++ks) {
+                    const int kabs = 2 * (t + tt) + ks;
+                    const int ridx = kabs & (PC_WD - 1);
+                    const bf16x8 bfrag = wring[ridx];
+                    __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+                    for (int m = 0; m < 16; ++m) {
+                        const int row = m * 16 + arow_l;
+                        const int chunk = (ks * 4 + alk) ^ (row & 7);
+                        const bf16x8 afrag =
+                            *(const __attribute__((address_space(3))) bf16x8*)(
+                                (const __attribute__((address_space(3))) char*)
+                                    buf +
+                                row * 128 + chunk * 16);
+                        acc[m] = gm_mfma(afrag, bfrag, acc[m]);
+                    }
+                    __builtin_amdgcn_sched_barrier(0);
+                    wring[ridx] = *wsrc(kabs + PC_WD);  // refill the slot
+                }
+                __syncthreads();
+            }
+        }
+        // tail tiles (ring indices still static: kabs & 7)
+        for (; t < ntiles; ++t) {
+            const char* buf = smem + (t % PC_NSLOT) * XB;
+#pragma unroll
+            for (int ks = 0; ks < 2; ++ks) {
+                const int kabs = 2 * t + ks;
+                bf16x8 bfrag;
+#pragma unroll
+                for (int d = 0; d < PC_WD; ++d)
+                    if (d == (kabs & (PC_WD - 1))) bfrag = wring[d];
+#pragma unroll
+                for (int m = 0; m < 16; ++m) {
+                    const int row = m * 16 + arow_l;
+                    const int chunk = (ks * 4 + alk) ^ (row & 7);
+                    const bf16x8 afrag =
+                        *(const __attribute__((address_space(3))) bf16x8*)(
+                            (const __attribute__((address_space(3))) char*)buf +
+                            row * 128 + chunk * 16);
+                    acc[m] = gm_mfma(afrag, bfrag, acc[m]);
+                }
+            }
+            __syncthreads();
+        }
+    }
+
+    // ---- epilogue ----
+    const int col = n0 + wave * 16 + (lane & 15);
+    if (col >= N) return;
+    if (SPLITK) {
+        float* slab = yw + (size_t)blockIdx.y * M * N;
+#pragma unroll
+        for (int m = 0; m < 16; ++m)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int row = m * 16 + (lane >> 4) * 4 + r;
+                if (row < M) slab[(size_t)row * N + col] = acc[m][r];
+            }
+    } else {
+#pragma unroll
+        for (int m = 0; m < 16; ++m)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int row = m * 16 + (lane >> 4) * 4 + r;
+                if (row < M) y[(size_t)row * N + col] = f2bf(acc[m][r]);
+            }
+    }
+}
+
 extern "C" hipError_t launch_gemm_reduce(void*, const float*, int64_t, int,
                                          hipStream_t);  // gemm_skinny.hip
 
 // nf: 4 (BN=64) or 8 (BN=128). M <= 256; N % (16*nf) == 0; K % 64 == 0.
 // variant: 0 = glds-staged (DMA pipeline, depth per `pipe`),
-//          1 = register-staged T14 (2 buffers, loads one tile ahead).
+//          1 = register-staged T14 (2 buffers, loads one tile ahead),
+//          2 = producer/consumer wave-specialized (BN=96; nf ignored).
 // pipe (variant 0 only): 0=(BK64,NBUF3) 1=(BK64,NBUF4,nf4)
 //          2=(BK32,NBUF4,nf8) 3=(BK32,NBUF6,nf8) — deeper rings keep more
 //          HBM latency covered on the W stream; 4=(BK64,NBUF2,nf4) and
@@ -432,9 +638,27 @@ extern "C" hipError_t launch_gemm_m256(
     void* y, float* workspace, const void* x, const void* w, int M, int N,
     int K, int nsk, int nf, int variant, int pipe, hipStream_t stream) {
     if (M <= 0 || M > 256) return hipErrorInvalidValue;
-    if (nf != 4 && nf != 8) return hipErrorInvalidValue;
-    if ((N % (16 * nf)) != 0 || (K % GM_BK) != 0) return hipErrorInvalidValue;
     if (nsk < 1 || (nsk > 1 && workspace == nullptr)) return hipErrorInvalidValue;
+    if ((K % GM_BK) != 0) return hipErrorInvalidValue;
+    if (variant == 2) {
+        if ((N % 16) != 0) return hipErrorInvalidValue;
+        dim3 grid((N + PC_BN - 1) / PC_BN, nsk);
+        dim3 block(PC_WAVES * WAVE_SIZE);
+        if (nsk > 1)
+            gemm_m256pc_kernel<true><<<grid, block, 0, stream>>>(
+                (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,
+                nsk);
+        else
+            gemm_m256pc_kernel<false><<<grid, block, 0, stream>>>(
+                (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,
+                nsk);
+        HIP_CHECK_LAST();
+        if (nsk > 1)
+            return launch_gemm_reduce(y, workspace, (int64_t)M * N, nsk, stream);
+        return hipSuccess;
+    }
+    if (nf != 4 && nf != 8) return hipErrorInvalidValue;
+    if ((N % (16 * nf)) != 0) return hipErrorInvalidValue;
     if (variant == 0) {
         if ((pipe == 1 || pipe == 4 || pipe == 5) && nf != 4)
             return hipErrorInvalidValue;

@@ -292,7 +292,7 @@ def test_gemm_m256(M, N, K, nf):
     w = (torch.randn(N, K, device=DEV) * 0.02).bfloat16()
     wf = ops.swizzle_weight_frag(w)
     ref = x.float() @ w.float().T
-    for variant in (0, 1):  # glds-staged and register-staged forms
+    for variant in (0, 1, 2):  # glds / register-staged / producer-consumer
         got = ops.gemm_m256(x, wf, nf=nf, variant=variant)
         err = (got.float() - ref).abs().max().item()
         scale = ref.abs().max().item() + 1e-3

@@ -60,6 +60,11 @@ def main():
                 for pipe in pipes:
                     combos.append((0, nf, pipe, nsk))
                 combos.append((1, nf, 0, nsk))
+        # producer/consumer variant (BN=96, nf ignored)
+        nsks2 = sorted({1, max(1, -(-256 // max(1, N // 96)))})
+        for nsk in nsks2:
+            if nsk <= max(1, (K // 64) // 2) and nsk <= 8:
+                combos.append((2, 4, 0, nsk))
         best = (None, 1e18)
         for (v, nf, pipe, nsk) in combos:
             try:
