@@ -449,7 +449,10 @@ __global__ void gemm_m256r_kernel(
 #define PC_NSLOT 3              // X LDS ring slots
 #define PC_WD 8                 // W register-ring depth in 32-deep k-steps
 
-template <bool SPLITK>
+// ABL: 0 = normal, 1 = loader-only (consumers only hit the barriers),
+// 2 = consumer-only (loaders only hit the barriers) — the ablation pair
+// that isolates which role is the critical path (guide §5 mistake 8).
+template <bool SPLITK, int ABL = 0>
 __launch_bounds__(PC_WAVES * WAVE_SIZE)
 __global__ void gemm_m256pc_kernel(
     bf16* __restrict__ y,        // [M, N] (!SPLITK)
@@ -502,12 +505,14 @@ __global__ void gemm_m256pc_kernel(
                 i * 8 * 128) = xr[i];                                          \
     } while (0)
         if (ntiles > 0) {
-            PC_LOAD(0);
-            PC_WRITE(0);
-            if (ntiles > 1) PC_LOAD(1);
+            if (ABL != 2) {
+                PC_LOAD(0);
+                PC_WRITE(0);
+                if (ntiles > 1) PC_LOAD(1);
+            }
             __syncthreads();
             for (int t = 0; t < ntiles; ++t) {
-                if (t + 1 < ntiles) {
+                if (ABL != 2 && t + 1 < ntiles) {
                     PC_WRITE(t + 1);
                     if (t + 2 < ntiles) PC_LOAD(t + 2);
                 }
@@ -538,7 +543,10 @@ __global__ void gemm_m256pc_kernel(
     const int arow_l = lane & 15;  // A-frag row within an m-tile
     const int alk = lane >> 4;
 
-    if (ntiles > 0) {
+    if (ntiles > 0 && ABL == 1) {
+        __syncthreads();
+        for (int t = 0; t < ntiles; ++t) __syncthreads();
+    } else if (ntiles > 0) {
         bf16x8 wring[PC_WD];
 #pragma unroll
         for (int d = 0; d < PC_WD; ++d) wring[d] = *wsrc(d);
@@ -644,14 +652,10 @@ extern "C" hipError_t launch_gemm_m256(
         if ((N % 16) != 0) return hipErrorInvalidValue;
         dim3 grid((N + PC_BN - 1) / PC_BN, nsk);
         dim3 block(PC_WAVES * WAVE_SIZE);
-        if (nsk > 1)
-            gemm_m256pc_kernel<true><<<grid, block, 0, stream>>>(
-                (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,
-                nsk);
-        else
-            gemm_m256pc_kernel<false><<<grid, block, 0, stream>>>(
-                (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,
-                nsk);
+#define PC_L(SPLIT)                                                                do {                                                                               if (pipe == 1)                                                                     gemm_m256pc_kernel<SPLIT, 1><<<grid, block, 0, stream>>>(                          (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,                  nsk);                                                                  else if (pipe == 2)                                                                gemm_m256pc_kernel<SPLIT, 2><<<grid, block, 0, stream>>>(                          (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,                  nsk);                                                                  else                                                                               gemm_m256pc_kernel<SPLIT, 0><<<grid, block, 0, stream>>>(                          (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,                  nsk);                                                              } while (0)
+        if (nsk > 1) PC_L(true);
+        else PC_L(false);
+#undef PC_L
         HIP_CHECK_LAST();
         if (nsk > 1)
             return launch_gemm_reduce(y, workspace, (int64_t)M * N, nsk, stream);
