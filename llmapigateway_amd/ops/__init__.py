@@ -370,7 +370,7 @@ def gemm_m256(
     pipe selects the DMA ring geometry for the glds variant:
     0=(BK64,N3) 1=(BK64,N4,nf4) 2=(BK32,N4,nf8) 3=(BK32,N6,nf8)."""
     M, K = x.shape
-    N = w_frag.shape[1] * 16
+    N = w_frag.shape[0] * 16
     if nf is None:
         if N % 128 == 0 and M > 64:
             nf = 8  # deep BK32 ring needs BN=128
