@@ -297,23 +297,19 @@ def swizzle_weight(w: torch.Tensor) -> torch.Tensor:
 
 
 def swizzle_weight_frag(w: torch.Tensor) -> torch.Tensor:
-    """[N, K] -> strip-major fragment layout [N/16, K/32, 64, 8] for
-    gemm_m256.
+    """[N, K] -> fragment-major [K/32, N/16, 64, 8] for gemm_m256.
 
-    Element [n16][k32][lane][e] = W[n16*16 + (lane&15)][k32*32 + (lane>>4)*8
+    Element [k32][n16][lane][e] = W[n16*16 + (lane&15)][k32*32 + (lane>>4)*8
     + e]: each 1 KiB row is exactly one wave's mfma_f32_16x16x32_bf16
-    B-fragment in lane order (staging is a straight contiguous copy and
-    ds_read_b128 is lane-linear, conflict-free per the gfx950 b128 lane
-    groups), and a 16-column strip's fragments are CONTIGUOUS along K —
-    the k-major alternative put consecutive k-fragments ~1 MB apart, which
-    broke the W stream into scattered 1 KiB HBM bursts and capped every
-    kernel variant near 2.5 TB/s."""
+    B-fragment in lane order, so the kernel's global_load_lds staging is a
+    straight contiguous copy and its ds_read_b128 is lane-linear
+    (conflict-free per the gfx950 b128 lane groups)."""
     N, K = w.shape
     assert K % 64 == 0 and N % 64 == 0
     return (
         w.view(N // 16, 16, K // 32, 4, 8)
-        .permute(0, 2, 3, 1, 4)
-        .reshape(N // 16, K // 32, 64, 8)
+        .permute(2, 0, 3, 1, 4)
+        .reshape(K // 32, N // 16, 64, 8)
         .contiguous()
     )
 
@@ -370,7 +366,7 @@ def gemm_m256(
     pipe selects the DMA ring geometry for the glds variant:
     0=(BK64,N3) 1=(BK64,N4,nf4) 2=(BK32,N4,nf8) 3=(BK32,N6,nf8)."""
     M, K = x.shape
-    N = w_frag.shape[0] * 16
+    N = w_frag.shape[1] * 16
     if nf is None:
         if N % 128 == 0 and M > 64:
             nf = 8  # deep BK32 ring needs BN=128

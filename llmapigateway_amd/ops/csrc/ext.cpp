@@ -287,11 +287,11 @@ void gemm_m256(torch::Tensor y, torch::Tensor x, torch::Tensor w_frag,
     TORCH_CHECK(x.is_contiguous() && w_frag.is_contiguous() && y.is_contiguous());
     TORCH_CHECK(x.dim() == 2 && y.dim() == 2);
     const int M = x.size(0), K = x.size(1);
-    // strip-major fragment twin: [N/16, K/32, 64, 8]
-    TORCH_CHECK(w_frag.dim() == 4 && w_frag.size(1) == K / 32 &&
+    // fragment-major twin: [K/32, N/16, 64, 8]
+    TORCH_CHECK(w_frag.dim() == 4 && w_frag.size(0) == K / 32 &&
                 w_frag.size(2) == 64 && w_frag.size(3) == 8,
-                "w_frag must be the strip-major [N/16, K/32, 64, 8] twin");
-    const int N = w_frag.size(0) * 16;
+                "w_frag must be the fragment-major [K/32, N/16, 64, 8] twin");
+    const int N = w_frag.size(1) * 16;
     TORCH_CHECK(y.size(0) == M && y.size(1) == N);
     float* ws = nullptr;
     if (workspace.has_value() && workspace->defined()) {

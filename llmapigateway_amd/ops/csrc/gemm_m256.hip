@@ -85,7 +85,7 @@ __global__ void gemm_m256_kernel(
     bf16* __restrict__ y,        // [M, N] (!SPLITK)
     float* __restrict__ yw,      // [nsk, M, N] fp32 slabs (SPLITK)
     const bf16* __restrict__ x,  // [M, K]
-    const bf16* __restrict__ w,  // strip-major [N/16][K/32][64][8]
+    const bf16* __restrict__ w,  // fragment-major [K/32][N/16][64][8]
     int M, int N, int K, int nsk) {
     constexpr int BM = MW * 32;
     constexpr int BN = NF * 16;
@@ -111,8 +111,6 @@ __global__ void gemm_m256_kernel(
     const int wave = threadIdx.x >> 6;
     const int n0 = blockIdx.x * BN;
     const int n16 = N / 16;
-    const int k32 = K / 32;
-    (void)n16;
 
     const int ktiles = K / BK;
     const int kt_per = SPLITK ? (ktiles + nsk - 1) / nsk : ktiles;
@@ -160,9 +158,8 @@ __global__ void gemm_m256_kernel(
                       xdst0 + i * 1024);                                       \
         _Pragma("unroll") for (int j = 0; j < WG_HI; ++j) {                    \
             if (wfi[j] < WU) {                                                 \
-                const size_t gfi__ =                                           \
-                    (size_t)(n0 / 16 + wfi[j] % NF) * k32 + KS * kt__ +        \
-                    wfi[j] / NF;                                               \
+                const size_t gfi__ = (size_t)(KS * kt__ + wfi[j] / NF) * n16 + \
+                                     n0 / 16 + wfi[j] % NF;                    \
                 gm_glds16(w + gfi__ * 512 + lane * 8, buf__,                   \
                           XB + wfi[j] * 1024);                                 \
             }                                                                  \
@@ -268,7 +265,7 @@ __launch_bounds__(MW * WAVE_SIZE)
 __global__ void gemm_m256r_kernel(
     bf16* __restrict__ y, float* __restrict__ yw,
     const bf16* __restrict__ x,  // [M, K]
-    const bf16* __restrict__ w,  // strip-major [N/16][K/32][64][8]
+    const bf16* __restrict__ w,  // fragment-major [K/32][N/16][64][8]
     int M, int N, int K, int nsk) {
     constexpr int BM = MW * 32;
     constexpr int BN = NF * 16;
@@ -285,8 +282,6 @@ __global__ void gemm_m256r_kernel(
     const int wave = threadIdx.x >> 6;
     const int n0 = blockIdx.x * BN;
     const int n16 = N / 16;
-    const int k32 = K / 32;
-    (void)n16;
 
     const int ktiles = K / GM_BK;
     const int kt_per = SPLITK ? (ktiles + nsk - 1) / nsk : ktiles;
@@ -325,8 +320,8 @@ __global__ void gemm_m256r_kernel(
             *(const __attribute__((address_space(1))) bf16x8*)(                \
                 xsrc[i] + (size_t)(kt__)*GM_BK);                               \
         _Pragma("unroll") for (int j = 0; j < WG_W; ++j) {                     \
-            const size_t gfi__ = (size_t)(n0 / 16 + wfi[j] % NF) * k32 +       \
-                                 2 * kt__ + wfi[j] / NF;                       \
+            const size_t gfi__ = (size_t)(2 * kt__ + wfi[j] / NF) * n16 +      \
+                                 n0 / 16 + wfi[j] % NF;                        \
             wr[j] = *(const __attribute__((address_space(1))) bf16x8*)(        \
                 w + gfi__ * 512 + lane * 8);                                   \
         }                                                                      \
@@ -463,7 +458,7 @@ __global__ void gemm_m256pc_kernel(
     bf16* __restrict__ y,        // [M, N] (!SPLITK)
     float* __restrict__ yw,      // [nsk, M, N] fp32 slabs (SPLITK)
     const bf16* __restrict__ x,  // [M, K]
-    const bf16* __restrict__ w,  // strip-major [N/16][K/32][64][8]
+    const bf16* __restrict__ w,  // fragment-major [K/32][N/16][64][8]
     int M, int N, int K, int nsk) {
     constexpr int XB = 256 * PC_BK * 2;  // one X slot: 32 KiB
     __shared__ __attribute__((aligned(16))) char smem[PC_NSLOT * XB];
@@ -544,12 +539,11 @@ __global__ void gemm_m256pc_kernel(
     // k-step, clamped to the last valid fragment for tail/padding reads
     // (results discarded).
     const int nfrag = min(n0 / 16 + wave, n16 - 1);
-    const int k32 = K / 32;
-    const bf16* wbase = w + ((size_t)nfrag * k32) * 512 + (size_t)lane * 8;
+    const bf16* wbase = w + (size_t)lane * 8;
     auto wsrc = [&](int ks) {  // ks = k-step index within this slice
         const int g = kt0 * 2 + min(ks, nk > 0 ? nk - 1 : 0);
         return (const __attribute__((address_space(1))) bf16x8*)(
-            wbase + (size_t)g * 512);
+            wbase + ((size_t)g * n16 + nfrag) * 512);
     };
 
     f32x4 acc[16];
