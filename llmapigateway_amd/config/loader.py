@@ -48,6 +48,7 @@ class EngineSpec(BaseModel):
     dtype: str = "bfloat16"
     max_batch_size: Optional[int] = None
     kv_block_size: Optional[int] = None
+    kv_dtype: str = "auto"               # "auto" (= compute dtype) | "fp8"
     fail_rate: float = 0.0               # failure injection: P(request fails)
     fail_requests: Optional[int] = None  # failure injection: fail first N requests
 

@@ -166,6 +166,7 @@ class LLMEngine:
         tp_size: int = 1,
         prefix_caching: bool = False,
         prefill_budget: int = 8192,
+        kv_dtype: str = "auto",
         tokenizer: Optional[object] = None,
         admit_min_batch: Optional[int] = None,
         admit_max_wait: Optional[float] = None,
@@ -194,13 +195,16 @@ class LLMEngine:
             tp_group=tp_group, tp_rank=tp_rank, tp_size=tp_size,
             full_config=full_config,
         )
+        if kv_dtype == "auto":
+            kv_dtype = os.environ.get("LLMAPI_KV_DTYPE", "auto")
         if num_blocks is None:
             num_blocks = PagedKVCache.fit_num_blocks(
-                config, block_size, self.device, dtype, hbm_fraction
+                config, block_size, self.device, dtype, hbm_fraction,
+                kv_dtype=kv_dtype,
             )
         self.kv = PagedKVCache(
             config, num_blocks, block_size, self.device, dtype,
-            prefix_caching=prefix_caching,
+            prefix_caching=prefix_caching, kv_dtype=kv_dtype,
         )
         self.prefix_caching = prefix_caching
         if admit_min_batch is None:
@@ -228,6 +232,8 @@ class LLMEngine:
                 self.kv.v_caches,
                 max_batch=max_batch_size,
                 max_blocks=(self.max_model_len + block_size - 1) // block_size,
+                k_scales=self.kv.k_scales if self.kv.fp8 else None,
+                v_scales=self.kv.v_scales if self.kv.fp8 else None,
             )
 
         # deferred sampling (GPU): the sampled-token fetch of decode step s
@@ -574,7 +580,11 @@ class LLMEngine:
             logits_indices=torch.from_numpy(logits_idx).to(device),
         )
         self.stats["mixed_steps" if nd else "prefill_steps"] += 1
-        logits = self.model.forward(batch, self.kv.k_caches, self.kv.v_caches)
+        logits = self.model.forward(
+            batch, self.kv.k_caches, self.kv.v_caches,
+            k_scales=self.kv.k_scales if self.kv.fp8 else None,
+            v_scales=self.kv.v_scales if self.kv.fp8 else None,
+        )
         sample_reqs = reqs + dec_reqs
         deliver_mask = [w[3] for w in work] + [True] * nd
         tokens = self._sample(logits, sample_reqs, deliver_mask)
@@ -709,7 +719,11 @@ class LLMEngine:
                 context_lens=torch.from_numpy(ctx).to(device),
                 logits_indices=None,
             )
-            logits = self.model.forward(batch, self.kv.k_caches, self.kv.v_caches)
+            logits = self.model.forward(
+                batch, self.kv.k_caches, self.kv.v_caches,
+                k_scales=self.kv.k_scales if self.kv.fp8 else None,
+                v_scales=self.kv.v_scales if self.kv.fp8 else None,
+            )
 
         tokens_dev = self._sample_dev(
             logits, reqs, noise_pos=[len(r.out_ids) + inflight for r in reqs]

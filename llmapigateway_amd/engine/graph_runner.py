@@ -26,10 +26,13 @@ logger = logging.getLogger(__name__)
 
 
 class DecodeGraphRunner:
-    def __init__(self, model, k_caches, v_caches, max_batch: int, max_blocks: int):
+    def __init__(self, model, k_caches, v_caches, max_batch: int, max_blocks: int,
+                 k_scales=None, v_scales=None):
         self.model = model
         self.k_caches = k_caches
         self.v_caches = v_caches
+        self.k_scales = k_scales
+        self.v_scales = v_scales
         self.device = model.device
         self.max_batch = max_batch
         self.max_blocks = max_blocks
@@ -75,13 +78,15 @@ class DecodeGraphRunner:
         s.wait_stream(torch.cuda.current_stream(self.device))
         with torch.cuda.stream(s):
             for _ in range(2):
-                self.model.forward(batch, self.k_caches, self.v_caches)
+                self.model.forward(batch, self.k_caches, self.v_caches,
+                                   k_scales=self.k_scales, v_scales=self.v_scales)
         torch.cuda.current_stream(self.device).wait_stream(s)
         torch.cuda.synchronize(self.device)
 
         graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(graph, pool=self.pool):
-            logits = self.model.forward(batch, self.k_caches, self.v_caches)
+            logits = self.model.forward(batch, self.k_caches, self.v_caches,
+                                        k_scales=self.k_scales, v_scales=self.v_scales)
         if self.pool is None:
             self.pool = graph.pool()  # share the memory pool across buckets
         self.graphs[bucket] = graph

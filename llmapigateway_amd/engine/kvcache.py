@@ -200,21 +200,47 @@ class PagedKVCache:
         device: torch.device | str,
         dtype: torch.dtype = torch.bfloat16,
         prefix_caching: bool = False,
+        kv_dtype: str = "auto",
     ):
+        """kv_dtype: "auto" (= compute dtype) or "fp8" — e4m3 bytes with
+        per-(token, head) fp32 row scales: half the decode KV traffic and
+        ~2x the resident-token capacity on the same HBM budget."""
         self.config = config
         self.block_size = block_size
         self.num_blocks = num_blocks
+        self.fp8 = kv_dtype == "fp8"
         shape = (num_blocks, config.num_kv_heads, block_size, config.head_dim)
+        store = torch.uint8 if self.fp8 else dtype
         self.k_caches = [
-            torch.zeros(shape, dtype=dtype, device=device) for _ in range(config.num_layers)
+            torch.zeros(shape, dtype=store, device=device) for _ in range(config.num_layers)
         ]
         self.v_caches = [
-            torch.zeros(shape, dtype=dtype, device=device) for _ in range(config.num_layers)
+            torch.zeros(shape, dtype=store, device=device) for _ in range(config.num_layers)
         ]
+        if self.fp8:
+            sshape = (num_blocks, config.num_kv_heads, block_size)
+            self.k_scales = [
+                torch.ones(sshape, dtype=torch.float32, device=device)
+                for _ in range(config.num_layers)
+            ]
+            self.v_scales = [
+                torch.ones(sshape, dtype=torch.float32, device=device)
+                for _ in range(config.num_layers)
+            ]
+        else:
+            self.k_scales = [None] * config.num_layers
+            self.v_scales = [None] * config.num_layers
         self.manager = BlockManager(num_blocks, block_size, prefix_caching=prefix_caching)
 
     @staticmethod
-    def block_bytes(config: ModelConfig, block_size: int, dtype: torch.dtype) -> int:
+    def block_bytes(
+        config: ModelConfig, block_size: int, dtype: torch.dtype,
+        kv_dtype: str = "auto",
+    ) -> int:
+        if kv_dtype == "fp8":
+            # 1 byte per element + a 4-byte scale per (token, head) row
+            per_row = config.head_dim + 4
+            return 2 * config.num_layers * config.num_kv_heads * block_size * per_row
         elem = torch.empty(0, dtype=dtype).element_size()
         return 2 * config.num_layers * config.num_kv_heads * block_size * config.head_dim * elem
 
@@ -227,8 +253,9 @@ class PagedKVCache:
         dtype: torch.dtype,
         hbm_fraction: float = 0.90,
         max_blocks: Optional[int] = None,
+        kv_dtype: str = "auto",
     ) -> int:
-        per_block = cls.block_bytes(config, block_size, dtype)
+        per_block = cls.block_bytes(config, block_size, dtype, kv_dtype)
         if device.type == "cuda":
             free, _total = torch.cuda.mem_get_info(device)
             budget = int(free * hbm_fraction)

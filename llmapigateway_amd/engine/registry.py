@@ -129,6 +129,8 @@ class EngineRegistry:
             key += f"/b{spec.max_batch_size}"
         if spec.kv_block_size:
             key += f"/k{spec.kv_block_size}"
+        if spec.kv_dtype not in ("auto", None):
+            key += f"/kv{spec.kv_dtype}"
         if spec.dtype not in ("bfloat16", None):
             key += f"/{spec.dtype}"
         return key
@@ -185,6 +187,7 @@ class EngineRegistry:
                 num_blocks=None if device.startswith("cuda") else 256,
                 prefix_caching=self.settings.engine_prefix_caching,
                 use_hipgraph=None if self.settings.engine_use_hipgraph else False,
+                kv_dtype=spec.kv_dtype,
             )
             if engine.graph_runner is not None:
                 logger.info("pre-capturing decode hipGraphs for %s", key)

@@ -180,9 +180,14 @@ class LlamaModel:
         batch: ForwardBatch,
         k_caches: List[torch.Tensor],
         v_caches: List[torch.Tensor],
+        k_scales: "Optional[List[torch.Tensor]]" = None,
+        v_scales: "Optional[List[torch.Tensor]]" = None,
     ) -> torch.Tensor:
-        """Returns logits [B, vocab] at batch.logits_indices."""
+        """Returns logits [B, vocab] at batch.logits_indices. With
+        k_scales/v_scales the caches are fp8 e4m3 (per-row scales)."""
         c = self.config
+        ksc = k_scales if k_scales is not None else [None] * len(self.layers)
+        vsc = v_scales if v_scales is not None else [None] * len(self.layers)
         h = F.embedding(batch.token_ids, self.embed)
         residual = h  # placeholder; layer 0 sets the real residual stream
         T = h.shape[0]
@@ -201,6 +206,7 @@ class LlamaModel:
             ops.rope_and_kv_write(
                 q, k, v, k_caches[i], v_caches[i],
                 batch.positions, self.cos_sin, batch.slot_mapping,
+                k_scale=ksc[i], v_scale=vsc[i],
             )
 
             if batch.kind == "prefill":
@@ -211,6 +217,8 @@ class LlamaModel:
                     v_cache=v_caches[i] if batch.cached_lens is not None else None,
                     block_tables=batch.block_tables,
                     cached_lens=batch.cached_lens,
+                    k_scale=ksc[i] if batch.cached_lens is not None else None,
+                    v_scale=vsc[i] if batch.cached_lens is not None else None,
                 )
             elif batch.kind == "mixed":
                 Tp = batch.n_prefill_tokens
@@ -221,15 +229,20 @@ class LlamaModel:
                     v_cache=v_caches[i] if batch.cached_lens is not None else None,
                     block_tables=batch.block_tables,
                     cached_lens=batch.cached_lens,
+                    k_scale=ksc[i] if batch.cached_lens is not None else None,
+                    v_scale=vsc[i] if batch.cached_lens is not None else None,
                 )
                 attn_d = ops.attention_decode(
                     q[Tp:], k_caches[i], v_caches[i],
                     batch.dec_block_tables, batch.dec_context_lens, self.scale,
+                    k_scale=ksc[i], v_scale=vsc[i],
                 )
                 attn = torch.cat([attn_p, attn_d], dim=0)
             else:
                 attn = ops.attention_decode(
-                    q, k_caches[i], v_caches[i], batch.block_tables, batch.context_lens, self.scale
+                    q, k_caches[i], v_caches[i], batch.block_tables,
+                    batch.context_lens, self.scale,
+                    k_scale=ksc[i], v_scale=vsc[i],
                 )
 
             h = self._maybe_all_reduce(

@@ -115,13 +115,19 @@ def rope_and_kv_write(
     positions: torch.Tensor,
     cos_sin: torch.Tensor,
     slot_mapping: torch.Tensor,
+    k_scale: Optional[torch.Tensor] = None,
+    v_scale: Optional[torch.Tensor] = None,
 ) -> None:
-    """Fused rope_inplace(q, k) + kv_cache_write(k, v): one launch on GPU."""
+    """Fused rope_inplace(q, k) + kv_cache_write(k, v): one launch on GPU.
+    With k_scale/v_scale the cache is fp8 e4m3 (per-row quantization)."""
     if q.is_cuda:
-        _native().rope_kv_write(q, k, v, k_cache, v_cache, positions, cos_sin, slot_mapping)
+        _native().rope_kv_write(
+            q, k, v, k_cache, v_cache, positions, cos_sin, slot_mapping,
+            k_scale, v_scale,
+        )
         return
     reference.rope_inplace(q, k, positions, cos_sin)
-    reference.kv_cache_write(k, v, k_cache, v_cache, slot_mapping)
+    reference.kv_cache_write(k, v, k_cache, v_cache, slot_mapping, k_scale, v_scale)
 
 
 def kv_cache_write(
@@ -130,11 +136,13 @@ def kv_cache_write(
     k_cache: torch.Tensor,
     v_cache: torch.Tensor,
     slot_mapping: torch.Tensor,
+    k_scale: Optional[torch.Tensor] = None,
+    v_scale: Optional[torch.Tensor] = None,
 ) -> None:
     if k.is_cuda:
-        _native().kv_cache_write(k, v, k_cache, v_cache, slot_mapping)
+        _native().kv_cache_write(k, v, k_cache, v_cache, slot_mapping, k_scale, v_scale)
         return
-    reference.kv_cache_write(k, v, k_cache, v_cache, slot_mapping)
+    reference.kv_cache_write(k, v, k_cache, v_cache, slot_mapping, k_scale, v_scale)
 
 
 QTILE = 64  # q rows per prefill workgroup (must match attention_prefill.hip)
@@ -167,6 +175,8 @@ def attention_prefill(
     v_cache: Optional[torch.Tensor] = None,
     block_tables: Optional[torch.Tensor] = None,
     cached_lens: Optional[torch.Tensor] = None,
+    k_scale: Optional[torch.Tensor] = None,
+    v_scale: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     """Varlen causal prefill. With cached_lens set, each sequence also
     attends (unmasked) to its first cached_lens[i] positions read from the
@@ -184,13 +194,14 @@ def attention_prefill(
         out = torch.empty_like(q)
         _native().attention_prefill(
             out, q, k, v, cu_seqlens.int(), tile_seq, tile_off, float(scale),
-            k_cache, v_cache, block_tables, cached_lens,
+            k_cache, v_cache, block_tables, cached_lens, k_scale, v_scale,
         )
         return out
     return reference.attention_prefill(
         q, k, v, cu_seqlens, scale, causal,
         k_cache=k_cache, v_cache=v_cache,
         block_tables=block_tables, cached_lens=cached_lens,
+        k_scale=k_scale, v_scale=v_scale,
     )
 
 
@@ -219,6 +230,8 @@ def attention_decode(
     block_tables: torch.Tensor,
     context_lens: torch.Tensor,
     scale: Optional[float] = None,
+    k_scale: Optional[torch.Tensor] = None,
+    v_scale: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     if scale is None:
         scale = float(q.shape[-1]) ** -0.5
@@ -238,15 +251,18 @@ def attention_decode(
             )
             _native().attention_decode(
                 out, q, k_cache, v_cache, block_tables, context_lens,
-                float(scale), pa, pm, nsplit,
+                float(scale), pa, pm, nsplit, k_scale, v_scale,
             )
         else:
             _native().attention_decode(
                 out, q, k_cache, v_cache, block_tables, context_lens,
-                float(scale), None, None, 1,
+                float(scale), None, None, 1, k_scale, v_scale,
             )
         return out
-    return reference.attention_decode(q, k_cache, v_cache, block_tables, context_lens, scale)
+    return reference.attention_decode(
+        q, k_cache, v_cache, block_tables, context_lens, scale,
+        k_scale=k_scale, v_scale=v_scale,
+    )
 
 
 # The 8-wave MF=2 depth-4 pipeline ties the tuned library when W is

@@ -33,13 +33,17 @@
 // B*Hkv underfills the chip but the context can be long (small-batch
 // long-context decode); the split count is derived from max_blocks so
 // it is hipGraph-capture-stable (idle splits exit on short contexts).
-template <int G, bool SPLIT>
+// FP8: caches hold e4m3 bytes with per-(token, head) row scales; the
+// scales factor out of both dot products (one multiply per key).
+template <int G, bool SPLIT, bool FP8>
 __launch_bounds__(NWAVES* WAVE_SIZE)
 __global__ void attention_decode_kernel(
     bf16* __restrict__ out,                 // [B, Hq, D]
     const bf16* __restrict__ q,             // [B, Hq, D]
-    const bf16* __restrict__ k_cache,       // [NB, Hkv, BS, D]
-    const bf16* __restrict__ v_cache,
+    const void* __restrict__ k_cache,       // [NB, Hkv, BS, D] bf16|fp8
+    const void* __restrict__ v_cache,
+    const float* __restrict__ k_scale,      // [NB, Hkv, BS] (FP8)
+    const float* __restrict__ v_scale,
     const int* __restrict__ block_tables,   // [B, max_blocks]
     const int* __restrict__ context_lens,   // [B]
     float scale,
@@ -118,16 +122,23 @@ __global__ void attention_decode_kernel(
         if (valid) {
             const int kblock = BT(pos / block_size);
             const int koff = pos % block_size;
-            const uint4* krow = reinterpret_cast<const uint4*>(
-                k_cache + (((size_t)kblock * Hkv + kvh) * block_size + koff) * D);
+            const size_t krow_i = ((size_t)kblock * Hkv + kvh) * block_size + koff;
 #pragma unroll 4
             for (int i = 0; i < D / 8; ++i) {
-                const uint4 kv8 = krow[i];
                 float kf[8];
-                unpack2(kv8.x, kf[0], kf[1]);
-                unpack2(kv8.y, kf[2], kf[3]);
-                unpack2(kv8.z, kf[4], kf[5]);
-                unpack2(kv8.w, kf[6], kf[7]);
+                if (FP8) {
+                    const uint2 kq = reinterpret_cast<const uint2*>(
+                        (const unsigned char*)k_cache + krow_i * D)[i];
+                    fp8x4_to_f32(kq.x, kf[0], kf[1], kf[2], kf[3]);
+                    fp8x4_to_f32(kq.y, kf[4], kf[5], kf[6], kf[7]);
+                } else {
+                    const uint4 kv8 = reinterpret_cast<const uint4*>(
+                        (const bf16*)k_cache + krow_i * D)[i];
+                    unpack2(kv8.x, kf[0], kf[1]);
+                    unpack2(kv8.y, kf[2], kf[3]);
+                    unpack2(kv8.z, kf[4], kf[5]);
+                    unpack2(kv8.w, kf[6], kf[7]);
+                }
 #pragma unroll
                 for (int g = 0; g < G; ++g) {
                     const float4 qa = *reinterpret_cast<const float4*>(&q_lds[g][i * 8]);
@@ -141,6 +152,11 @@ __global__ void attention_decode_kernel(
                     s[g] = fmaf(qb.z, kf[6], s[g]);
                     s[g] = fmaf(qb.w, kf[7], s[g]);
                 }
+            }
+            if (FP8) {
+                const float ksc = k_scale[krow_i];
+#pragma unroll
+                for (int g = 0; g < G; ++g) s[g] *= ksc;
             }
         }
 
@@ -172,24 +188,37 @@ __global__ void attention_decode_kernel(
 #pragma unroll 2
         for (int j = 0; j < WAVE_SIZE; j += 8) {
             uint2 vp[4];
+            uint32_t vp8[4];
+            float vsc[4];
 #pragma unroll
             for (int u = 0; u < 4; ++u) {
                 const int pos = min(c * WAVE_SIZE + j + 2 * u + khalf, span1 - 1);
                 const int vb = BT(pos / block_size);
                 const int vo = pos % block_size;
-                vp[u] = *reinterpret_cast<const uint2*>(
-                    v_cache + (((size_t)vb * Hkv + kvh) * block_size + vo) * D +
-                    dbase);
+                const size_t vrow_i = ((size_t)vb * Hkv + kvh) * block_size + vo;
+                if (FP8) {
+                    vp8[u] = *reinterpret_cast<const uint32_t*>(
+                        (const unsigned char*)v_cache + vrow_i * D + dbase);
+                    vsc[u] = v_scale[vrow_i];
+                } else {
+                    vp[u] = *reinterpret_cast<const uint2*>(
+                        (const bf16*)v_cache + vrow_i * D + dbase);
+                }
             }
 #pragma unroll
             for (int u = 0; u < 4; ++u) {
                 float v0, v1, v2, v3;
-                unpack2(vp[u].x, v0, v1);
-                unpack2(vp[u].y, v2, v3);
+                if (FP8) {
+                    fp8x4_to_f32(vp8[u], v0, v1, v2, v3);
+                } else {
+                    unpack2(vp[u].x, v0, v1);
+                    unpack2(vp[u].y, v2, v3);
+                }
                 const int key = j + 2 * u + khalf;
 #pragma unroll
                 for (int g = 0; g < G; ++g) {
-                    const float pj = p_lds[wave][g][key];
+                    // fp8: fold the V row scale into p once per key
+                    const float pj = p_lds[wave][g][key] * (FP8 ? vsc[u] : 1.f);
                     acc[g][0] = fmaf(pj, v0, acc[g][0]);
                     acc[g][1] = fmaf(pj, v1, acc[g][1]);
                     acc[g][2] = fmaf(pj, v2, acc[g][2]);
@@ -287,6 +316,7 @@ __global__ void decode_combine_kernel(
 
 extern "C" hipError_t launch_attention_decode(
     void* out, const void* q, const void* k_cache, const void* v_cache,
+    const float* k_scale, const float* v_scale,
     const int* block_tables, const int* context_lens, float scale, int B,
     int Hq, int Hkv, int block_size, int max_blocks, int D, int64_t q_stride,
     float* part_acc, float* part_ml, int nsplit, hipStream_t stream) {
@@ -295,13 +325,18 @@ extern "C" hipError_t launch_attention_decode(
     if (nsplit > 1 && (part_acc == nullptr || part_ml == nullptr))
         return hipErrorInvalidValue;
     const int G = Hq / Hkv;
+    const bool fp8 = k_scale != nullptr;
     dim3 grid(B, Hkv, nsplit > 1 ? nsplit : 1);
     dim3 block(NWAVES * WAVE_SIZE);
-#define LAUNCH_G2(GV, SPLIT)                                                   \
-    attention_decode_kernel<GV, SPLIT><<<grid, block, 0, stream>>>(            \
-        (bf16*)out, (const bf16*)q, (const bf16*)k_cache,                      \
-        (const bf16*)v_cache, block_tables, context_lens, scale, Hq, Hkv,     \
+#define LAUNCH_G3(GV, SPLIT, FP8V)                                             \
+    attention_decode_kernel<GV, SPLIT, FP8V><<<grid, block, 0, stream>>>(      \
+        (bf16*)out, (const bf16*)q, k_cache, v_cache, k_scale, v_scale,        \
+        block_tables, context_lens, scale, Hq, Hkv,                            \
         block_size, max_blocks, q_stride, part_acc, part_ml, nsplit)
+#define LAUNCH_G2(GV, SPLIT)                                                   \
+    do {                                                                       \
+        if (fp8) LAUNCH_G3(GV, SPLIT, true); else LAUNCH_G3(GV, SPLIT, false);\
+    } while (0)
 #define LAUNCH_G(GV)                                                           \
     do {                                                                       \
         if (nsplit > 1) LAUNCH_G2(GV, true); else LAUNCH_G2(GV, false);        \
@@ -319,6 +354,7 @@ extern "C" hipError_t launch_attention_decode(
     }
 #undef LAUNCH_G
 #undef LAUNCH_G2
+#undef LAUNCH_G3
     if (nsplit > 1) {
         decode_combine_kernel<<<dim3(B * Hq), dim3(WAVE_SIZE), 0, stream>>>(
             (bf16*)out, part_acc, part_ml, nsplit);

@@ -40,6 +40,10 @@ __device__ __forceinline__ f32x4 mfma16(bf16x8 a, bf16x8 b, f32x4 c) {
 __device__ __forceinline__ int swz16(int chunk, int row) { return chunk ^ (row & 15); }
 __device__ __forceinline__ int swz8(int chunk, int row) { return chunk ^ (row & 7); }
 
+// FP8C: the PAGED-CACHE side (phase 0) holds e4m3 bytes with per-row
+// scales; fresh K/V stay bf16. Dequant happens at LDS staging, so the
+// compute phases are unchanged.
+template <bool FP8C>
 __launch_bounds__(PF_WAVES* WAVE_SIZE)
 __global__ void attention_prefill_kernel(
     bf16* __restrict__ out,             // [T, Hq, D]
@@ -58,8 +62,10 @@ __global__ void attention_prefill_kernel(
     // cached-context phase (prefix caching / chunked prefill): per-seq
     // prior KV already resident in the paged cache; every fresh q row
     // attends to ALL of it (no mask — cached positions precede the tile)
-    const bf16* __restrict__ k_cache,        // [NB, Hkv, BS, D] or null
-    const bf16* __restrict__ v_cache,
+    const void* __restrict__ k_cache,        // [NB, Hkv, BS, D] or null
+    const void* __restrict__ v_cache,
+    const float* __restrict__ k_scale,       // [NB, Hkv, BS] (FP8C)
+    const float* __restrict__ v_scale,
     const int* __restrict__ block_tables,    // [B, max_blocks] or null
     const int* __restrict__ cached_lens,     // [B] or null
     int block_size,
@@ -126,12 +132,35 @@ __global__ void attention_prefill_kernel(
             uint4 kraw = {0, 0, 0, 0}, vraw = {0, 0, 0, 0};
             if (kk < keys_here) {
                 const int pos = key0 + kk;
-                const size_t base =
-                    (((size_t)bt[pos / block_size] * Hkv + kvh) * block_size +
-                     pos % block_size) *
-                    PF_D;
-                kraw = *reinterpret_cast<const uint4*>(k_cache + base + d0);
-                vraw = *reinterpret_cast<const uint4*>(v_cache + base + d0);
+                const size_t row =
+                    ((size_t)bt[pos / block_size] * Hkv + kvh) * block_size +
+                    pos % block_size;
+                if (FP8C) {
+                    // dequantize 8 e4m3 bytes (+ row scale) to bf16
+                    const uint2 kq = *reinterpret_cast<const uint2*>(
+                        (const unsigned char*)k_cache + row * PF_D + d0);
+                    const uint2 vq = *reinterpret_cast<const uint2*>(
+                        (const unsigned char*)v_cache + row * PF_D + d0);
+                    const float ksc = k_scale[row], vsc = v_scale[row];
+                    float kf[8], vf[8];
+                    fp8x4_to_f32(kq.x, kf[0], kf[1], kf[2], kf[3]);
+                    fp8x4_to_f32(kq.y, kf[4], kf[5], kf[6], kf[7]);
+                    fp8x4_to_f32(vq.x, vf[0], vf[1], vf[2], vf[3]);
+                    fp8x4_to_f32(vq.y, vf[4], vf[5], vf[6], vf[7]);
+                    uint32_t* kp = reinterpret_cast<uint32_t*>(&kraw);
+                    uint32_t* vp = reinterpret_cast<uint32_t*>(&vraw);
+#pragma unroll
+                    for (int j = 0; j < 4; ++j) {
+                        kp[j] = pack2(kf[2 * j] * ksc, kf[2 * j + 1] * ksc);
+                        vp[j] = pack2(vf[2 * j] * vsc, vf[2 * j + 1] * vsc);
+                    }
+                } else {
+                    const size_t base = row * PF_D;
+                    kraw = *reinterpret_cast<const uint4*>(
+                        (const bf16*)k_cache + base + d0);
+                    vraw = *reinterpret_cast<const uint4*>(
+                        (const bf16*)v_cache + base + d0);
+                }
             }
             *reinterpret_cast<uint4*>(&k_lds[kk][swz16(d0 / 8, kk) * 8]) = kraw;
             const bf16* v8 = reinterpret_cast<const bf16*>(&vraw);
@@ -333,16 +362,24 @@ extern "C" hipError_t launch_attention_prefill(
     const int* cu_seqlens, const int* tile_seq, const int* tile_off,
     int ntiles, float scale, int Hq, int Hkv, int D, int64_t q_stride,
     int64_t k_stride, int64_t v_stride, const void* k_cache,
-    const void* v_cache, const int* block_tables, const int* cached_lens,
+    const void* v_cache, const float* k_scale, const float* v_scale,
+    const int* block_tables, const int* cached_lens,
     int block_size, int max_blocks, hipStream_t stream) {
     if (D != PF_D) return hipErrorNotSupported;
     dim3 grid(ntiles, Hq);
     dim3 block(PF_WAVES * WAVE_SIZE);
-    attention_prefill_kernel<<<grid, block, 0, stream>>>(
-        (bf16*)out, (const bf16*)q, (const bf16*)k, (const bf16*)v, cu_seqlens,
-        tile_seq, tile_off, scale, Hq, Hkv, q_stride, k_stride, v_stride,
-        (const bf16*)k_cache, (const bf16*)v_cache, block_tables, cached_lens,
-        block_size, max_blocks);
+    if (k_scale != nullptr)
+        attention_prefill_kernel<true><<<grid, block, 0, stream>>>(
+            (bf16*)out, (const bf16*)q, (const bf16*)k, (const bf16*)v,
+            cu_seqlens, tile_seq, tile_off, scale, Hq, Hkv, q_stride, k_stride,
+            v_stride, k_cache, v_cache, k_scale, v_scale, block_tables,
+            cached_lens, block_size, max_blocks);
+    else
+        attention_prefill_kernel<false><<<grid, block, 0, stream>>>(
+            (bf16*)out, (const bf16*)q, (const bf16*)k, (const bf16*)v,
+            cu_seqlens, tile_seq, tile_off, scale, Hq, Hkv, q_stride, k_stride,
+            v_stride, k_cache, v_cache, nullptr, nullptr, block_tables,
+            cached_lens, block_size, max_blocks);
     HIP_CHECK_LAST();
     return hipSuccess;
 }

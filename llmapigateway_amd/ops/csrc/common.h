@@ -11,6 +11,28 @@
 
 using bf16 = __hip_bfloat16;
 
+// fp8 KV cache (OCP e4m3fn — gfx950's native format, NOT MI300X fnuz).
+// Quantization: per (token, head) row scale = amax/448; dequant scales
+// factor out of the attention dot products so the kernels multiply once
+// per key instead of per element.
+#include <hip/hip_fp8.h>
+#define FP8_MAX 448.0f
+typedef __attribute__((__vector_size__(2 * sizeof(float)))) float f32x2_t;
+
+__device__ __forceinline__ unsigned char f2fp8(float v) {
+    return __hip_fp8_e4m3(v).__x;
+}
+// 4 packed e4m3 bytes -> 4 floats (hardware cvt_pk)
+__device__ __forceinline__ void fp8x4_to_f32(uint32_t q, float& a, float& b,
+                                             float& c, float& d) {
+    const f32x2_t lo = __builtin_amdgcn_cvt_pk_f32_fp8(q, false);
+    const f32x2_t hi = __builtin_amdgcn_cvt_pk_f32_fp8(q, true);
+    a = lo[0];
+    b = lo[1];
+    c = hi[0];
+    d = hi[1];
+}
+
 __device__ __forceinline__ float bf2f(bf16 v) { return __bfloat162float(v); }
 __device__ __forceinline__ bf16 f2bf(float v) { return __float2bfloat16(v); }
 

@@ -246,12 +246,50 @@ extern "C" hipError_t launch_swiglu(
 // cache-copy phase reads them back (same CU, own L1 — coherent).
 // ---------------------------------------------------------------------------
 
+
+// ---------------------------------------------------------------------------
+// fp8 KV write helper: quantize one (token, head) row of D elements to
+// e4m3 with a per-row scale (amax/448). One 64-lane wave per head row:
+// lane covers D/64 elements, wave-reduced amax, packed byte stores.
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ void fp8_write_row(
+    const bf16* __restrict__ src,      // D contiguous bf16
+    unsigned char* __restrict__ dst,   // D bytes
+    float* __restrict__ scale_out,     // 1 float
+    int D, int lane) {
+    const int per = D / WAVE_SIZE;     // 2 at D=128
+    float vals[4];                     // per <= 4 supported (D <= 256)
+    float amax = 0.f;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+        if (j < per) {
+            vals[j] = bf2f(src[lane * per + j]);
+            amax = fmaxf(amax, fabsf(vals[j]));
+        }
+    }
+    amax = wave_reduce_max(amax);
+    const float sc = amax > 0.f ? amax / FP8_MAX : 1.f;
+    const float inv = 1.f / sc;
+    if (lane == 0) *scale_out = sc;
+    // pack per bytes (2 at D=128) and store
+    if (per == 2) {
+        unsigned short pk = (unsigned short)f2fp8(vals[0]* inv) |
+                            ((unsigned short)f2fp8(vals[1]* inv) << 8);
+        reinterpret_cast<unsigned short*>(dst)[lane] = pk;
+    } else {
+        for (int j = 0; j < per; ++j) dst[lane * per + j] = f2fp8(vals[j] * inv);
+    }
+}
+
+template <bool FP8>
 __global__ void rope_kv_kernel(
     bf16* __restrict__ q,
     bf16* __restrict__ k,
     const bf16* __restrict__ v,
-    bf16* __restrict__ k_cache,
-    bf16* __restrict__ v_cache,
+    void* __restrict__ k_cache,        // bf16 or fp8 bytes
+    void* __restrict__ v_cache,
+    float* __restrict__ k_scale,       // [NB, Hkv, BS] (FP8)
+    float* __restrict__ v_scale,
     const int64_t* __restrict__ positions,     // [T]
     const float* __restrict__ cos_sin,         // [max_pos, D]
     const int64_t* __restrict__ slot_mapping,  // [T]
@@ -288,6 +326,22 @@ __global__ void rope_kv_kernel(
 
     const int64_t block = slot / block_size;
     const int64_t off = slot % block_size;
+    if (FP8) {
+        // one wave per head row (k then v): blockDim 256 = 4 waves
+        const int lane = threadIdx.x & (WAVE_SIZE - 1);
+        const int wave = threadIdx.x >> 6;
+        for (int hh = wave; hh < 2 * Hkv; hh += blockDim.x / WAVE_SIZE) {
+            const int h = hh % Hkv;
+            const bool is_k = hh < Hkv;
+            const size_t row = ((size_t)block * Hkv + h) * block_size + off;
+            fp8_write_row(
+                (is_k ? k + (size_t)t * k_stride : v + (size_t)t * v_stride) +
+                    (size_t)h * D,
+                (unsigned char*)(is_k ? k_cache : v_cache) + row * D,
+                (is_k ? k_scale : v_scale) + row, D, lane);
+        }
+        return;
+    }
     const int nvec = (Hkv * D) / 8;
     const uint4* ksrc = reinterpret_cast<const uint4*>(k + (size_t)t * k_stride);
     const uint4* vsrc = reinterpret_cast<const uint4*>(v + (size_t)t * v_stride);
@@ -295,22 +349,31 @@ __global__ void rope_kv_kernel(
         const int h = (i * 8) / D;
         const int d = (i * 8) % D;
         const size_t dst = (((size_t)block * Hkv + h) * block_size + off) * D + d;
-        reinterpret_cast<uint4*>(k_cache + dst)[0] = ksrc[i];
-        reinterpret_cast<uint4*>(v_cache + dst)[0] = vsrc[i];
+        reinterpret_cast<uint4*>((bf16*)k_cache + dst)[0] = ksrc[i];
+        reinterpret_cast<uint4*>((bf16*)v_cache + dst)[0] = vsrc[i];
     }
 }
 
 extern "C" hipError_t launch_rope_kv(
     void* q, void* k, const void* v, void* k_cache, void* v_cache,
+    float* k_scale, float* v_scale,
     const int64_t* positions, const float* cos_sin,
     const int64_t* slot_mapping, int T, int64_t q_stride, int64_t k_stride,
     int64_t v_stride, int Hq, int Hkv, int D, int block_size,
     hipStream_t stream) {
     if (D % 8 != 0) return hipErrorInvalidValue;
-    rope_kv_kernel<<<T, 256, 0, stream>>>(
-        (bf16*)q, (bf16*)k, (const bf16*)v, (bf16*)k_cache, (bf16*)v_cache,
-        positions, cos_sin, slot_mapping, q_stride, k_stride, v_stride, Hq,
-        Hkv, D, block_size);
+    if (k_scale != nullptr) {
+        if (D % WAVE_SIZE != 0 || D > 256) return hipErrorInvalidValue;
+        rope_kv_kernel<true><<<T, 256, 0, stream>>>(
+            (bf16*)q, (bf16*)k, (const bf16*)v, k_cache, v_cache, k_scale,
+            v_scale, positions, cos_sin, slot_mapping, q_stride, k_stride,
+            v_stride, Hq, Hkv, D, block_size);
+    } else {
+        rope_kv_kernel<false><<<T, 256, 0, stream>>>(
+            (bf16*)q, (bf16*)k, (const bf16*)v, k_cache, v_cache, nullptr,
+            nullptr, positions, cos_sin, slot_mapping, q_stride, k_stride,
+            v_stride, Hq, Hkv, D, block_size);
+    }
     HIP_CHECK_LAST();
     return hipSuccess;
 }
@@ -321,11 +384,14 @@ extern "C" hipError_t launch_rope_kv(
 // One block per token; vectorized 16 B per lane.
 // ---------------------------------------------------------------------------
 
+template <bool FP8>
 __global__ void kv_cache_write_kernel(
     const bf16* __restrict__ k,
     const bf16* __restrict__ v,
-    bf16* __restrict__ k_cache,
-    bf16* __restrict__ v_cache,
+    void* __restrict__ k_cache,
+    void* __restrict__ v_cache,
+    float* __restrict__ k_scale,
+    float* __restrict__ v_scale,
     const int64_t* __restrict__ slot_mapping,  // [T]
     int64_t k_stride,
     int64_t v_stride,
@@ -337,6 +403,21 @@ __global__ void kv_cache_write_kernel(
     if (slot < 0) return;
     const int64_t block = slot / block_size;
     const int64_t off = slot % block_size;
+    if (FP8) {
+        const int lane = threadIdx.x & (WAVE_SIZE - 1);
+        const int wave = threadIdx.x >> 6;
+        for (int hh = wave; hh < 2 * Hkv; hh += blockDim.x / WAVE_SIZE) {
+            const int h = hh % Hkv;
+            const bool is_k = hh < Hkv;
+            const size_t row = ((size_t)block * Hkv + h) * block_size + off;
+            fp8_write_row(
+                (is_k ? k + (size_t)t * k_stride : v + (size_t)t * v_stride) +
+                    (size_t)h * D,
+                (unsigned char*)(is_k ? k_cache : v_cache) + row * D,
+                (is_k ? k_scale : v_scale) + row, D, lane);
+        }
+        return;
+    }
     const int nvec = (Hkv * D) / 8;  // uint4 elements per token
 
     const uint4* ksrc = reinterpret_cast<const uint4*>(k + (size_t)t * k_stride);
@@ -346,19 +427,27 @@ __global__ void kv_cache_write_kernel(
         const int d = (i * 8) % D;
         const size_t dst =
             (((size_t)block * Hkv + h) * block_size + off) * D + d;
-        reinterpret_cast<uint4*>(k_cache + dst)[0] = ksrc[i];
-        reinterpret_cast<uint4*>(v_cache + dst)[0] = vsrc[i];
+        reinterpret_cast<uint4*>((bf16*)k_cache + dst)[0] = ksrc[i];
+        reinterpret_cast<uint4*>((bf16*)v_cache + dst)[0] = vsrc[i];
     }
 }
 
 extern "C" hipError_t launch_kv_cache_write(
     const void* k, const void* v, void* k_cache, void* v_cache,
+    float* k_scale, float* v_scale,
     const int64_t* slot_mapping, int T, int64_t k_stride, int64_t v_stride,
     int Hkv, int block_size, int D, hipStream_t stream) {
     if (D % 8 != 0) return hipErrorInvalidValue;
-    kv_cache_write_kernel<<<T, 256, 0, stream>>>(
-        (const bf16*)k, (const bf16*)v, (bf16*)k_cache, (bf16*)v_cache,
-        slot_mapping, k_stride, v_stride, Hkv, block_size, D);
+    if (k_scale != nullptr) {
+        if (D % WAVE_SIZE != 0 || D > 256) return hipErrorInvalidValue;
+        kv_cache_write_kernel<true><<<T, 256, 0, stream>>>(
+            (const bf16*)k, (const bf16*)v, k_cache, v_cache, k_scale,
+            v_scale, slot_mapping, k_stride, v_stride, Hkv, block_size, D);
+    } else {
+        kv_cache_write_kernel<false><<<T, 256, 0, stream>>>(
+            (const bf16*)k, (const bf16*)v, k_cache, v_cache, nullptr,
+            nullptr, slot_mapping, k_stride, v_stride, Hkv, block_size, D);
+    }
     HIP_CHECK_LAST();
     return hipSuccess;
 }

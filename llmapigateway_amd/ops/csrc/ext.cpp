@@ -16,10 +16,10 @@ hipError_t launch_rmsnorm(void*, const void*, const void*, float, int, int, hipS
 hipError_t launch_rmsnorm_residual(void*, const void*, void*, const void*, float, int, int, hipStream_t);
 hipError_t launch_rope(void*, void*, const int64_t*, const float*, int, int64_t, int64_t, int, int, int, hipStream_t);
 hipError_t launch_swiglu(void*, const void*, int, int, hipStream_t);
-hipError_t launch_kv_cache_write(const void*, const void*, void*, void*, const int64_t*, int, int64_t, int64_t, int, int, int, hipStream_t);
-hipError_t launch_rope_kv(void*, void*, const void*, void*, void*, const int64_t*, const float*, const int64_t*, int, int64_t, int64_t, int64_t, int, int, int, int, hipStream_t);
-hipError_t launch_attention_decode(void*, const void*, const void*, const void*, const int*, const int*, float, int, int, int, int, int, int, int64_t, float*, float*, int, hipStream_t);
-hipError_t launch_attention_prefill(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, float, int, int, int, int64_t, int64_t, int64_t, const void*, const void*, const int*, const int*, int, int, hipStream_t);
+hipError_t launch_kv_cache_write(const void*, const void*, void*, void*, float*, float*, const int64_t*, int, int64_t, int64_t, int, int, int, hipStream_t);
+hipError_t launch_rope_kv(void*, void*, const void*, void*, void*, float*, float*, const int64_t*, const float*, const int64_t*, int, int64_t, int64_t, int64_t, int, int, int, int, hipStream_t);
+hipError_t launch_attention_decode(void*, const void*, const void*, const void*, const float*, const float*, const int*, const int*, float, int, int, int, int, int, int, int64_t, float*, float*, int, hipStream_t);
+hipError_t launch_attention_prefill(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, float, int, int, int, int64_t, int64_t, int64_t, const void*, const void*, const float*, const float*, const int*, const int*, int, int, hipStream_t);
 hipError_t launch_sample(int64_t*, const float*, const float*, const float*, float*, int*, int, int, hipStream_t);
 hipError_t launch_topk_topp_filter(float*, const float*, const int*, int, int, hipStream_t);
 hipError_t launch_gemm_skinny(void*, float*, const void*, const void*, int, int, int, int, int, hipStream_t);
@@ -42,6 +42,20 @@ hipStream_t current_stream() {
 void check_bf16(const torch::Tensor& t, const char* name) {
     TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
     TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+}
+
+// fp8 KV caches are stored as uint8 (e4m3 bits) with fp32 row scales
+void check_cache(const torch::Tensor& t, const char* name) {
+    TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+    TORCH_CHECK(t.scalar_type() == torch::kBFloat16 ||
+                t.scalar_type() == torch::kUInt8,
+                name, " must be bf16 or uint8(e4m3)");
+}
+
+float* scale_ptr(const c10::optional<torch::Tensor>& s) {
+    if (!s.has_value() || !s->defined()) return nullptr;
+    TORCH_CHECK(s->scalar_type() == torch::kFloat32 && s->is_contiguous());
+    return s->data_ptr<float>();
 }
 
 void rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor weight, double eps) {
@@ -92,9 +106,11 @@ void swiglu(torch::Tensor out, torch::Tensor x) {
 }
 
 void kv_cache_write(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
-                    torch::Tensor v_cache, torch::Tensor slot_mapping) {
+                    torch::Tensor v_cache, torch::Tensor slot_mapping,
+                    c10::optional<torch::Tensor> k_scale,
+                    c10::optional<torch::Tensor> v_scale) {
     check_bf16(k, "k");
-    check_bf16(k_cache, "k_cache");
+    check_cache(k_cache, "k_cache");
     TORCH_CHECK(k.dim() == 3 && k_cache.dim() == 4);
     TORCH_CHECK(k.stride(2) == 1 && k.stride(1) == k.size(2));
     TORCH_CHECK(v.stride(2) == 1 && v.stride(1) == v.size(2));
@@ -104,8 +120,12 @@ void kv_cache_write(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
     const int Hkv = k_cache.size(1);
     const int block_size = k_cache.size(2);
     const int D = k_cache.size(3);
+    TORCH_CHECK((k_cache.scalar_type() == torch::kUInt8) ==
+                (k_scale.has_value() && k_scale->defined()),
+                "fp8 cache needs scales; bf16 cache must not have them");
     CHECK_HIP(launch_kv_cache_write(
         k.data_ptr(), v.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+        scale_ptr(k_scale), scale_ptr(v_scale),
         slot_mapping.data_ptr<int64_t>(), T, k.stride(0), v.stride(0), Hkv,
         block_size, D, current_stream()));
 }
@@ -113,10 +133,12 @@ void kv_cache_write(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
 void rope_kv_write(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                    torch::Tensor k_cache, torch::Tensor v_cache,
                    torch::Tensor positions, torch::Tensor cos_sin,
-                   torch::Tensor slot_mapping) {
+                   torch::Tensor slot_mapping,
+                   c10::optional<torch::Tensor> k_scale,
+                   c10::optional<torch::Tensor> v_scale) {
     check_bf16(q, "q");
     check_bf16(k, "k");
-    check_bf16(k_cache, "k_cache");
+    check_cache(k_cache, "k_cache");
     TORCH_CHECK(q.dim() == 3 && k.dim() == 3 && v.dim() == 3 && k_cache.dim() == 4);
     TORCH_CHECK(q.stride(2) == 1 && k.stride(2) == 1 && v.stride(2) == 1);
     TORCH_CHECK(q.stride(1) == q.size(2) && k.stride(1) == k.size(2) &&
@@ -128,7 +150,8 @@ void rope_kv_write(torch::Tensor q, torch::Tensor k, torch::Tensor v,
     const int T = q.size(0);
     CHECK_HIP(launch_rope_kv(
         q.data_ptr(), k.data_ptr(), v.data_ptr(), k_cache.data_ptr(),
-        v_cache.data_ptr(), positions.data_ptr<int64_t>(),
+        v_cache.data_ptr(), scale_ptr(k_scale), scale_ptr(v_scale),
+        positions.data_ptr<int64_t>(),
         cos_sin.data_ptr<float>(), slot_mapping.data_ptr<int64_t>(), T,
         q.stride(0), k.stride(0), v.stride(0), q.size(1), k_cache.size(1),
         q.size(2), k_cache.size(2), current_stream()));
@@ -138,7 +161,13 @@ void attention_decode(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
                       torch::Tensor v_cache, torch::Tensor block_tables,
                       torch::Tensor context_lens, double scale,
                       c10::optional<torch::Tensor> part_acc,
-                      c10::optional<torch::Tensor> part_ml, int64_t nsplit) {
+                      c10::optional<torch::Tensor> part_ml, int64_t nsplit,
+                      c10::optional<torch::Tensor> k_scale,
+                      c10::optional<torch::Tensor> v_scale) {
+    check_cache(k_cache, "k_cache");
+    TORCH_CHECK((k_cache.scalar_type() == torch::kUInt8) ==
+                (k_scale.has_value() && k_scale->defined()),
+                "fp8 cache needs scales; bf16 cache must not have them");
     check_bf16(q, "q");
     check_bf16(out, "out");
     TORCH_CHECK(out.is_contiguous());
@@ -164,6 +193,7 @@ void attention_decode(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
     }
     CHECK_HIP(launch_attention_decode(
         out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+        scale_ptr(k_scale), scale_ptr(v_scale),
         block_tables.data_ptr<int>(), context_lens.data_ptr<int>(), (float)scale,
         B, Hq, Hkv, block_size, max_blocks, D, q.stride(0), pa, pm, (int)nsplit,
         current_stream()));
@@ -176,7 +206,9 @@ void attention_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                        c10::optional<torch::Tensor> k_cache,
                        c10::optional<torch::Tensor> v_cache,
                        c10::optional<torch::Tensor> block_tables,
-                       c10::optional<torch::Tensor> cached_lens) {
+                       c10::optional<torch::Tensor> cached_lens,
+                       c10::optional<torch::Tensor> k_scale,
+                       c10::optional<torch::Tensor> v_scale) {
     check_bf16(q, "q");
     check_bf16(out, "out");
     TORCH_CHECK(out.is_contiguous());
@@ -209,8 +241,9 @@ void attention_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
         out.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
         cu_seqlens.data_ptr<int>(), tile_seq.data_ptr<int>(),
         tile_off.data_ptr<int>(), ntiles, (float)scale, q.size(1), k.size(1),
-        q.size(2), q.stride(0), k.stride(0), v.stride(0), kc, vc, bt, cl,
-        block_size, max_blocks, current_stream()));
+        q.size(2), q.stride(0), k.stride(0), v.stride(0), kc, vc,
+        kc ? scale_ptr(k_scale) : nullptr, vc ? scale_ptr(v_scale) : nullptr,
+        bt, cl, block_size, max_blocks, current_stream()));
 }
 
 void sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temperature,

@@ -72,12 +72,31 @@ def swiglu(x: torch.Tensor) -> torch.Tensor:
     return (g * torch.sigmoid(g) * u).to(x.dtype)
 
 
+FP8_MAX = 448.0  # e4m3fn
+
+
+def fp8_quantize_rows(x: torch.Tensor):
+    """Per-row (last-dim) e4m3 quantization: returns (uint8 bits, scales).
+    x: [..., D] -> bits [..., D] uint8, scales [...] fp32."""
+    xf = x.float()
+    amax = xf.abs().amax(dim=-1)
+    scale = torch.where(amax > 0, amax / FP8_MAX, torch.ones_like(amax))
+    q = (xf / scale.unsqueeze(-1)).to(torch.float8_e4m3fn)
+    return q.view(torch.uint8), scale
+
+
+def fp8_dequantize_rows(bits: torch.Tensor, scale: torch.Tensor) -> torch.Tensor:
+    return bits.view(torch.float8_e4m3fn).float() * scale.unsqueeze(-1).float()
+
+
 def kv_cache_write(
     k: torch.Tensor,
     v: torch.Tensor,
     k_cache: torch.Tensor,
     v_cache: torch.Tensor,
     slot_mapping: torch.Tensor,
+    k_scale: "Optional[torch.Tensor]" = None,
+    v_scale: "Optional[torch.Tensor]" = None,
 ) -> None:
     """Scatter new K/V rows into the paged cache.
 
@@ -87,6 +106,14 @@ def kv_cache_write(
     block_size = k_cache.shape[2]
     blocks = torch.div(slot_mapping, block_size, rounding_mode="floor")
     offs = slot_mapping % block_size
+    if k_scale is not None:  # fp8 e4m3 cache with per-row scales
+        kq, ks = fp8_quantize_rows(k)
+        vq, vs = fp8_quantize_rows(v)
+        k_cache[blocks, :, offs, :] = kq
+        v_cache[blocks, :, offs, :] = vq
+        k_scale[blocks, :, offs] = ks
+        v_scale[blocks, :, offs] = vs
+        return
     k_cache[blocks, :, offs, :] = k.to(k_cache.dtype)
     v_cache[blocks, :, offs, :] = v.to(v_cache.dtype)
 
@@ -102,6 +129,8 @@ def attention_prefill(
     v_cache: Optional[torch.Tensor] = None,
     block_tables: Optional[torch.Tensor] = None,
     cached_lens: Optional[torch.Tensor] = None,
+    k_scale: Optional[torch.Tensor] = None,
+    v_scale: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     """Varlen causal attention with GQA over fresh K/V; with cached_lens,
     each sequence also attends (unmasked) to its cached prefix gathered
@@ -125,8 +154,13 @@ def attention_prefill(
             rows_k, rows_v = [], []
             for pos in range(nc):
                 blk = int(block_tables[i, pos // bs])
-                rows_k.append(k_cache[blk, :, pos % bs, :])
-                rows_v.append(v_cache[blk, :, pos % bs, :])
+                kr = k_cache[blk, :, pos % bs, :]
+                vr = v_cache[blk, :, pos % bs, :]
+                if k_scale is not None:
+                    kr = fp8_dequantize_rows(kr, k_scale[blk, :, pos % bs])
+                    vr = fp8_dequantize_rows(vr, v_scale[blk, :, pos % bs])
+                rows_k.append(kr)
+                rows_v.append(vr)
             kc = torch.stack(rows_k).float().repeat_interleave(group, dim=1)
             vc = torch.stack(rows_v).float().repeat_interleave(group, dim=1)
             ki = torch.cat([kc, ki], dim=0)
@@ -154,6 +188,8 @@ def attention_decode(
     block_tables: torch.Tensor,
     context_lens: torch.Tensor,
     scale: Optional[float] = None,
+    k_scale: Optional[torch.Tensor] = None,
+    v_scale: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     """Single-token decode attention over the paged KV cache."""
     B, Hq, D = q.shape
@@ -168,8 +204,14 @@ def attention_decode(
         nblocks = (L + block_size - 1) // block_size
         blocks = block_tables[b, :nblocks].long()
         # gather [L, Hkv, D]
-        kk = k_cache[blocks].permute(0, 2, 1, 3).reshape(-1, Hkv, D)[:L].float()
-        vv = v_cache[blocks].permute(0, 2, 1, 3).reshape(-1, Hkv, D)[:L].float()
+        if k_scale is not None:
+            kd = fp8_dequantize_rows(k_cache[blocks], k_scale[blocks])
+            vd = fp8_dequantize_rows(v_cache[blocks], v_scale[blocks])
+            kk = kd.permute(0, 2, 1, 3).reshape(-1, Hkv, D)[:L].float()
+            vv = vd.permute(0, 2, 1, 3).reshape(-1, Hkv, D)[:L].float()
+        else:
+            kk = k_cache[blocks].permute(0, 2, 1, 3).reshape(-1, Hkv, D)[:L].float()
+            vv = v_cache[blocks].permute(0, 2, 1, 3).reshape(-1, Hkv, D)[:L].float()
         kk = kk.repeat_interleave(group, dim=1)  # [L, Hq, D]
         vv = vv.repeat_interleave(group, dim=1)
         qb = q[b].float()  # [Hq, D]

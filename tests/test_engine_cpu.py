@@ -246,3 +246,29 @@ def test_admission_batch_admits_when_idle():
     )
     r = eng.generate([1, 2, 3], SamplingParams(max_tokens=3, ignore_eos=True))
     assert r.state == "finished"  # idle engine never waits for a batch
+
+
+def test_fp8_kv_cache_cpu_close_to_bf16():
+    """fp8 KV engine output stays close to the fp32-cache run on CPU (the
+    quantization is per-row e4m3: ~2 decimal digits). Greedy argmax over
+    tiny random weights is tolerance-fragile, so compare a short horizon
+    and only require the FIRST tokens to agree."""
+    a = LLMEngine(model="tiny-llama", device="cpu", dtype=torch.float32,
+                  num_blocks=64, seed=7)
+    b = LLMEngine(model="tiny-llama", device="cpu", dtype=torch.float32,
+                  num_blocks=64, seed=7, kv_dtype="fp8")
+    assert b.kv.fp8 and b.kv.k_caches[0].dtype == torch.uint8
+    assert b.kv.k_scales[0] is not None
+    pa = a.generate(list(range(5, 21)), SamplingParams(max_tokens=4, ignore_eos=True))
+    pb = b.generate(list(range(5, 21)), SamplingParams(max_tokens=4, ignore_eos=True))
+    assert pa.out_ids[0] == pb.out_ids[0]
+
+
+def test_fp8_kv_capacity_gain():
+    from llmapigateway_amd.engine.kvcache import PagedKVCache
+    from llmapigateway_amd.models.configs import get_model_config
+
+    cfg = get_model_config("llama-3-8b")
+    bf16 = PagedKVCache.block_bytes(cfg, 64, torch.bfloat16)
+    fp8 = PagedKVCache.block_bytes(cfg, 64, torch.bfloat16, kv_dtype="fp8")
+    assert fp8 < 0.54 * bf16  # ~1.94x the tokens per byte
