@@ -269,15 +269,15 @@ __device__ __forceinline__ void fp8_write_row(
     }
     amax = wave_reduce_max(amax);
     const float sc = amax > 0.f ? amax / FP8_MAX : 1.f;
-    const float inv = 1.f / sc;
     if (lane == 0) *scale_out = sc;
-    // pack per bytes (2 at D=128) and store
+    // exact division (not reciprocal-multiply): bitwise-matches the torch
+    // reference quantizer, so tests compare exactly
     if (per == 2) {
-        unsigned short pk = (unsigned short)f2fp8(vals[0]* inv) |
-                            ((unsigned short)f2fp8(vals[1]* inv) << 8);
+        unsigned short pk = (unsigned short)f2fp8(vals[0] / sc) |
+                            ((unsigned short)f2fp8(vals[1] / sc) << 8);
         reinterpret_cast<unsigned short*>(dst)[lane] = pk;
     } else {
-        for (int j = 0; j < per; ++j) dst[lane * per + j] = f2fp8(vals[j] * inv);
+        for (int j = 0; j < per; ++j) dst[lane * per + j] = f2fp8(vals[j] / sc);
     }
 }
 

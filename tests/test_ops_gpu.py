@@ -609,10 +609,10 @@ def test_fp8_engine_decode_close_to_bf16_gpu():
     to the bf16-KV engine (same seed/weights): first tokens agree."""
     import llmapigateway_amd.engine as E
 
-    a = E.LLMEngine(model="llama-1b", device=DEV, dtype=torch.bfloat16,
+    a = E.LLMEngine(model="llama-3-8b", device=DEV, dtype=torch.bfloat16,
                     max_batch_size=4, max_model_len=128, seed=3,
                     num_blocks=64)
-    b = E.LLMEngine(model="llama-1b", device=DEV, dtype=torch.bfloat16,
+    b = E.LLMEngine(model="llama-3-8b", device=DEV, dtype=torch.bfloat16,
                     max_batch_size=4, max_model_len=128, seed=3,
                     num_blocks=64, kv_dtype="fp8")
     prompt = list(range(5, 37))
