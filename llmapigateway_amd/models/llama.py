@@ -174,6 +174,29 @@ class LlamaModel:
             dist.all_reduce(x, group=self.tp_group)
         return x
 
+    def _row_parallel(self, x: torch.Tensor, w: torch.Tensor,
+                      w_swz=None) -> torch.Tensor:
+        """Row-parallel linear with comm/compute overlap: split the token
+        batch in two, launch the first half's all-reduce asynchronously
+        while the second half's GEMM runs (overlap the 2
+        all-reduces/layer with compute). Falls back to the plain form for
+        tiny batches or TP=1."""
+        if self.tp_group is None:
+            return ops.linear(x, w, w_swz)
+        T = x.shape[0]
+        if T < 32:  # split overhead exceeds the overlap win
+            y = ops.linear(x, w, w_swz)
+            dist.all_reduce(y, group=self.tp_group)
+            return y
+        half = T // 2
+        y1 = ops.linear(x[:half], w, w_swz)
+        h1 = dist.all_reduce(y1, group=self.tp_group, async_op=True)
+        y2 = ops.linear(x[half:], w, w_swz)
+        h2 = dist.all_reduce(y2, group=self.tp_group, async_op=True)
+        h1.wait()
+        h2.wait()
+        return torch.cat([y1, y2], dim=0)
+
     @torch.inference_mode()
     def forward(
         self,
@@ -245,16 +268,14 @@ class LlamaModel:
                     k_scale=ksc[i], v_scale=vsc[i],
                 )
 
-            h = self._maybe_all_reduce(
-                ops.linear(attn.reshape(T, c.q_size), layer["o"], layer.get("o_swz"))
+            h = self._row_parallel(
+                attn.reshape(T, c.q_size), layer["o"], layer.get("o_swz")
             )
 
             x, residual = ops.rmsnorm_residual(h, residual, layer["post_norm"], c.rms_eps)
-            h = self._maybe_all_reduce(
-                ops.linear(
-                    ops.swiglu(ops.linear(x, layer["gate_up"], layer.get("gate_up_swz"))),
-                    layer["down"], layer.get("down_swz"),
-                )
+            h = self._row_parallel(
+                ops.swiglu(ops.linear(x, layer["gate_up"], layer.get("gate_up_swz"))),
+                layer["down"], layer.get("down_swz"),
             )
 
         x, _ = ops.rmsnorm_residual(h, residual, self.final_norm, c.rms_eps)
