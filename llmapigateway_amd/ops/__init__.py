@@ -206,6 +206,9 @@ def attention_prefill(
 
 
 _SPLIT_SPAN = 2048  # keep in sync with SPLIT_SPAN in attention_decode.hip
+# decode-attention geometry: 2 = shared-LDS-chunk / wave-per-head form
+# (coalesced staging, no cross-wave combine); 1 = the round-1 form
+_DECODE_VER = int(os.environ.get("LLMAPI_DECODE_VER", "2"))
 _decode_ws: dict = {}
 _decode_ws_retired: list = []
 
@@ -251,12 +254,12 @@ def attention_decode(
             )
             _native().attention_decode(
                 out, q, k_cache, v_cache, block_tables, context_lens,
-                float(scale), pa, pm, nsplit, k_scale, v_scale,
+                float(scale), pa, pm, nsplit, k_scale, v_scale, _DECODE_VER,
             )
         else:
             _native().attention_decode(
                 out, q, k_cache, v_cache, block_tables, context_lens,
-                float(scale), None, None, 1, k_scale, v_scale,
+                float(scale), None, None, 1, k_scale, v_scale, _DECODE_VER,
             )
         return out
     return reference.attention_decode(

@@ -18,7 +18,7 @@ hipError_t launch_rope(void*, void*, const int64_t*, const float*, int, int64_t,
 hipError_t launch_swiglu(void*, const void*, int, int, hipStream_t);
 hipError_t launch_kv_cache_write(const void*, const void*, void*, void*, float*, float*, const int64_t*, int, int64_t, int64_t, int, int, int, hipStream_t);
 hipError_t launch_rope_kv(void*, void*, const void*, void*, void*, float*, float*, const int64_t*, const float*, const int64_t*, int, int64_t, int64_t, int64_t, int, int, int, int, hipStream_t);
-hipError_t launch_attention_decode(void*, const void*, const void*, const void*, const float*, const float*, const int*, const int*, float, int, int, int, int, int, int, int64_t, float*, float*, int, hipStream_t);
+hipError_t launch_attention_decode(void*, const void*, const void*, const void*, const float*, const float*, const int*, const int*, float, int, int, int, int, int, int, int64_t, float*, float*, int, int, hipStream_t);
 hipError_t launch_attention_prefill(void*, const void*, const void*, const void*, const int*, const int*, const int*, int, float, int, int, int, int64_t, int64_t, int64_t, const void*, const void*, const float*, const float*, const int*, const int*, int, int, hipStream_t);
 hipError_t launch_sample(int64_t*, const float*, const float*, const float*, float*, int*, int, int, hipStream_t);
 hipError_t launch_topk_topp_filter(float*, const float*, const int*, int, int, hipStream_t);
@@ -163,7 +163,7 @@ void attention_decode(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
                       c10::optional<torch::Tensor> part_acc,
                       c10::optional<torch::Tensor> part_ml, int64_t nsplit,
                       c10::optional<torch::Tensor> k_scale,
-                      c10::optional<torch::Tensor> v_scale) {
+                      c10::optional<torch::Tensor> v_scale, int64_t version) {
     check_cache(k_cache, "k_cache");
     TORCH_CHECK((k_cache.scalar_type() == torch::kUInt8) ==
                 (k_scale.has_value() && k_scale->defined()),
@@ -196,7 +196,7 @@ void attention_decode(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
         scale_ptr(k_scale), scale_ptr(v_scale),
         block_tables.data_ptr<int>(), context_lens.data_ptr<int>(), (float)scale,
         B, Hq, Hkv, block_size, max_blocks, D, q.stride(0), pa, pm, (int)nsplit,
-        current_stream()));
+        (int)version, current_stream()));
 }
 
 void attention_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,

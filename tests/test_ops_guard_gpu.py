@@ -105,7 +105,8 @@ def test_attention_decode_guards():
     pa, pacheck = guarded((B * Hq * nsplit * D,), torch.float32)
     pm, pmcheck = guarded((B * Hq * nsplit * 2,), torch.float32)
     ops._native().attention_decode(
-        out, q, kc, vc, bt, ctx, float(D) ** -0.5, pa, pm, nsplit
+        out, q, kc, vc, bt, ctx, float(D) ** -0.5, pa, pm, nsplit, None, None,
+        ops._DECODE_VER,
     )
     check("decode out")
     pacheck("decode part_acc")

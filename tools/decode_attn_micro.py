@@ -31,5 +31,6 @@ if __name__ == "__main__":
     if len(sys.argv) > 1:
         run(int(sys.argv[1]), int(sys.argv[2]) if len(sys.argv) > 2 else 256)
     else:
+        print("decode version", ops._DECODE_VER)
         for c in (192, 512, 2048, 8192):
             run(c)
