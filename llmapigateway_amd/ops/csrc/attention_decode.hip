@@ -405,16 +405,16 @@ __global__ void attention_decode_v2_kernel(
                 fp8x4_to_f32(vq.x, vf[0], vf[1], vf[2], vf[3]);
                 fp8x4_to_f32(vq.y, vf[4], vf[5], vf[6], vf[7]);
                 uint2 kw, vw;
-                kw.x = pack2(kf[0] * ksc, kf[1] * ksc);
-                kw.y = pack2(kf[2] * ksc, kf[3] * ksc);
+                kw.x = pack2_trunc(kf[0] * ksc, kf[1] * ksc);
+                kw.y = pack2_trunc(kf[2] * ksc, kf[3] * ksc);
                 uint2 kw2;
-                kw2.x = pack2(kf[4] * ksc, kf[5] * ksc);
-                kw2.y = pack2(kf[6] * ksc, kf[7] * ksc);
-                vw.x = pack2(vf[0] * vsc, vf[1] * vsc);
-                vw.y = pack2(vf[2] * vsc, vf[3] * vsc);
+                kw2.x = pack2_trunc(kf[4] * ksc, kf[5] * ksc);
+                kw2.y = pack2_trunc(kf[6] * ksc, kf[7] * ksc);
+                vw.x = pack2_trunc(vf[0] * vsc, vf[1] * vsc);
+                vw.y = pack2_trunc(vf[2] * vsc, vf[3] * vsc);
                 uint2 vw2;
-                vw2.x = pack2(vf[4] * vsc, vf[5] * vsc);
-                vw2.y = pack2(vf[6] * vsc, vf[7] * vsc);
+                vw2.x = pack2_trunc(vf[4] * vsc, vf[5] * vsc);
+                vw2.y = pack2_trunc(vf[6] * vsc, vf[7] * vsc);
                 uint4 kq4 = {kw.x, kw.y, kw2.x, kw2.y};
                 uint4 vq4 = {vw.x, vw.y, vw2.x, vw2.y};
                 *reinterpret_cast<uint4*>(&k_lds[key][kchunk * 8]) = kq4;

@@ -60,6 +60,14 @@ __device__ __forceinline__ uint32_t pack2(float lo, float hi) {
     return static_cast<uint32_t>(f2bfbits(lo)) |
            (static_cast<uint32_t>(f2bfbits(hi)) << 16);
 }
+// truncating pack (round-to-zero): 2 ops instead of ~10 — for paths where
+// the input already carries quantization noise (fp8 dequant staging)
+__device__ __forceinline__ uint32_t pack2_trunc(float lo, float hi) {
+    union { float f; uint32_t u; } a, b;
+    a.f = lo;
+    b.f = hi;
+    return (a.u >> 16) | (b.u & 0xFFFF0000u);
+}
 
 // wave-wide reductions (64 lanes)
 __device__ __forceinline__ float wave_reduce_sum(float v) {
