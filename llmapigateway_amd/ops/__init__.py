@@ -252,9 +252,11 @@ def attention_decode(
             pa, pm = _decode_scratch(
                 q.device, B * Hq * nsplit * q.shape[2], B * Hq * nsplit * 2
             )
+            # measured: the v1 geometry wins once the flash-decode split
+            # path engages (small-batch long-context), v2 everywhere else
             _native().attention_decode(
                 out, q, k_cache, v_cache, block_tables, context_lens,
-                float(scale), pa, pm, nsplit, k_scale, v_scale, _DECODE_VER,
+                float(scale), pa, pm, nsplit, k_scale, v_scale, 1,
             )
         else:
             _native().attention_decode(
