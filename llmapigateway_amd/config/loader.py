@@ -78,6 +78,8 @@ class ProviderDetails(BaseModel):
                     updates[key] = int(q[key][0])
             if "dtype" in q:
                 updates["dtype"] = q["dtype"][0]
+            if "kv_dtype" in q:
+                updates["kv_dtype"] = q["kv_dtype"][0]
             if "fail_rate" in q:
                 updates["fail_rate"] = float(q["fail_rate"][0])
             if updates:

@@ -156,3 +156,12 @@ def test_duplicate_gateway_model_last_wins():
 def test_engine_spec_defaults():
     s = EngineSpec()
     assert s.model == "llama-3-8b" and s.tp == 1 and s.fail_rate == 0.0
+
+
+def test_local_url_kv_dtype():
+    from llmapigateway_amd.config.loader import ProviderDetails
+
+    p = ProviderDetails(baseUrl="local://llama-3-8b?device=1&kv_dtype=fp8")
+    spec = p.engine_spec()
+    assert spec.model == "llama-3-8b" and spec.device == 1
+    assert spec.kv_dtype == "fp8"
