@@ -759,6 +759,15 @@ class LLMEngine:
         far (with deferred sampling the in-flight token lags one step —
         an accepted approximation). Returns logits, cloned only if any
         request uses penalties."""
+        # fast path: the common all-greedy/plain batch skips the per-row
+        # python loop entirely (this runs every decode step at b256)
+        if not any(
+            r.params.presence_penalty != 0.0
+            or r.params.frequency_penalty != 0.0
+            or r.params.logit_bias
+            for r in reqs
+        ):
+            return logits
         rows, idxs, vals = [], [], []
         from collections import Counter
 
@@ -810,7 +819,8 @@ class LLMEngine:
             )
             # per-request seeds (OpenAI `seed`): deterministic per output
             # position, independent of batch composition
-            for i, r in enumerate(reqs):
+            seeded = any(r.params.seed is not None for r in reqs)
+            for i, r in enumerate(reqs) if seeded else ():
                 if r.params.seed is not None and r.params.temperature > 0:
                     pos = noise_pos[i] if noise_pos is not None else len(r.out_ids)
                     g = torch.Generator(device=logits.device).manual_seed(
