@@ -86,7 +86,7 @@ def test_attention_prefill_guards():
     tile_seq, tile_off = ops.build_prefill_tiles([L], DEV)
     ops._native().attention_prefill(
         out, q, k, v, cu, tile_seq, tile_off, float(D) ** -0.5,
-        None, None, None, None,
+        None, None, None, None, None, None,
     )
     check("prefill out")
 
