@@ -378,76 +378,54 @@ __global__ void attention_decode_v2_kernel(
 
     const int c0 = span0 / WAVE_SIZE;
     const int nchunks = (span1 + WAVE_SIZE - 1) / WAVE_SIZE;
-    // T14 staging: issue the next chunk's (coalesced) loads right after
-    // re-writing the current chunk, so HBM latency hides under compute.
-    // thread -> (key = pass*16 + tid/16, 16-B chunk = tid%16): each wave
-    // reads 1 KiB contiguous per pass.
-    uint4 kreg[4], vreg[4];          // bf16 path
-    uint2 kreg8[4], vreg8[4];        // fp8 path (8-B rows) + row scales
-    float kscr[4], vscr[4];
-#define DEC_LOAD(C)                                                            \
-    do {                                                                       \
-        _Pragma("unroll") for (int pass = 0; pass < 4; ++pass) {               \
-            const int key__ = (pass << 4) | (tid >> 4);                        \
-            const int chunk__ = tid & 15;                                      \
-            const int pos__ = (C)*WAVE_SIZE + key__;                           \
-            const int cpos__ = min(pos__, span1 - 1);                          \
-            const int blk__ = BT2(cpos__ / block_size);                        \
-            const size_t row__ = ((size_t)blk__ * Hkv + kvh) * block_size +    \
-                                 cpos__ % block_size;                          \
-            if (FP8) {                                                         \
-                kreg8[pass] = reinterpret_cast<const uint2*>(                  \
-                    (const unsigned char*)k_cache + row__ * D)[chunk__];       \
-                vreg8[pass] = reinterpret_cast<const uint2*>(                  \
-                    (const unsigned char*)v_cache + row__ * D)[chunk__];       \
-                kscr[pass] = k_scale[row__];                                   \
-                vscr[pass] = v_scale[row__];                                   \
-            } else {                                                           \
-                kreg[pass] = reinterpret_cast<const uint4*>(                   \
-                    (const bf16*)k_cache + row__ * D)[chunk__];                \
-                vreg[pass] = reinterpret_cast<const uint4*>(                   \
-                    (const bf16*)v_cache + row__ * D)[chunk__];                \
-            }                                                                  \
-        }                                                                      \
-    } while (0)
-#define DEC_WRITE()                                                            \
-    do {                                                                       \
-        _Pragma("unroll") for (int pass = 0; pass < 4; ++pass) {               \
-            const int key__ = (pass << 4) | (tid >> 4);                        \
-            const int chunk__ = tid & 15;                                      \
-            const int kchunk__ = chunk__ ^ (key__ & 15);                       \
-            if (FP8) {                                                         \
-                float kf[8], vf[8];                                            \
-                fp8x4_to_f32(kreg8[pass].x, kf[0], kf[1], kf[2], kf[3]);       \
-                fp8x4_to_f32(kreg8[pass].y, kf[4], kf[5], kf[6], kf[7]);       \
-                fp8x4_to_f32(vreg8[pass].x, vf[0], vf[1], vf[2], vf[3]);       \
-                fp8x4_to_f32(vreg8[pass].y, vf[4], vf[5], vf[6], vf[7]);       \
-                const float ks__ = kscr[pass], vs__ = vscr[pass];              \
-                uint4 kq4, vq4;                                                \
-                kq4.x = pack2_trunc(kf[0] * ks__, kf[1] * ks__);               \
-                kq4.y = pack2_trunc(kf[2] * ks__, kf[3] * ks__);               \
-                kq4.z = pack2_trunc(kf[4] * ks__, kf[5] * ks__);               \
-                kq4.w = pack2_trunc(kf[6] * ks__, kf[7] * ks__);               \
-                vq4.x = pack2_trunc(vf[0] * vs__, vf[1] * vs__);               \
-                vq4.y = pack2_trunc(vf[2] * vs__, vf[3] * vs__);               \
-                vq4.z = pack2_trunc(vf[4] * vs__, vf[5] * vs__);               \
-                vq4.w = pack2_trunc(vf[6] * vs__, vf[7] * vs__);               \
-                *reinterpret_cast<uint4*>(&k_lds[key__][kchunk__ * 8]) = kq4;  \
-                *reinterpret_cast<uint4*>(&v_lds[key__][chunk__ * 8]) = vq4;   \
-            } else {                                                           \
-                *reinterpret_cast<uint4*>(&k_lds[key__][kchunk__ * 8]) =       \
-                    kreg[pass];                                                \
-                *reinterpret_cast<uint4*>(&v_lds[key__][chunk__ * 8]) =        \
-                    vreg[pass];                                                \
-            }                                                                  \
-        }                                                                      \
-    } while (0)
-    if (c0 < nchunks) DEC_LOAD(c0);
     for (int c = c0; c < nchunks; ++c) {
-        // regs hold chunk c (loaded last iteration): publish, then issue
-        // chunk c+1's loads to fly under this chunk's compute
-        DEC_WRITE();
-        if (c + 1 < nchunks) DEC_LOAD(c + 1);
+        // ---- cooperative coalesced staging of the 64-key K/V chunk ----
+        // thread i covers key i/4, 16-B piece i%4 per pass (4 passes of
+        // 64 B per key row = 256 B); K lands XOR-swizzled, V linear.
+        __syncthreads();  // previous chunk's reads complete
+        for (int pass = 0; pass < 4; ++pass) {
+            // 16 lanes per 256-B key row: each wave reads 1 KiB contiguous
+            const int key = (pass << 4) | (tid >> 4);
+            const int chunk = tid & 15;  // 16-B chunk 0..15
+            const int pos = c * WAVE_SIZE + key;
+            const int cpos = min(pos, span1 - 1);
+            const int blk = BT2(cpos / block_size);
+            const size_t row =
+                ((size_t)blk * Hkv + kvh) * block_size + cpos % block_size;
+            const int kchunk = chunk ^ (key & 15);
+            if (FP8) {
+                const uint2 kq = reinterpret_cast<const uint2*>(
+                    (const unsigned char*)k_cache + row * D)[chunk];
+                const uint2 vq = reinterpret_cast<const uint2*>(
+                    (const unsigned char*)v_cache + row * D)[chunk];
+                const float ksc = k_scale[row], vsc = v_scale[row];
+                float kf[8], vf[8];
+                fp8x4_to_f32(kq.x, kf[0], kf[1], kf[2], kf[3]);
+                fp8x4_to_f32(kq.y, kf[4], kf[5], kf[6], kf[7]);
+                fp8x4_to_f32(vq.x, vf[0], vf[1], vf[2], vf[3]);
+                fp8x4_to_f32(vq.y, vf[4], vf[5], vf[6], vf[7]);
+                uint2 kw, vw;
+                kw.x = pack2_trunc(kf[0] * ksc, kf[1] * ksc);
+                kw.y = pack2_trunc(kf[2] * ksc, kf[3] * ksc);
+                uint2 kw2;
+                kw2.x = pack2_trunc(kf[4] * ksc, kf[5] * ksc);
+                kw2.y = pack2_trunc(kf[6] * ksc, kf[7] * ksc);
+                vw.x = pack2_trunc(vf[0] * vsc, vf[1] * vsc);
+                vw.y = pack2_trunc(vf[2] * vsc, vf[3] * vsc);
+                uint2 vw2;
+                vw2.x = pack2_trunc(vf[4] * vsc, vf[5] * vsc);
+                vw2.y = pack2_trunc(vf[6] * vsc, vf[7] * vsc);
+                uint4 kq4 = {kw.x, kw.y, kw2.x, kw2.y};
+                uint4 vq4 = {vw.x, vw.y, vw2.x, vw2.y};
+                *reinterpret_cast<uint4*>(&k_lds[key][kchunk * 8]) = kq4;
+                *reinterpret_cast<uint4*>(&v_lds[key][chunk * 8]) = vq4;
+            } else {
+                *reinterpret_cast<uint4*>(&k_lds[key][kchunk * 8]) =
+                    reinterpret_cast<const uint4*>((const bf16*)k_cache + row * D)[chunk];
+                *reinterpret_cast<uint4*>(&v_lds[key][chunk * 8]) =
+                    reinterpret_cast<const uint4*>((const bf16*)v_cache + row * D)[chunk];
+            }
+        }
         __syncthreads();
 
         // ---- phase A: lane = key (LDS, swizzle-matched reads) ----
@@ -517,11 +495,7 @@ __global__ void attention_decode_v2_kernel(
                 acc[hh][3] = fmaf(pj, v3, acc[hh][3]);
             }
         }
-        // all waves' LDS reads complete before the next DEC_WRITE refill
-        __syncthreads();
     }
-#undef DEC_LOAD
-#undef DEC_WRITE
 
     // fold the half-wave key subsets; each wave writes its own heads
 #pragma unroll
