@@ -207,6 +207,9 @@ class LLMEngine:
             prefix_caching=prefix_caching, kv_dtype=kv_dtype,
         )
         self.prefix_caching = prefix_caching
+        if prefill_budget == 8192:  # default: allow env tuning
+            prefill_budget = int(os.environ.get("LLMAPI_PREFILL_BUDGET", "8192"))
+        self.prefill_budget = prefill_budget
         if admit_min_batch is None:
             admit_min_batch = int(os.environ.get("LLMAPI_ADMIT_MIN", "64"))
         self.admit_min_batch = max(1, admit_min_batch)
