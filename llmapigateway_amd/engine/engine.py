@@ -165,7 +165,7 @@ class LLMEngine:
         tp_rank: int = 0,
         tp_size: int = 1,
         prefix_caching: bool = False,
-        prefill_budget: int = 8192,
+        prefill_budget: int = 4096,
         kv_dtype: str = "auto",
         tokenizer: Optional[object] = None,
         admit_min_batch: Optional[int] = None,
@@ -207,8 +207,8 @@ class LLMEngine:
             prefix_caching=prefix_caching, kv_dtype=kv_dtype,
         )
         self.prefix_caching = prefix_caching
-        if prefill_budget == 8192:  # default: allow env tuning
-            prefill_budget = int(os.environ.get("LLMAPI_PREFILL_BUDGET", "8192"))
+        if prefill_budget == 4096:  # default: allow env tuning
+            prefill_budget = int(os.environ.get("LLMAPI_PREFILL_BUDGET", "4096"))
         self.prefill_budget = prefill_budget
         if admit_min_batch is None:
             admit_min_batch = int(os.environ.get("LLMAPI_ADMIT_MIN", "64"))
