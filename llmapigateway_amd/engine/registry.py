@@ -718,6 +718,7 @@ class EngineRegistry:
                     running=len(eng.running),
                     kv_blocks_free=eng.kv.manager.num_free_blocks,
                     kv_blocks_total=eng.kv.num_blocks,
+                    kv_dtype="fp8" if eng.kv.fp8 else str(eng.dtype).replace("torch.", ""),
                     prefix_cache_hits=eng.kv.manager.stats_prefix_hits,
                     prefix_cached_tokens=eng.kv.manager.stats_prefix_tokens,
                 )
