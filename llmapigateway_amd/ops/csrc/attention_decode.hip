@@ -527,6 +527,240 @@ __global__ void attention_decode_v2_kernel(
     }
 }
 
+template <int G, bool SPLIT, bool FP8>
+__launch_bounds__(NWAVES* WAVE_SIZE)
+__global__ void attention_decode_v3_kernel(
+    bf16* __restrict__ out,                 // [B, Hq, D]
+    const bf16* __restrict__ q,             // [B, Hq, D]
+    const void* __restrict__ k_cache,       // [NB, Hkv, BS, D] bf16|fp8
+    const void* __restrict__ v_cache,
+    const float* __restrict__ k_scale,      // [NB, Hkv, BS] (FP8)
+    const float* __restrict__ v_scale,
+    const int* __restrict__ block_tables,   // [B, max_blocks]
+    const int* __restrict__ context_lens,   // [B]
+    float scale,
+    int Hq,
+    int Hkv,
+    int block_size,
+    int max_blocks,
+    int64_t q_stride,
+    float* __restrict__ part_acc,           // [B, Hq, NSPLIT, D] (SPLIT)
+    float* __restrict__ part_ml,            // [B, Hq, NSPLIT, 2] (SPLIT)
+    int nsplit) {
+    const int seq = blockIdx.x;
+    const int kvh = blockIdx.y;
+    const int split = SPLIT ? blockIdx.z : 0;
+    const int D = DECODE_D;
+    const int L = context_lens[seq];
+    if (L <= 0) return;
+    const int span0 = SPLIT ? split * SPLIT_SPAN : 0;
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE_SIZE - 1);
+    const int wave = tid >> 6;
+    // wave w owns heads {w, w+NWAVES, ...} < G
+    constexpr int HPW = (G + NWAVES - 1) / NWAVES;  // heads per wave (max)
+    if (SPLIT && span0 >= L) {  // idle split: publish empty partials
+#pragma unroll
+        for (int hh = 0; hh < HPW; ++hh) {
+            const int g = wave + hh * NWAVES;
+            if (g >= G) break;
+            const size_t pb = (((size_t)seq * Hq + kvh * G + g) * nsplit + split);
+            if (lane < 2) part_ml[pb * 2 + lane] = lane == 0 ? -INFINITY : 0.f;
+            for (int d = lane; d < D; d += WAVE_SIZE) part_acc[pb * D + d] = 0.f;
+        }
+        return;
+    }
+    const int span1 = SPLIT ? min(L, span0 + SPLIT_SPAN) : L;
+
+    __shared__ float q_lds[G][DECODE_D];
+    __shared__ bf16 k_lds[WAVE_SIZE][DECODE_D];   // 16 KiB (XOR-swizzled)
+    __shared__ float p_lds[G][WAVE_SIZE];
+    __shared__ int bt_lds[1024];
+
+    for (int i = tid; i < G * D; i += blockDim.x) {
+        const int g = i / D, d = i % D;
+        q_lds[g][d] =
+            bf2f(q[(size_t)seq * q_stride + (size_t)(kvh * G + g) * D + d]) * scale;
+    }
+    const int nblocks = (L + block_size - 1) / block_size;
+    for (int i = tid; i < nblocks && i < 1024; i += blockDim.x)
+        bt_lds[i] = block_tables[(size_t)seq * max_blocks + i];
+    __syncthreads();
+    const int* __restrict__ bt_global = block_tables + (size_t)seq * max_blocks;
+#define BT2(idx) ((idx) < 1024 ? bt_lds[(idx)] : bt_global[(idx)])
+
+    float m[HPW], l[HPW], acc[HPW][4];
+#pragma unroll
+    for (int h = 0; h < HPW; ++h) {
+        m[h] = -INFINITY;
+        l[h] = 0.f;
+#pragma unroll
+        for (int d = 0; d < 4; ++d) acc[h][d] = 0.f;
+    }
+
+    const int c0 = span0 / WAVE_SIZE;
+    const int nchunks = (span1 + WAVE_SIZE - 1) / WAVE_SIZE;
+    for (int c = c0; c < nchunks; ++c) {
+        // ---- cooperative coalesced staging of the 64-key K/V chunk ----
+        // thread i covers key i/4, 16-B piece i%4 per pass (4 passes of
+        // 64 B per key row = 256 B); K lands XOR-swizzled, V linear.
+        __syncthreads();  // previous chunk's reads complete
+        for (int pass = 0; pass < 4; ++pass) {
+            // 16 lanes per 256-B key row: each wave reads 1 KiB contiguous
+            const int key = (pass << 4) | (tid >> 4);
+            const int chunk = tid & 15;  // 16-B chunk 0..15
+            const int pos = c * WAVE_SIZE + key;
+            const int cpos = min(pos, span1 - 1);
+            const int blk = BT2(cpos / block_size);
+            const size_t row =
+                ((size_t)blk * Hkv + kvh) * block_size + cpos % block_size;
+            const int kchunk = chunk ^ (key & 15);
+            if (FP8) {
+                const uint2 kq = reinterpret_cast<const uint2*>(
+                    (const unsigned char*)k_cache + row * D)[chunk];
+                const float ksc = k_scale[row];
+                float kf[8];
+                fp8x4_to_f32(kq.x, kf[0], kf[1], kf[2], kf[3]);
+                fp8x4_to_f32(kq.y, kf[4], kf[5], kf[6], kf[7]);
+                uint4 kq4;
+                kq4.x = pack2_trunc(kf[0] * ksc, kf[1] * ksc);
+                kq4.y = pack2_trunc(kf[2] * ksc, kf[3] * ksc);
+                kq4.z = pack2_trunc(kf[4] * ksc, kf[5] * ksc);
+                kq4.w = pack2_trunc(kf[6] * ksc, kf[7] * ksc);
+                *reinterpret_cast<uint4*>(&k_lds[key][kchunk * 8]) = kq4;
+            } else {
+                *reinterpret_cast<uint4*>(&k_lds[key][kchunk * 8]) =
+                    reinterpret_cast<const uint4*>((const bf16*)k_cache + row * D)[chunk];
+            }
+        }
+        __syncthreads();
+
+        // ---- phase A: lane = key (LDS, swizzle-matched reads) ----
+        const int pos = c * WAVE_SIZE + lane;
+        const bool valid = pos < span1;
+#pragma unroll
+        for (int hh = 0; hh < HPW; ++hh) {
+            const int g = wave + hh * NWAVES;
+            if (g >= G) break;
+            float sg = valid ? 0.f : -INFINITY;
+            if (valid) {
+#pragma unroll 4
+                for (int i = 0; i < D / 8; ++i) {
+                    const bf16x8_d kv8 = *reinterpret_cast<const bf16x8_d*>(
+                        &k_lds[lane][(i ^ (lane & 15)) * 8]);
+                    const uint4 kvu = *reinterpret_cast<const uint4*>(&kv8);
+                    float kf[8];
+                    unpack2(kvu.x, kf[0], kf[1]);
+                    unpack2(kvu.y, kf[2], kf[3]);
+                    unpack2(kvu.z, kf[4], kf[5]);
+                    unpack2(kvu.w, kf[6], kf[7]);
+                    const float4 qa = *reinterpret_cast<const float4*>(&q_lds[g][i * 8]);
+                    const float4 qb = *reinterpret_cast<const float4*>(&q_lds[g][i * 8 + 4]);
+                    sg = fmaf(qa.x, kf[0], sg);
+                    sg = fmaf(qa.y, kf[1], sg);
+                    sg = fmaf(qa.z, kf[2], sg);
+                    sg = fmaf(qa.w, kf[3], sg);
+                    sg = fmaf(qb.x, kf[4], sg);
+                    sg = fmaf(qb.y, kf[5], sg);
+                    sg = fmaf(qb.z, kf[6], sg);
+                    sg = fmaf(qb.w, kf[7], sg);
+                }
+            }
+            const float cmax = wave_reduce_max(sg);
+            const float m_new = fmaxf(m[hh], cmax);
+            float pv = 0.f;
+            if (valid && m_new != -INFINITY) pv = __expf(sg - m_new);
+            const float factor = (m[hh] == -INFINITY) ? 0.f : __expf(m[hh] - m_new);
+            const float csum = wave_reduce_sum(pv);
+            l[hh] = l[hh] * factor + csum;
+#pragma unroll
+            for (int d = 0; d < 4; ++d) acc[hh][d] *= factor;
+            m[hh] = m_new;
+            p_lds[g][lane] = pv;
+        }
+        // p_lds is wave-private per head (one writer wave), read below by
+        // the same wave only — no barrier needed before phase B
+
+        // ---- phase B: half-wave per key, lane owns a dim quad, V read
+        // straight from global (coalesced 256-B half-wave segments; the
+        // LDS round trip bought nothing for a once-read operand) ----
+        const int dbase = (lane & 31) * 4;
+        const int khalf = lane >> 5;
+#pragma unroll 2
+        for (int j = 0; j < WAVE_SIZE; j += 8) {
+            uint2 vp[4];
+            uint32_t vp8[4];
+            float vsc4[4];
+#pragma unroll
+            for (int u = 0; u < 4; ++u) {
+                const int pos2 = min(c * WAVE_SIZE + j + 2 * u + khalf, span1 - 1);
+                const int vb = BT2(pos2 / block_size);
+                const int vo = pos2 % block_size;
+                const size_t vrow = ((size_t)vb * Hkv + kvh) * block_size + vo;
+                if (FP8) {
+                    vp8[u] = *reinterpret_cast<const uint32_t*>(
+                        (const unsigned char*)v_cache + vrow * D + dbase);
+                    vsc4[u] = v_scale[vrow];
+                } else {
+                    vp[u] = *reinterpret_cast<const uint2*>(
+                        (const bf16*)v_cache + vrow * D + dbase);
+                }
+            }
+#pragma unroll
+            for (int u = 0; u < 4; ++u) {
+                float v0, v1, v2, v3;
+                if (FP8) {
+                    fp8x4_to_f32(vp8[u], v0, v1, v2, v3);
+                } else {
+                    unpack2(vp[u].x, v0, v1);
+                    unpack2(vp[u].y, v2, v3);
+                }
+                const int key = j + 2 * u + khalf;
+#pragma unroll
+                for (int hh = 0; hh < HPW; ++hh) {
+                    const int g = wave + hh * NWAVES;
+                    if (g >= G) break;
+                    const float pj = p_lds[g][key] * (FP8 ? vsc4[u] : 1.f);
+                    acc[hh][0] = fmaf(pj, v0, acc[hh][0]);
+                    acc[hh][1] = fmaf(pj, v1, acc[hh][1]);
+                    acc[hh][2] = fmaf(pj, v2, acc[hh][2]);
+                    acc[hh][3] = fmaf(pj, v3, acc[hh][3]);
+                }
+            }
+        }
+    }
+
+    // fold the half-wave key subsets; each wave writes its own heads
+#pragma unroll
+    for (int hh = 0; hh < HPW; ++hh) {
+        const int g = wave + hh * NWAVES;
+        if (g >= G) break;
+#pragma unroll
+        for (int d = 0; d < 4; ++d) acc[hh][d] += __shfl_xor(acc[hh][d], 32, WAVE_SIZE);
+        if (lane < 32) {
+            if (SPLIT) {
+                const size_t pb = (((size_t)seq * Hq + kvh * G + g) * nsplit + split);
+#pragma unroll
+                for (int d = 0; d < 4; ++d)
+                    part_acc[pb * DECODE_D + (lane & 31) * 4 + d] = acc[hh][d];
+                if (lane == 0) {
+                    part_ml[pb * 2] = m[hh];
+                    part_ml[pb * 2 + 1] = l[hh];
+                }
+            } else {
+                const float inv_l = (l[hh] > 0.f) ? 1.f / l[hh] : 0.f;
+                uint2 o2;
+                o2.x = pack2(acc[hh][0] * inv_l, acc[hh][1] * inv_l);
+                o2.y = pack2(acc[hh][2] * inv_l, acc[hh][3] * inv_l);
+                *reinterpret_cast<uint2*>(
+                    out + (size_t)seq * Hq * DECODE_D +
+                    (size_t)(kvh * G + g) * DECODE_D + (lane & 31) * 4) = o2;
+            }
+        }
+    }
+}
+
+
 // merge the per-split partials: out[b,h] = sum_z exp(m_z - M) acc_z / L
 __global__ void decode_combine_kernel(
     bf16* __restrict__ out,            // [B, Hq, D]
