@@ -187,6 +187,7 @@ class LLMEngine:
             # real BPE for production presets, byte stand-in for tiny ones
             tokenizer = get_tokenizer("auto", full_config.vocab_size)
         self.tokenizer = tokenizer  # stop-string decode + EOS detection
+        self._eos_np = np.asarray(sorted(tokenizer.eos_ids), dtype=np.int64)
 
         if self.device.type == "cuda":
             torch.cuda.set_device(self.device)
@@ -271,6 +272,15 @@ class LLMEngine:
         maxb = (self.max_model_len + block_size - 1) // block_size
         self._bt_np = np.zeros((max_batch_size, maxb), dtype=np.int32)
         self._slot_pool = list(range(max_batch_size - 1, -1, -1))
+        # vectorized per-slot bookkeeping (the python per-request loops in
+        # the decode hot path measured ~0.4 ms/step at b256): committed
+        # token count, last committed token, absolute caps, block count
+        self._slot_ntok = np.zeros(max_batch_size, dtype=np.int64)
+        self._slot_last = np.zeros(max_batch_size, dtype=np.int64)
+        self._slot_maxt = np.zeros(max_batch_size, dtype=np.int64)  # prompt+max_tokens
+        self._slot_nblk = np.zeros(max_batch_size, dtype=np.int32)
+        # bit0 = ignore_eos, bit1 = has stop strings
+        self._slot_flags = np.zeros(max_batch_size, dtype=np.int8)
 
         # optional batched token-event sink (set by the gateway registry):
         # called under the engine lock with [(req, token), ...] per step
@@ -383,6 +393,14 @@ class LLMEngine:
             req.prefill_pos = req.num_cached
             req.bt_slot = self._slot_pool.pop()
             self._bt_np[req.bt_slot, : len(req.block_table)] = req.block_table
+            self._slot_ntok[req.bt_slot] = len(req.prompt_ids)
+            self._slot_last[req.bt_slot] = req.prompt_ids[-1]
+            self._slot_maxt[req.bt_slot] = len(req.prompt_ids) + req.params.max_tokens
+            self._slot_nblk[req.bt_slot] = len(req.block_table)
+            self._slot_flags[req.bt_slot] = (
+                (1 if req.params.ignore_eos else 0)
+                | (2 if req.params.stop else 0)
+            )
             admitted.append(req)
         return admitted
 
@@ -509,6 +527,7 @@ class LLMEngine:
                             self._bt_np[req.bt_slot, before : len(req.block_table)] = (
                                 req.block_table[before:]
                             )
+                            self._slot_nblk[req.bt_slot] = len(req.block_table)
                         i += 1
                     except RuntimeError:
                         if not self._preempt_youngest():
@@ -639,11 +658,17 @@ class LLMEngine:
         # list; any mismatch or limit-crossing request forces a flush
         if self._pending is not None:
             same = self._pending[0] == reqs_now
-            capped = same and any(
-                len(r.out_ids) + 1 >= r.params.max_tokens
-                or r.num_tokens + 2 > self.max_model_len
-                for r in reqs_now
-            )
+            if same:
+                rows0 = np.fromiter(
+                    (r.bt_slot for r in reqs_now), dtype=np.intp, count=len(reqs_now)
+                )
+                nt0 = self._slot_ntok[rows0]
+                capped = bool(
+                    ((nt0 + 1 >= self._slot_maxt[rows0])
+                     | (nt0 + 2 > self.max_model_len)).any()
+                )
+            else:
+                capped = False
             if not same or capped:
                 self._flush_pending()
                 with self._lock:
@@ -653,27 +678,52 @@ class LLMEngine:
         inflight = 1 if self._pending is not None else 0
 
         with self._lock:
-            # ensure every running seq has a block for the incoming token
-            i = 0
-            while i < len(self.running):
-                req = self.running[i]
-                nt = req.num_tokens + inflight
+            # ensure every running seq has a block for the incoming token:
+            # vectorized need-check, python only for the rows that grow
+            nrun = len(self.running)
+            rows_all = np.fromiter(
+                (r.bt_slot for r in self.running), dtype=np.intp, count=nrun
+            )
+            nt_all = self._slot_ntok[rows_all] + inflight
+            need = (nt_all + bs) // bs  # blocks_needed(nt + 1)
+            grow = np.nonzero(need > self._slot_nblk[rows_all])[0]
+            i = -1
+            for gi in grow.tolist():
+                req = self.running[gi] if gi < len(self.running) else None
+                if req is None or req.bt_slot is None:
+                    continue
+                nt = int(self._slot_ntok[req.bt_slot]) + inflight
                 try:
                     before = len(req.block_table)
                     self.kv.manager.extend(req.block_table, nt, nt + 1)
-                    if len(req.block_table) != before:  # keep the bt row in sync
+                    if len(req.block_table) != before:
                         self._bt_np[req.bt_slot, before : len(req.block_table)] = (
                             req.block_table[before:]
                         )
-                    i += 1
+                        self._slot_nblk[req.bt_slot] = len(req.block_table)
                 except RuntimeError:
                     if inflight:
-                        break  # flush below, then retry the whole step
-                    if not self._preempt_youngest():
-                        raise
-                    # if we preempted the request we were extending, skip it
-            else:
-                i = -1  # completed without break
+                        i = 0  # flush below, then retry the whole step
+                        break
+                    # rare pressure path: fall back to the full python
+                    # loop with preemption (unchanged semantics)
+                    j = 0
+                    while j < len(self.running):
+                        r2 = self.running[j]
+                        nt2 = r2.num_tokens + inflight
+                        try:
+                            b2 = len(r2.block_table)
+                            self.kv.manager.extend(r2.block_table, nt2, nt2 + 1)
+                            if len(r2.block_table) != b2:
+                                self._bt_np[r2.bt_slot, b2 : len(r2.block_table)] = (
+                                    r2.block_table[b2:]
+                                )
+                                self._slot_nblk[r2.bt_slot] = len(r2.block_table)
+                            j += 1
+                        except RuntimeError:
+                            if not self._preempt_youngest():
+                                raise
+                    break
             reqs = list(self.running)
         if i != -1:  # needed preemption while tokens were in flight
             self._flush_pending()
@@ -682,10 +732,8 @@ class LLMEngine:
             return 0
 
         n = len(reqs)
-        pos = np.fromiter(
-            (req.num_tokens + inflight - 1 for req in reqs), dtype=np.int64, count=n
-        )
         slot_rows = np.fromiter((req.bt_slot for req in reqs), dtype=np.intp, count=n)
+        pos = self._slot_ntok[slot_rows] + (inflight - 1)
         tables_np = self._bt_np[slot_rows]  # [n, maxb] vectorized gather
         blk = pos // bs
         slots = tables_np[np.arange(n), blk].astype(np.int64) * bs + pos % bs
@@ -693,10 +741,7 @@ class LLMEngine:
         if inflight:
             last_tokens = self._pend_tokens_dev  # device int64 [n], aligned
         else:
-            last_tokens = np.fromiter(
-                ((req.out_ids[-1] if req.out_ids else req.prompt_ids[-1]) for req in reqs),
-                dtype=np.int64, count=n,
-            )
+            last_tokens = self._slot_last[slot_rows].copy()
 
         self.stats["decode_steps"] += 1
         logits = None
@@ -915,10 +960,15 @@ class LLMEngine:
         # event (the gateway registers batch_notifier — per-token
         # call_soon_threadsafe wakeups saturate the event loop ~10K/s)
         events = [] if self.batch_notifier is not None else None
+        rows: List[int] = []
+        row_toks: List[int] = []
         for req, tok in zip(reqs, tokens):
             if req.state != "running":
                 continue
             req.out_ids.append(int(tok))
+            if req.bt_slot is not None:
+                rows.append(req.bt_slot)
+                row_toks.append(int(tok))
             if req.params.stop:
                 req.text += self.tokenizer.decode([int(tok)])
             if req.first_token_time is None:
@@ -930,13 +980,40 @@ class LLMEngine:
                     logger.exception("on_token callback failed for %s", req.id)
             elif events is not None:
                 events.append((req, int(tok)))
+        if rows:
+            r = np.asarray(rows, dtype=np.intp)
+            self._slot_ntok[r] += 1
+            self._slot_last[r] = row_toks
         if events:
             try:
                 self.batch_notifier(events)
             except Exception:
                 logger.exception("batch_notifier failed")
-        # pass 2: finish checks (finish callbacks are queued after the
-        # batched token event, preserving token->finish order per request)
+        # pass 2: finish checks. The vectorized prefilter covers the
+        # common finish causes (length caps, EOS for non-ignore_eos);
+        # requests with stop strings always take the python path.
+        if rows and len(reqs) == len(rows):
+            nt = self._slot_ntok[r]
+            flags = self._slot_flags[r]
+            tarr = np.asarray(row_toks, dtype=np.int64)
+            needs_py = (
+                (nt >= self._slot_maxt[r])
+                | (nt >= self.max_model_len)
+                | (np.isin(tarr, self._eos_np) & ((flags & 1) == 0))
+                | ((flags & 2) != 0)
+            )
+            idxs = np.nonzero(needs_py)[0]
+            if idxs.size == 0:
+                return
+            for i in idxs.tolist():
+                req, tok = reqs[i], tokens[i]
+                if req.state != "running":
+                    continue
+                reason = self._finish_reason(req, int(tok))
+                if reason:
+                    self.running.remove(req)
+                    self._finish(req, reason)
+            return
         for req, tok in zip(reqs, tokens):
             if req.state != "running":
                 continue
