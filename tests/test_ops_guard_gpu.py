@@ -159,3 +159,47 @@ def test_topk_topp_filter_guards():
     topk = torch.full((B,), 40, dtype=torch.int32, device=DEV)
     ops._native().topk_topp_filter(logits, topp, topk)
     check("filter logits")
+
+
+def test_fp8_kv_write_guards():
+    T, Hq, Hkv, D, BS, NB = 23, 8, 2, 128, 16, 8
+    qkv = torch.randn(T, (Hq + 2 * Hkv) * D, dtype=torch.bfloat16, device=DEV)
+    q = qkv[:, : Hq * D].view(T, Hq, D)
+    k = qkv[:, Hq * D : (Hq + Hkv) * D].view(T, Hkv, D)
+    v = qkv[:, (Hq + Hkv) * D :].view(T, Hkv, D)
+    kc, kcheck = guarded((NB, Hkv, BS, D), torch.uint8)
+    vc, vcheck = guarded((NB, Hkv, BS, D), torch.uint8)
+    ks, kscheck = guarded((NB, Hkv, BS), torch.float32)
+    vs, vscheck = guarded((NB, Hkv, BS), torch.float32)
+    pos = torch.arange(T, dtype=torch.int64, device=DEV)
+    slots = torch.randperm(NB * BS, device=DEV)[:T]
+    slots[0] = -1
+    slots[1] = NB * BS - 1  # boundary row + scale
+    cs = ops.build_rope_cache(64, D, 10000.0, device=DEV)
+    ops.rope_and_kv_write(q, k, v, kc, vc, pos, cs, slots, k_scale=ks, v_scale=vs)
+    kcheck("fp8 k_cache")
+    vcheck("fp8 v_cache")
+    kscheck("fp8 k_scale")
+    vscheck("fp8 v_scale")
+
+
+def test_fp8_decode_v_scale_guards():
+    B, Hq, Hkv, BS, D = 4, 8, 2, 16, 128
+    maxb = 4
+    NB = B * maxb + 2
+    bits = torch.randint(1, 126, (NB, Hkv, BS, D), dtype=torch.uint8, device=DEV)
+    ks, kscheck = guarded((NB, Hkv, BS), torch.float32)
+    vs, vscheck = guarded((NB, Hkv, BS), torch.float32)
+    ks.fill_(0.01)
+    vs.fill_(0.01)
+    bt = torch.randperm(NB)[: B * maxb].view(B, maxb).int().to(DEV)
+    q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=DEV)
+    ctx = torch.tensor([5, 16, 33, 60], dtype=torch.int32, device=DEV)
+    out, ocheck = guarded((B, Hq, D), torch.bfloat16)
+    ops._native().attention_decode(
+        out, q, bits, bits, bt, ctx, float(D) ** -0.5, None, None, 1,
+        ks, vs, ops._DECODE_VER,
+    )
+    ocheck("fp8 decode out")
+    kscheck("fp8 decode k_scale")
+    vscheck("fp8 decode v_scale")
