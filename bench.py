@@ -62,8 +62,10 @@ def parse_args():
     p.add_argument("--debug-stats", action="store_true")
     p.add_argument(
         "--qps", default="auto",
-        help="offered request rate per GPU for --mode qps; 'auto' = 1.1x "
-        "capacity estimated from the warmup waves",
+        help="offered request rate per GPU for --mode qps; 'auto' runs an "
+        "untimed open-loop calibration segment after the wave warmups and "
+        "offers exactly the achieved open-loop capacity, so the timed "
+        "region is saturated with a bounded queue (stable TTFT)",
     )
     return p.parse_args()
 
@@ -190,8 +192,19 @@ def main():
     qps = None
     if args.mode == "qps":
         if args.qps == "auto":
+            # calibration: a short untimed open-loop segment at 1.2x the
+            # wave estimate measures the OPEN-LOOP capacity (lower than
+            # wave capacity: continuous prefill/decode interleaving);
+            # offering exactly that keeps the timed queue near-critical
+            # so p50/p95 TTFT are stable properties, not functions of
+            # the run length
             cap = max(wave_rates) if wave_rates else 50.0
-            qps = cap * 1.1  # saturate: achieved req/s measures capacity
+            _, cal_elapsed = run_open_loop(
+                engine,
+                make_prompts(args.batch, args.prompt_len, vocab, 555 + rank),
+                args.gen_tokens, cap * 1.2, seed=31 + rank, sp=sp,
+            )
+            qps = args.batch / cal_elapsed
         else:
             qps = float(args.qps)
         # every rank must offer the same load: agree on rank 0's value
