@@ -26,6 +26,11 @@ SHAPES = [
 
 
 def bench(fn, iters=30):
+    # warmup OUTSIDE the timed region: first-call hipBLASLt algo selection /
+    # TunableOp tuning on a never-seen shape costs 100s of ms and would
+    # otherwise dominate the 30-iter average.
+    for i in range(3):
+        fn(i)
     torch.cuda.synchronize()
     t0 = time.monotonic()
     for i in range(iters):
