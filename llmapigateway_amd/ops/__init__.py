@@ -341,10 +341,18 @@ def _m256_config(M: int, N: int, K: int) -> Optional[dict]:
     when the tuned library wins the shape (gate_up N=28672, lm_head)."""
     if M > 256:
         return None
+    if N == 8192 and K % 64 == 0 and K >= 8192:
+        # 70B-class projections at tp=1 (hidden 8192): o-proj K=8192
+        # (51.2 vs library 58.2 us) and down-proj K=28672 (147.5 vs
+        # 192.4 us) — register-staged variant wins the deep-K shape
+        if K > 8192:
+            return {"nf": 8, "nsk": 4, "variant": 1, "pipe": 0}  # 70B down 1.30x
+        return {"nf": 8, "nsk": 4, "variant": 0, "pipe": 0}      # 70B o 1.14x
     if N > 4096:
-        # library wins the wide shapes: gate_up N=28672 (custom best
-        # 73.3 vs 70.3 us) and qkv N=6144 (36.4 vs ~28 us — hipBLASLt's
-        # MT112x256 kernel is strong exactly there)
+        # library wins the very wide shapes: gate_up N=28672/57344
+        # (custom best 73.3 vs 70.3 us / 287.9 vs 258.0), qkv N=6144
+        # (36.4 vs ~28 us — hipBLASLt's MT112x256 kernel is strong
+        # exactly there) and N=10240 (77.7 vs 66.7), lm_head
         return None
     if K > 8192 and N % 128 == 0:
         return {"nf": 8, "nsk": 8, "variant": 0, "pipe": 0}  # down 1.34x
