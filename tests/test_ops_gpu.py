@@ -313,6 +313,24 @@ def test_gemm_m256_via_linear_twin_dispatch():
     assert err < 0.02
 
 
+@pytest.mark.parametrize("N,K", [(8192, 8192), (8192, 28672)])
+def test_gemm_m256_70b_dispatch(N, K):
+    # 70B-class projections (hidden 8192) route to the custom kernel per
+    # the measured table (profiles/r02_gemm_m256_sweep.md)
+    from llmapigateway_amd.ops import _m256_config
+
+    assert _m256_config(256, N, K) is not None
+    M = 256
+    torch.manual_seed(N + K)
+    x = (torch.randn(M, K, device=DEV) * 0.5).bfloat16()
+    w = (torch.randn(N, K, device=DEV) * 0.02).bfloat16()
+    wf = ops.swizzle_weight_frag(w)
+    got = ops.linear(x, w, wf)
+    ref = x.float() @ w.float().T
+    err = (got.float() - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
+    assert err < 0.02
+
+
 def test_gemm_m256_repeat_launches_stable():
     # split-K slab reuse across launches (hipGraph replay pattern)
     M, N, K = 64, 4096, 14336
