@@ -157,7 +157,15 @@ class LlamaModel:
                     if legacy:
                         if w.shape[0] % 64 == 0 and w.shape[1] % 32 == 0:
                             layer[name + "_swz"] = ops.swizzle_weight(w)
-                    elif w.shape[0] % 64 == 0 and w.shape[1] % 64 == 0:
+                    elif (
+                        w.shape[0] % 64 == 0
+                        and w.shape[1] % 64 == 0
+                        # only shapes the measured dispatch actually routes to
+                        # the custom kernel get a twin — qkv/gate_up twins
+                        # would be dead HBM (the library wins those shapes),
+                        # ~9 GB at 8B and ~89 GB at 70B
+                        and ops._m256_config(256, w.shape[0], w.shape[1]) is not None
+                    ):
                         layer[name + "_swz"] = ops.swizzle_weight_frag(w)
 
     def param_bytes(self) -> int:
