@@ -1,0 +1,39 @@
+"""The measured GEMM dispatch table (ops._m256_config) is data the
+serving hot path depends on: these CPU tests pin which shapes route to
+the custom macro-tile kernel vs the tuned library so a table edit can't
+silently drop a measured entry (profiles/r02_gemm_m256_sweep.md)."""
+
+import pytest
+
+from llmapigateway_amd.ops import _m256_config
+
+
+# llama-3-8b tp=1 decode projections (hidden 4096, inter 14336)
+def test_8b_routing():
+    assert _m256_config(256, 6144, 4096) is None        # qkv -> library
+    assert _m256_config(256, 4096, 4096) is not None    # o -> custom
+    assert _m256_config(256, 28672, 4096) is None       # gate_up -> library
+    assert _m256_config(256, 4096, 14336) is not None   # down -> custom
+
+
+# llama-3-70b tp=1 decode projections (hidden 8192, inter 28672)
+def test_70b_routing():
+    assert _m256_config(256, 10240, 8192) is None       # qkv -> library
+    cfg = _m256_config(256, 8192, 8192)                 # o -> custom
+    assert cfg == {"nf": 8, "nsk": 4, "variant": 0, "pipe": 0}
+    assert _m256_config(256, 57344, 8192) is None       # gate_up -> library
+    cfg = _m256_config(256, 8192, 28672)                # down -> custom
+    assert cfg == {"nf": 8, "nsk": 4, "variant": 1, "pipe": 0}
+
+
+def test_prefill_m_goes_to_library():
+    # chunked-prefill token counts exceed the macro-tile M ceiling
+    assert _m256_config(4096, 4096, 4096) is None
+
+
+@pytest.mark.parametrize("N,K", [(4096, 4096), (4096, 14336), (8192, 8192), (8192, 28672)])
+def test_custom_configs_are_launchable_shapes(N, K):
+    cfg = _m256_config(256, N, K)
+    assert cfg is not None
+    assert N % (16 * cfg["nf"]) == 0          # whole column tiles
+    assert (K // 64) % cfg["nsk"] == 0 or cfg["nsk"] <= (K // 64)  # split-K fits
