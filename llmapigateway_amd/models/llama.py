@@ -167,6 +167,17 @@ class LlamaModel:
                         and ops._m256_config(256, w.shape[0], w.shape[1]) is not None
                     ):
                         layer[name + "_swz"] = ops.swizzle_weight_frag(w)
+                gu = layer["gate_up"]
+                if (
+                    not legacy
+                    and ops._m256_swiglu_config(256, gu.shape[0], gu.shape[1])
+                    is not None
+                ):
+                    # block-16 interleaved twin for the FUSED gate_up+swiglu
+                    # decode path (ops.swiglu_linear)
+                    layer["gate_up_int"] = ops.swizzle_weight_frag(
+                        ops.interleave_gate_up(gu)
+                    )
 
     def param_bytes(self) -> int:
         total = self.embed.numel() + self.final_norm.numel()
@@ -282,7 +293,10 @@ class LlamaModel:
 
             x, residual = ops.rmsnorm_residual(h, residual, layer["post_norm"], c.rms_eps)
             h = self._row_parallel(
-                ops.swiglu(ops.linear(x, layer["gate_up"], layer.get("gate_up_swz"))),
+                ops.swiglu_linear(
+                    x, layer["gate_up"], layer.get("gate_up_swz"),
+                    layer.get("gate_up_int"),
+                ),
                 layer["down"], layer.get("down_swz"),
             )
 

@@ -416,8 +416,64 @@ def gemm_m256(
             pipe = 0
     y = torch.empty((M, N), dtype=torch.bfloat16, device=x.device)
     ws = _skinny_scratch(x.device, nsk * M * N) if nsk > 1 else None
-    _native().gemm_m256(y, x, w_frag, ws, nsk, nf, variant, pipe)
+    _native().gemm_m256(y, x, w_frag, ws, nsk, nf, variant, pipe, 0)
     return y
+
+
+def interleave_gate_up(w: torch.Tensor) -> torch.Tensor:
+    """Row-permute a [gate; up] stacked weight [2I, K] into block-16
+    interleaved order [g0..15, u0..15, g16..31, ...]: after
+    swizzle_weight_frag, the gate/up accumulators for the same output
+    column land in the SAME lane of adjacent 16-col fragments, which is
+    what gemm_m256's fused swiglu epilogue consumes."""
+    two_i, K = w.shape
+    assert two_i % 32 == 0
+    return w.view(2, two_i // 32, 16, K).transpose(0, 1).reshape(two_i, K)
+
+
+def gemm_m256_swiglu(
+    x: torch.Tensor, w_frag: torch.Tensor, nf: int = 8, variant: int = 1,
+    pipe: int = 0,
+) -> torch.Tensor:
+    """silu(x @ gate.T) * (x @ up.T) in ONE kernel: w_frag is
+    swizzle_weight_frag(interleave_gate_up(gate_up)) and the swiglu runs
+    in the GEMM epilogue on the fp32 accumulators — the [M, 2I]
+    intermediate round trip and the separate swiglu kernel disappear
+    (~72 MB of HBM traffic per llama-3-8b layer at M=256)."""
+    M, K = x.shape
+    N = w_frag.shape[1] * 16
+    y = torch.empty((M, N // 2), dtype=torch.bfloat16, device=x.device)
+    _native().gemm_m256(y, x, w_frag, None, 1, nf, variant, pipe, 1)
+    return y
+
+
+def _m256_swiglu_config(M: int, N: int, K: int) -> Optional[dict]:
+    """Measured dispatch for the FUSED gate_up+swiglu decode MLP front
+    (profiles/r02_gemm_m256_sweep.md). N is the stacked 2*intermediate.
+    Only the 8B shape measured ahead of library+swiglu (77.2 vs 79.4 us
+    cold-L3); the 70B shape loses (304.8 vs 226.6) and stays unfused."""
+    if os.environ.get("LLMAPI_NO_FUSED_SWIGLU"):
+        return None
+    if not (8 < M <= 256):
+        return None
+    if N == 28672 and K == 4096:  # llama-3-8b tp=1
+        return {"nf": 8, "variant": 1, "pipe": 0}
+    return None
+
+
+def swiglu_linear(
+    x: torch.Tensor, w: torch.Tensor,
+    w_swz: Optional[torch.Tensor] = None,
+    w_int: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """MLP front: swiglu(x @ w.T) with w = [gate; up] stacked. Routes to
+    the fused gemm_m256 epilogue when the interleaved twin exists and the
+    measured table says it wins; otherwise library GEMM + swiglu kernel."""
+    if w_int is not None and x.dim() == 2:
+        cfg = _m256_swiglu_config(x.shape[0], w_int.shape[1] * 16, x.shape[1])
+        if cfg is not None:
+            return gemm_m256_swiglu(x, w_int, **cfg)
+    return swiglu(linear(x, w, w_swz))
 
 
 def linear(

@@ -23,7 +23,7 @@ hipError_t launch_attention_prefill(void*, const void*, const void*, const void*
 hipError_t launch_sample(int64_t*, const float*, const float*, const float*, float*, int*, int, int, hipStream_t);
 hipError_t launch_topk_topp_filter(float*, const float*, const int*, int, int, hipStream_t);
 hipError_t launch_gemm_skinny(void*, float*, const void*, const void*, int, int, int, int, int, hipStream_t);
-hipError_t launch_gemm_m256(void*, float*, const void*, const void*, int, int, int, int, int, int, int, hipStream_t);
+hipError_t launch_gemm_m256(void*, float*, const void*, const void*, int, int, int, int, int, int, int, int, hipStream_t);
 }
 
 namespace {
@@ -313,7 +313,7 @@ void gemm_skinny(torch::Tensor y, torch::Tensor x, torch::Tensor w,
 
 void gemm_m256(torch::Tensor y, torch::Tensor x, torch::Tensor w_frag,
                c10::optional<torch::Tensor> workspace, int64_t nsk,
-               int64_t nf, int64_t variant, int64_t pipe) {
+               int64_t nf, int64_t variant, int64_t pipe, int64_t swiglu) {
     check_bf16(x, "x");
     check_bf16(w_frag, "w_frag");
     check_bf16(y, "y");
@@ -325,7 +325,9 @@ void gemm_m256(torch::Tensor y, torch::Tensor x, torch::Tensor w_frag,
                 w_frag.size(2) == 64 && w_frag.size(3) == 8,
                 "w_frag must be the fragment-major [K/32, N/16, 64, 8] twin");
     const int N = w_frag.size(1) * 16;
-    TORCH_CHECK(y.size(0) == M && y.size(1) == N);
+    // fused swiglu epilogue: w_frag is the block-16 interleaved gate/up
+    // twin and y holds silu(gate)*up, half the columns
+    TORCH_CHECK(y.size(0) == M && y.size(1) == (swiglu ? N / 2 : N));
     float* ws = nullptr;
     if (workspace.has_value() && workspace->defined()) {
         TORCH_CHECK(workspace->scalar_type() == torch::kFloat32 &&
@@ -335,7 +337,8 @@ void gemm_m256(torch::Tensor y, torch::Tensor x, torch::Tensor w_frag,
     }
     CHECK_HIP(launch_gemm_m256(y.data_ptr(), ws, x.data_ptr(),
                                w_frag.data_ptr(), M, N, K, (int)nsk, (int)nf,
-                               (int)variant, (int)pipe, current_stream()));
+                               (int)variant, (int)pipe, (int)swiglu,
+                               current_stream()));
 }
 
 }  // namespace

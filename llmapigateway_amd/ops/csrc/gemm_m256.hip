@@ -65,6 +65,10 @@ __device__ __forceinline__ void gm_vmwait() {
 // compiler memory barrier)
 __device__ __forceinline__ void gm_cfence() { asm volatile("" ::: "memory"); }
 
+__device__ __forceinline__ float gm_silu(float g) {
+    return g / (1.f + __expf(-g));
+}
+
 // counted wait for "ahead" tiles of GPW glds each still in flight
 // (compile-time unrolled: vmcnt immediates must be constants)
 template <int GPW, int NBUF, int A = 0>
@@ -79,14 +83,16 @@ __device__ __forceinline__ void gm_wait_ahead(int ahead) {
 // 16*NF), BK = staged k-depth per tile (32 or 64), NBUF = LDS ring depth
 // (NBUF-1 tiles of DMA in flight across the barriers — the knob that
 // covers HBM latency on the W stream).
-template <int MW, int NF, int BK, int NBUF, bool SPLITK>
+template <int MW, int NF, int BK, int NBUF, bool SPLITK, bool SWIGLU = false>
 __launch_bounds__(MW * WAVE_SIZE)
 __global__ void gemm_m256_kernel(
-    bf16* __restrict__ y,        // [M, N] (!SPLITK)
+    bf16* __restrict__ y,        // [M, N] (!SPLITK; [M, N/2] when SWIGLU)
     float* __restrict__ yw,      // [nsk, M, N] fp32 slabs (SPLITK)
     const bf16* __restrict__ x,  // [M, K]
     const bf16* __restrict__ w,  // fragment-major [K/32][N/16][64][8]
     int M, int N, int K, int nsk) {
+    static_assert(!(SWIGLU && SPLITK), "fused swiglu epilogue needs nsk=1");
+    static_assert(!SWIGLU || NF % 2 == 0, "fused swiglu pairs fragments");
     constexpr int BM = MW * 32;
     constexpr int BN = NF * 16;
     constexpr int KS = BK / 32;          // MFMA k-steps per tile
@@ -229,6 +235,24 @@ __global__ void gemm_m256_kernel(
                     slab[(size_t)row * N + n0 + n * 16 + (lane & 15)] =
                         acc[f][n][r];
             }
+    } else if constexpr (SWIGLU) {
+        // block-16 interleaved gate/up twin (ops.interleave_gate_up):
+        // even fragment n holds the GATE columns, fragment n+1 the UP
+        // columns of the SAME 16 output columns — the pair lands in the
+        // same lane, so silu(g)*u needs no cross-lane traffic and the
+        // [M, N] intermediate + separate swiglu kernel disappear.
+#pragma unroll
+        for (int f = 0; f < 2; ++f)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int row = m_base + f * 16 + (lane >> 4) * 4 + r;
+                if (row >= M) continue;
+#pragma unroll
+                for (int n = 0; n < NF; n += 2)
+                    y[(size_t)row * (N >> 1) + (n0 >> 1) + (n >> 1) * 16 +
+                      (lane & 15)] =
+                        f2bf(gm_silu(acc[f][n][r]) * acc[f][n + 1][r]);
+            }
     } else {
 #pragma unroll
         for (int f = 0; f < 2; ++f)
@@ -260,13 +284,15 @@ __global__ void gemm_m256_kernel(
 // contiguous store groups.
 // ---------------------------------------------------------------------------
 
-template <int MW, int NF, bool SPLITK>
+template <int MW, int NF, bool SPLITK, bool SWIGLU = false>
 __launch_bounds__(MW * WAVE_SIZE)
 __global__ void gemm_m256r_kernel(
     bf16* __restrict__ y, float* __restrict__ yw,
     const bf16* __restrict__ x,  // [M, K]
     const bf16* __restrict__ w,  // fragment-major [K/32][N/16][64][8]
     int M, int N, int K, int nsk) {
+    static_assert(!(SWIGLU && SPLITK), "fused swiglu epilogue needs nsk=1");
+    static_assert(!SWIGLU || NF % 2 == 0, "fused swiglu pairs fragments");
     constexpr int BM = MW * 32;
     constexpr int BN = NF * 16;
     constexpr int XB = BM * GM_BK * 2;
@@ -401,6 +427,20 @@ __global__ void gemm_m256r_kernel(
                 for (int n = 0; n < NF; ++n)
                     slab[(size_t)row * N + n0 + n * 16 + (lane & 15)] =
                         acc[f][n][r];
+            }
+    } else if constexpr (SWIGLU) {
+        // see gemm_m256_kernel's SWIGLU epilogue (block-16 interleaved twin)
+#pragma unroll
+        for (int f = 0; f < 2; ++f)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int row = m_base + f * 16 + (lane >> 4) * 4 + r;
+                if (row >= M) continue;
+#pragma unroll
+                for (int n = 0; n < NF; n += 2)
+                    y[(size_t)row * (N >> 1) + (n0 >> 1) + (n >> 1) * 16 +
+                      (lane & 15)] =
+                        f2bf(gm_silu(acc[f][n][r]) * acc[f][n + 1][r]);
             }
     } else {
 #pragma unroll
@@ -652,10 +692,14 @@ extern "C" hipError_t launch_gemm_reduce(void*, const float*, int64_t, int,
 //          barrier-lockstep wait gaps.
 extern "C" hipError_t launch_gemm_m256(
     void* y, float* workspace, const void* x, const void* w, int M, int N,
-    int K, int nsk, int nf, int variant, int pipe, hipStream_t stream) {
+    int K, int nsk, int nf, int variant, int pipe, int swiglu,
+    hipStream_t stream) {
     if (M <= 0 || M > 256) return hipErrorInvalidValue;
     if (nsk < 1 || (nsk > 1 && workspace == nullptr)) return hipErrorInvalidValue;
     if ((K % GM_BK) != 0) return hipErrorInvalidValue;
+    // fused swiglu epilogue: y is [M, N/2], single-pass only (the fp32
+    // split-K slabs hold pre-activation partials — no fusion point)
+    if (swiglu && (nsk != 1 || variant == 2)) return hipErrorInvalidValue;
     if (variant == 2) {
         if ((N % 16) != 0) return hipErrorInvalidValue;
         dim3 grid((N + PC_BN - 1) / PC_BN, nsk);
@@ -681,47 +725,49 @@ extern "C" hipError_t launch_gemm_m256(
     const int tiles = N / (16 * nf);
     dim3 grid(tiles, nsk);
     dim3 block(mw * WAVE_SIZE);
-#define GM_L2(MWV, NFV, SPLIT)                                                 \
+#define GM_L2(MWV, NFV, SPLIT, SW)                                             \
     do {                                                                       \
         if (variant == 1)                                                      \
-            gemm_m256r_kernel<MWV, NFV, SPLIT><<<grid, block, 0, stream>>>(    \
+            gemm_m256r_kernel<MWV, NFV, SPLIT, SW><<<grid, block, 0,           \
+                                                     stream>>>(               \
                 (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,  \
                 nsk);                                                          \
         else if (pipe == 1 && NFV == 4)                                        \
-            gemm_m256_kernel<MWV, 4, 64, 4, SPLIT><<<grid, block, 0,           \
-                                                     stream>>>(               \
+            gemm_m256_kernel<MWV, 4, 64, 4, SPLIT, SW><<<grid, block, 0,       \
+                                                         stream>>>(           \
                 (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,  \
                 nsk);                                                          \
         else if (pipe == 2 && NFV == 8)                                        \
-            gemm_m256_kernel<MWV, 8, 32, 4, SPLIT><<<grid, block, 0,           \
-                                                     stream>>>(               \
+            gemm_m256_kernel<MWV, 8, 32, 4, SPLIT, SW><<<grid, block, 0,       \
+                                                         stream>>>(           \
                 (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,  \
                 nsk);                                                          \
         else if (pipe == 3 && NFV == 8)                                        \
-            gemm_m256_kernel<MWV, 8, 32, 6, SPLIT><<<grid, block, 0,           \
-                                                     stream>>>(               \
+            gemm_m256_kernel<MWV, 8, 32, 6, SPLIT, SW><<<grid, block, 0,       \
+                                                         stream>>>(           \
                 (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,  \
                 nsk);                                                          \
         else if (pipe == 4 && NFV == 4)                                        \
-            gemm_m256_kernel<MWV, 4, 64, 2, SPLIT><<<grid, block, 0,           \
-                                                     stream>>>(               \
+            gemm_m256_kernel<MWV, 4, 64, 2, SPLIT, SW><<<grid, block, 0,       \
+                                                         stream>>>(           \
                 (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,  \
                 nsk);                                                          \
         else if (pipe == 5 && NFV == 4)                                        \
-            gemm_m256_kernel<MWV, 4, 32, 3, SPLIT><<<grid, block, 0,           \
-                                                     stream>>>(               \
+            gemm_m256_kernel<MWV, 4, 32, 3, SPLIT, SW><<<grid, block, 0,       \
+                                                         stream>>>(           \
                 (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,  \
                 nsk);                                                          \
         else                                                                   \
-            gemm_m256_kernel<MWV, NFV, 64, 3, SPLIT><<<grid, block, 0,         \
-                                                       stream>>>(             \
+            gemm_m256_kernel<MWV, NFV, 64, 3, SPLIT, SW><<<grid, block, 0,     \
+                                                           stream>>>(         \
                 (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,  \
                 nsk);                                                          \
     } while (0)
 #define GM_L1(MWV, NFV)                                                        \
     do {                                                                       \
-        if (nsk > 1) GM_L2(MWV, NFV, true);                                    \
-        else GM_L2(MWV, NFV, false);                                           \
+        if (swiglu) GM_L2(MWV, NFV, false, true);                              \
+        else if (nsk > 1) GM_L2(MWV, NFV, true, false);                        \
+        else GM_L2(MWV, NFV, false, false);                                    \
     } while (0)
 #define GM_L0(MWV)                                                             \
     do {                                                                       \

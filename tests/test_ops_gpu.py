@@ -331,6 +331,37 @@ def test_gemm_m256_70b_dispatch(N, K):
     assert err < 0.02
 
 
+@pytest.mark.parametrize("M,variant,nf", [(256, 0, 8), (256, 1, 8), (100, 1, 8), (64, 0, 4)])
+def test_gemm_m256_fused_swiglu(M, variant, nf):
+    # fused gate_up+swiglu epilogue vs plain torch: silu(x@g.T) * (x@u.T)
+    N, K = 28672, 4096
+    torch.manual_seed(M + variant)
+    x = (torch.randn(M, K, device=DEV) * 0.5).bfloat16()
+    w = (torch.randn(N, K, device=DEV) * 0.02).bfloat16()
+    wi = ops.swizzle_weight_frag(ops.interleave_gate_up(w))
+    got = ops.gemm_m256_swiglu(x, wi, nf=nf, variant=variant)
+    g, u = (x.float() @ w.float().T).chunk(2, dim=-1)
+    ref = torch.nn.functional.silu(g) * u
+    err = (got.float() - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
+    assert got.shape == (M, N // 2)
+    assert err < 0.02
+
+
+def test_swiglu_linear_routes_fused():
+    M, N, K = 128, 28672, 4096
+    torch.manual_seed(11)
+    x = (torch.randn(M, K, device=DEV) * 0.5).bfloat16()
+    w = (torch.randn(N, K, device=DEV) * 0.02).bfloat16()
+    wi = ops.swizzle_weight_frag(ops.interleave_gate_up(w))
+    assert ops._m256_swiglu_config(M, N, K) is not None
+    got = ops.swiglu_linear(x, w, None, wi)
+    ref = ops.swiglu(torch.nn.functional.linear(x, w))
+    err = (got.float() - ref.float()).abs().max().item() / (
+        ref.float().abs().max().item() + 1e-3
+    )
+    assert err < 0.02
+
+
 def test_gemm_m256_repeat_launches_stable():
     # split-K slab reuse across launches (hipGraph replay pattern)
     M, N, K = 64, 4096, 14336

@@ -121,12 +121,24 @@ def test_gemm_m256_guards(M, N, K, nf, nsk):
     y, ycheck = guarded((M, N), torch.bfloat16)
     ws, wscheck = guarded((nsk * M * N,), torch.float32)
     for variant in (0, 1):
-        ops._native().gemm_m256(y, x, wf, ws, nsk, nf, variant, 0 if nf == 4 else 0)
+        ops._native().gemm_m256(y, x, wf, ws, nsk, nf, variant, 0 if nf == 4 else 0, 0)
         ycheck(f"gemm_m256 y v{variant}")
         wscheck(f"gemm_m256 slab v{variant}")
     ref = x.float() @ w.float().T
     err = (y.float() - ref).abs().max().item() / (ref.abs().max().item() + 1e-3)
     assert err < 0.02
+
+
+def test_gemm_m256_swiglu_guards():
+    # fused epilogue writes [M, N/2] — the guard catches any write past it
+    M, N, K = 200, 28672, 4096
+    x = (torch.randn(M, K, device=DEV) * 0.5).bfloat16()
+    w = (torch.randn(N, K, device=DEV) * 0.02).bfloat16()
+    wi = ops.swizzle_weight_frag(ops.interleave_gate_up(w))
+    y, ycheck = guarded((M, N // 2), torch.bfloat16)
+    for variant in (0, 1):
+        ops._native().gemm_m256(y, x, wi, None, 1, 8, variant, 0, 1)
+        ycheck(f"gemm_m256 swiglu y v{variant}")
 
 
 def test_gemm_skinny_guards():
