@@ -348,6 +348,9 @@ def _m256_config(M: int, N: int, K: int) -> Optional[dict]:
         if K > 8192:
             return {"nf": 8, "nsk": 4, "variant": 1, "pipe": 0}  # 70B down 1.30x
         return {"nf": 8, "nsk": 4, "variant": 0, "pipe": 0}      # 70B o 1.14x
+    if N == 7168 and K <= 8192:
+        # tp=4 gate_up shard (unfused path): 37.3 vs library 43.4 us
+        return {"nf": 4, "nsk": 3, "variant": 0, "pipe": 4}
     if N > 4096:
         # library wins the very wide shapes: gate_up N=28672/57344
         # (custom best 73.3 vs 70.3 us / 287.9 vs 258.0), qkv N=6144
@@ -356,6 +359,15 @@ def _m256_config(M: int, N: int, K: int) -> Optional[dict]:
         return None
     if K > 8192 and N % 128 == 0:
         return {"nf": 8, "nsk": 8, "variant": 0, "pipe": 0}  # down 1.34x
+    if N < 2048:
+        # narrow column shards (tp=4 qkv N=1536): too few tiles to fill
+        # the chip — library 1.8x faster (21.7 vs 39.7 us)
+        return None
+    if K < 2048:
+        # shallow-K row shards (tp=4 o-proj K=1024): single-pass
+        # register-staged wins 1.67x (12.7 vs 21.2 us); split-K slab
+        # traffic would dominate this little work
+        return {"nf": 4, "nsk": 1, "variant": 1, "pipe": 0}
     # o-proj class (N<=4096, K<=8192): BK64/NBUF2 2-blocks/CU + split-K
     tiles = N // 64
     nsk = max(1, min(-(-256 // tiles), (K // 64) // 2, 8))

@@ -26,6 +26,16 @@ def test_70b_routing():
     assert cfg == {"nf": 8, "nsk": 4, "variant": 1, "pipe": 0}
 
 
+# llama-3-8b tp=4 per-rank shard shapes
+def test_8b_tp4_routing():
+    assert _m256_config(256, 1536, 4096) is None        # qkv shard -> library
+    cfg = _m256_config(256, 4096, 1024)                 # o shard -> custom
+    assert cfg == {"nf": 4, "nsk": 1, "variant": 1, "pipe": 0}
+    cfg = _m256_config(256, 7168, 4096)                 # gate_up shard -> custom
+    assert cfg == {"nf": 4, "nsk": 3, "variant": 0, "pipe": 4}
+    assert _m256_config(256, 4096, 3584) is not None    # down shard -> custom
+
+
 def test_prefill_m_goes_to_library():
     # chunked-prefill token counts exceed the macro-tile M ceiling
     assert _m256_config(4096, 4096, 4096) is None
