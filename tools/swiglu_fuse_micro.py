@@ -1,7 +1,7 @@
 """Fused gate_up+swiglu (gemm_m256 SWIGLU epilogue) vs library GEMM +
 separate swiglu kernel, cold-L3 (2 GB weight rotation). Run on the GPU box:
 
-    python tools/swiglu_fuse_micro.py [M]
+    python tools/swiglu_fuse_micro.py [M [N K]]
 """
 
 import os
@@ -33,7 +33,8 @@ def main():
     assert torch.cuda.is_available()
     dev = "cuda:0"
     M = int(sys.argv[1]) if len(sys.argv) > 1 else 256
-    for (N, K) in SHAPES:
+    shapes = [tuple(int(a) for a in sys.argv[2:4])] if len(sys.argv) == 4 else SHAPES
+    for (N, K) in shapes:
         x = (torch.randn(M, K, device=dev) * 0.5).bfloat16()
         ncopies = max(1, min(16, (2048 << 20) // (N * K * 2)))
         ws = [(torch.randn(N, K, device=dev) * 0.02).bfloat16() for _ in range(ncopies)]
