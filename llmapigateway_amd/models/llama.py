@@ -165,6 +165,15 @@ class LlamaModel:
                         # would be dead HBM (the library wins those shapes),
                         # ~9 GB at 8B and ~89 GB at 70B
                         and ops._m256_config(256, w.shape[0], w.shape[1]) is not None
+                        # gate_up with a fused-swiglu twin never uses the
+                        # plain twin — don't hold both in HBM
+                        and not (
+                            name == "gate_up"
+                            and ops._m256_swiglu_config(
+                                256, w.shape[0], w.shape[1]
+                            )
+                            is not None
+                        )
                     ):
                         layer[name + "_swz"] = ops.swizzle_weight_frag(w)
                 gu = layer["gate_up"]
