@@ -331,7 +331,10 @@ def test_gemm_m256_70b_dispatch(N, K):
     assert err < 0.02
 
 
-@pytest.mark.parametrize("M,variant,nf", [(256, 0, 8), (256, 1, 8), (100, 1, 8), (64, 0, 4)])
+@pytest.mark.parametrize(
+    "M,variant,nf",
+    [(256, 0, 8), (256, 1, 8), (100, 1, 8), (100, 0, 8), (64, 0, 4)],
+)
 def test_gemm_m256_fused_swiglu(M, variant, nf):
     # fused gate_up+swiglu epilogue vs plain torch: silu(x@g.T) * (x@u.T)
     N, K = 28672, 4096
