@@ -116,13 +116,97 @@ __global__ void rmsnorm_kernel(
     }
 }
 
+// Wave-per-row variant for the decode regime: T <= ~1024 rows means the
+// block-per-row form leaves the chip latency-bound (measured ~5.2 us flat
+// for T <= 256 vs ~0.3 us of L2 traffic — tools/rmsnorm_micro.py). One
+// 64-lane wave owns a whole row: the block barrier and the LDS reduce
+// round trip disappear (wave_reduce only), 4 independent rows share a
+// block. H <= 4096 so the row fits the per-lane register cache (H/64 <= 64
+// floats); the launcher falls back to the block form above otherwise.
+template <bool FUSED_RESIDUAL>
+__global__ void rmsnorm_wave_kernel(
+    bf16* __restrict__ out, const bf16* __restrict__ x,
+    bf16* __restrict__ residual, const bf16* __restrict__ weight, float eps,
+    int T, int H) {
+    const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+    if (row >= T) return;
+    const int lane = threadIdx.x & (WAVE_SIZE - 1);
+    const uint4* xrow = reinterpret_cast<const uint4*>(x + (size_t)row * H);
+    uint4* rrow = FUSED_RESIDUAL
+                      ? reinterpret_cast<uint4*>(residual + (size_t)row * H)
+                      : nullptr;
+    uint4* orow = reinterpret_cast<uint4*>(out + (size_t)row * H);
+    const uint4* wv = reinterpret_cast<const uint4*>(weight);
+    const int nvec = H / 8;
+
+    float vals[64];  // H/64 <= 64 per-lane floats (launcher-enforced)
+    int held = 0;
+    float ssq = 0.f;
+    for (int i = lane; i < nvec; i += WAVE_SIZE) {
+        uint4 vx = xrow[i];
+        float f[8];
+        unpack2(vx.x, f[0], f[1]);
+        unpack2(vx.y, f[2], f[3]);
+        unpack2(vx.z, f[4], f[5]);
+        unpack2(vx.w, f[6], f[7]);
+        if (FUSED_RESIDUAL) {
+            uint4 vr = rrow[i];
+            float g[8];
+            unpack2(vr.x, g[0], g[1]);
+            unpack2(vr.y, g[2], g[3]);
+            unpack2(vr.z, g[4], g[5]);
+            unpack2(vr.w, g[6], g[7]);
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+                f[j] = bfbits2f(f2bfbits(f[j] + g[j]));
+            uint4 vw;
+            vw.x = pack2(f[0], f[1]);
+            vw.y = pack2(f[2], f[3]);
+            vw.z = pack2(f[4], f[5]);
+            vw.w = pack2(f[6], f[7]);
+            rrow[i] = vw;
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            ssq += f[j] * f[j];
+            vals[held + j] = f[j];
+        }
+        held += 8;
+    }
+    const float inv = rsqrtf(wave_reduce_sum(ssq) / (float)H + eps);
+    held = 0;
+    for (int i = lane; i < nvec; i += WAVE_SIZE) {
+        uint4 vw = wv[i];
+        float w8[8];
+        unpack2(vw.x, w8[0], w8[1]);
+        unpack2(vw.y, w8[2], w8[3]);
+        unpack2(vw.z, w8[4], w8[5]);
+        unpack2(vw.w, w8[6], w8[7]);
+        uint4 vo;
+        vo.x = pack2(vals[held] * inv * w8[0], vals[held + 1] * inv * w8[1]);
+        vo.y = pack2(vals[held + 2] * inv * w8[2], vals[held + 3] * inv * w8[3]);
+        vo.z = pack2(vals[held + 4] * inv * w8[4], vals[held + 5] * inv * w8[5]);
+        vo.w = pack2(vals[held + 6] * inv * w8[6], vals[held + 7] * inv * w8[7]);
+        orow[i] = vo;
+        held += 8;
+    }
+}
+
+static inline bool rms_wave_fits(int T, int H) {
+    return T <= 1024 && H <= 4096 && H % 8 == 0;
+}
+
 extern "C" hipError_t launch_rmsnorm(
     void* out, const void* x, const void* weight, float eps, int T, int H,
     hipStream_t stream) {
     if (H % 8 != 0) return hipErrorInvalidValue;
-    const int threads = 256;
-    rmsnorm_kernel<false><<<T, threads, 0, stream>>>(
-        (bf16*)out, (const bf16*)x, nullptr, (const bf16*)weight, eps, H);
+    if (rms_wave_fits(T, H)) {
+        rmsnorm_wave_kernel<false><<<(T + 3) / 4, 256, 0, stream>>>(
+            (bf16*)out, (const bf16*)x, nullptr, (const bf16*)weight, eps, T, H);
+    } else {
+        rmsnorm_kernel<false><<<T, 256, 0, stream>>>(
+            (bf16*)out, (const bf16*)x, nullptr, (const bf16*)weight, eps, H);
+    }
     HIP_CHECK_LAST();
     return hipSuccess;
 }
@@ -131,9 +215,15 @@ extern "C" hipError_t launch_rmsnorm_residual(
     void* out, const void* x, void* residual, const void* weight, float eps,
     int T, int H, hipStream_t stream) {
     if (H % 8 != 0) return hipErrorInvalidValue;
-    const int threads = 256;
-    rmsnorm_kernel<true><<<T, threads, 0, stream>>>(
-        (bf16*)out, (const bf16*)x, (bf16*)residual, (const bf16*)weight, eps, H);
+    if (rms_wave_fits(T, H)) {
+        rmsnorm_wave_kernel<true><<<(T + 3) / 4, 256, 0, stream>>>(
+            (bf16*)out, (const bf16*)x, (bf16*)residual, (const bf16*)weight,
+            eps, T, H);
+    } else {
+        rmsnorm_kernel<true><<<T, 256, 0, stream>>>(
+            (bf16*)out, (const bf16*)x, (bf16*)residual, (const bf16*)weight,
+            eps, H);
+    }
     HIP_CHECK_LAST();
     return hipSuccess;
 }
