@@ -121,9 +121,11 @@ __global__ void rmsnorm_kernel(
 // for T <= 256 vs ~0.3 us of L2 traffic — tools/rmsnorm_micro.py). One
 // 64-lane wave owns a whole row: the block barrier and the LDS reduce
 // round trip disappear (wave_reduce only), 4 independent rows share a
-// block. H <= 4096 so the row fits the per-lane register cache (H/64 <= 64
-// floats); the launcher falls back to the block form above otherwise.
-template <bool FUSED_RESIDUAL>
+// block. VPL = row uint4-chunks per lane (H / 512), a template constant so
+// the register row cache fully unrolls — with a runtime bound the compiler
+// demoted vals[] to 272 B/lane of scratch and the kernel measured SLOWER
+// than the block form (8.2 vs 5.2 us); compile-time it is pure VGPRs.
+template <bool FUSED_RESIDUAL, int VPL>
 __global__ void rmsnorm_wave_kernel(
     bf16* __restrict__ out, const bf16* __restrict__ x,
     bf16* __restrict__ residual, const bf16* __restrict__ weight, float eps,
@@ -137,12 +139,12 @@ __global__ void rmsnorm_wave_kernel(
                       : nullptr;
     uint4* orow = reinterpret_cast<uint4*>(out + (size_t)row * H);
     const uint4* wv = reinterpret_cast<const uint4*>(weight);
-    const int nvec = H / 8;
 
-    float vals[64];  // H/64 <= 64 per-lane floats (launcher-enforced)
-    int held = 0;
+    float vals[VPL * 8];
     float ssq = 0.f;
-    for (int i = lane; i < nvec; i += WAVE_SIZE) {
+#pragma unroll
+    for (int t = 0; t < VPL; ++t) {
+        const int i = lane + t * WAVE_SIZE;
         uint4 vx = xrow[i];
         float f[8];
         unpack2(vx.x, f[0], f[1]);
@@ -169,13 +171,13 @@ __global__ void rmsnorm_wave_kernel(
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
             ssq += f[j] * f[j];
-            vals[held + j] = f[j];
+            vals[t * 8 + j] = f[j];
         }
-        held += 8;
     }
     const float inv = rsqrtf(wave_reduce_sum(ssq) / (float)H + eps);
-    held = 0;
-    for (int i = lane; i < nvec; i += WAVE_SIZE) {
+#pragma unroll
+    for (int t = 0; t < VPL; ++t) {
+        const int i = lane + t * WAVE_SIZE;
         uint4 vw = wv[i];
         float w8[8];
         unpack2(vw.x, w8[0], w8[1]);
@@ -183,27 +185,49 @@ __global__ void rmsnorm_wave_kernel(
         unpack2(vw.z, w8[4], w8[5]);
         unpack2(vw.w, w8[6], w8[7]);
         uint4 vo;
-        vo.x = pack2(vals[held] * inv * w8[0], vals[held + 1] * inv * w8[1]);
-        vo.y = pack2(vals[held + 2] * inv * w8[2], vals[held + 3] * inv * w8[3]);
-        vo.z = pack2(vals[held + 4] * inv * w8[4], vals[held + 5] * inv * w8[5]);
-        vo.w = pack2(vals[held + 6] * inv * w8[6], vals[held + 7] * inv * w8[7]);
+        vo.x = pack2(vals[t * 8] * inv * w8[0], vals[t * 8 + 1] * inv * w8[1]);
+        vo.y = pack2(vals[t * 8 + 2] * inv * w8[2], vals[t * 8 + 3] * inv * w8[3]);
+        vo.z = pack2(vals[t * 8 + 4] * inv * w8[4], vals[t * 8 + 5] * inv * w8[5]);
+        vo.w = pack2(vals[t * 8 + 6] * inv * w8[6], vals[t * 8 + 7] * inv * w8[7]);
         orow[i] = vo;
-        held += 8;
     }
 }
 
+template <bool FUSED>
+static hipError_t launch_rms_wave(bf16* out, const bf16* x, bf16* residual,
+                                  const bf16* w, float eps, int T, int H,
+                                  hipStream_t stream) {
+    const dim3 grid((T + 3) / 4);
+#define RMS_W(V)                                                               \
+    rmsnorm_wave_kernel<FUSED, V>                                              \
+        <<<grid, 256, 0, stream>>>(out, x, residual, w, eps, T, H)
+    switch (H / 512) {
+        case 1: RMS_W(1); break;
+        case 2: RMS_W(2); break;
+        case 3: RMS_W(3); break;
+        case 4: RMS_W(4); break;
+        case 5: RMS_W(5); break;
+        case 6: RMS_W(6); break;
+        case 7: RMS_W(7); break;
+        default: RMS_W(8); break;
+    }
+#undef RMS_W
+    HIP_CHECK_LAST();
+    return hipSuccess;
+}
+
 static inline bool rms_wave_fits(int T, int H) {
-    return T <= 1024 && H <= 4096 && H % 8 == 0;
+    return T <= 1024 && H <= 4096 && H % 512 == 0;
 }
 
 extern "C" hipError_t launch_rmsnorm(
     void* out, const void* x, const void* weight, float eps, int T, int H,
     hipStream_t stream) {
     if (H % 8 != 0) return hipErrorInvalidValue;
-    if (rms_wave_fits(T, H)) {
-        rmsnorm_wave_kernel<false><<<(T + 3) / 4, 256, 0, stream>>>(
-            (bf16*)out, (const bf16*)x, nullptr, (const bf16*)weight, eps, T, H);
-    } else {
+    if (rms_wave_fits(T, H))
+        return launch_rms_wave<false>((bf16*)out, (const bf16*)x, nullptr,
+                                      (const bf16*)weight, eps, T, H, stream);
+    {
         rmsnorm_kernel<false><<<T, 256, 0, stream>>>(
             (bf16*)out, (const bf16*)x, nullptr, (const bf16*)weight, eps, H);
     }
@@ -215,11 +239,11 @@ extern "C" hipError_t launch_rmsnorm_residual(
     void* out, const void* x, void* residual, const void* weight, float eps,
     int T, int H, hipStream_t stream) {
     if (H % 8 != 0) return hipErrorInvalidValue;
-    if (rms_wave_fits(T, H)) {
-        rmsnorm_wave_kernel<true><<<(T + 3) / 4, 256, 0, stream>>>(
-            (bf16*)out, (const bf16*)x, (bf16*)residual, (const bf16*)weight,
-            eps, T, H);
-    } else {
+    if (rms_wave_fits(T, H))
+        return launch_rms_wave<true>((bf16*)out, (const bf16*)x,
+                                     (bf16*)residual, (const bf16*)weight,
+                                     eps, T, H, stream);
+    {
         rmsnorm_kernel<true><<<T, 256, 0, stream>>>(
             (bf16*)out, (const bf16*)x, (bf16*)residual, (const bf16*)weight,
             eps, H);
