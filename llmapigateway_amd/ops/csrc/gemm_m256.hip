@@ -284,7 +284,9 @@ __global__ void gemm_m256_kernel(
 // contiguous store groups.
 // ---------------------------------------------------------------------------
 
-template <int MW, int NF, bool SPLITK, bool SWIGLU = false>
+// NTW: stream W through non-temporal loads — the weight stream has zero
+// reuse, so keeping it out of L2 leaves the cache to X and the consumers.
+template <int MW, int NF, bool SPLITK, bool SWIGLU = false, bool NTW = false>
 __launch_bounds__(MW * WAVE_SIZE)
 __global__ void gemm_m256r_kernel(
     bf16* __restrict__ y, float* __restrict__ yw,
@@ -348,8 +350,10 @@ __global__ void gemm_m256r_kernel(
         _Pragma("unroll") for (int j = 0; j < WG_W; ++j) {                     \
             const size_t gfi__ = (size_t)(2 * kt__ + wfi[j] / NF) * n16 +      \
                                  n0 / 16 + wfi[j] % NF;                        \
-            wr[j] = *(const __attribute__((address_space(1))) bf16x8*)(        \
-                w + gfi__ * 512 + lane * 8);                                   \
+            const __attribute__((address_space(1))) bf16x8* wp__ =             \
+                (const __attribute__((address_space(1))) bf16x8*)(             \
+                    w + gfi__ * 512 + lane * 8);                               \
+            wr[j] = NTW ? __builtin_nontemporal_load(wp__) : *wp__;            \
         }                                                                      \
     } while (0)
 #define GMR_WRITE(T)                                                           \
@@ -727,7 +731,12 @@ extern "C" hipError_t launch_gemm_m256(
     dim3 block(mw * WAVE_SIZE);
 #define GM_L2(MWV, NFV, SPLIT, SW)                                             \
     do {                                                                       \
-        if (variant == 1)                                                      \
+        if (variant == 1 && pipe == 1)                                         \
+            gemm_m256r_kernel<MWV, NFV, SPLIT, SW, true><<<grid, block, 0,     \
+                                                           stream>>>(         \
+                (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,  \
+                nsk);                                                          \
+        else if (variant == 1)                                                 \
             gemm_m256r_kernel<MWV, NFV, SPLIT, SW><<<grid, block, 0,           \
                                                      stream>>>(               \
                 (bf16*)y, workspace, (const bf16*)x, (const bf16*)w, M, N, K,  \

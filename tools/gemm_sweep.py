@@ -65,6 +65,7 @@ def main():
                 for pipe in pipes:
                     combos.append((0, nf, pipe, nsk))
                 combos.append((1, nf, 0, nsk))
+                combos.append((1, nf, 1, nsk))  # non-temporal W stream
         # producer/consumer variant (BN=96, nf ignored)
         nsks2 = sorted({1, max(1, -(-256 // max(1, N // 96)))})
         for nsk in nsks2:

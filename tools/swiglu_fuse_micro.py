@@ -45,7 +45,7 @@ def main():
         print(f"--- M={M} N={N} K={K}  library+swiglu {t_lib:.1f} us ---")
         for nf in (4, 8):
             pipes = {4: [0, 1, 4, 5], 8: [0, 2, 3]}[nf]
-            for variant, pipe in [(1, 0)] + [(0, p) for p in pipes]:
+            for variant, pipe in [(1, 0), (1, 1)] + [(0, p) for p in pipes]:
                 try:
                     got = ops.gemm_m256_swiglu(x, wi[0], nf=nf, variant=variant, pipe=pipe)
                 except Exception as e:
