@@ -272,3 +272,55 @@ def test_fp8_kv_capacity_gain():
     bf16 = PagedKVCache.block_bytes(cfg, 64, torch.bfloat16)
     fp8 = PagedKVCache.block_bytes(cfg, 64, torch.bfloat16, kv_dtype="fp8")
     assert fp8 < 0.54 * bf16  # ~1.94x the tokens per byte
+
+
+def test_abort_during_chunked_prefill_defers_block_frees():
+    # Regression for the round-2 soak finding: aborting a request whose
+    # prompt is mid-chunk must not free its KV blocks while the current
+    # forward still writes them (the engine defers the free to the next
+    # step top) and must not corrupt a concurrent request's blocks.
+    eng = make_engine(prefill_budget=16, num_blocks=32)
+    long_req = EngineRequest(
+        list(range(1, 49)), SamplingParams(max_tokens=4, ignore_eos=True)
+    )
+    probe_prompt = [2, 4, 6, 8, 10]
+    eng.add_request(long_req)
+    eng.step()  # first chunk issued; long_req is mid-prompt
+    assert long_req.state == "running" and long_req.prefill_pos > 0
+    eng.abort_request(long_req)
+    # the abort must leave the engine steppable and the blocks deferred
+    probe = EngineRequest(
+        list(probe_prompt), SamplingParams(max_tokens=6, ignore_eos=True)
+    )
+    eng.add_request(probe)
+    for _ in range(20):
+        if probe.state == "finished":
+            break
+        eng.step()
+    assert probe.state == "finished"
+    assert long_req.finish_reason == "aborted"
+    # all blocks back: a fresh max-size request must be admittable
+    assert eng.kv.manager.num_free_blocks + len(eng.kv.manager._evictable) \
+        == eng.kv.manager.allocator.num_blocks
+    # and the probe's output matches a clean engine (no KV corruption)
+    eng2 = make_engine(prefill_budget=16, num_blocks=32)
+    probe2 = EngineRequest(
+        list(probe_prompt), SamplingParams(max_tokens=6, ignore_eos=True)
+    )
+    eng2.add_request(probe2)
+    for _ in range(20):
+        if probe2.state == "finished":
+            break
+        eng2.step()
+    assert probe.out_ids == probe2.out_ids
+
+
+def test_max_completion_tokens_precedence():
+    # OpenAI: max_completion_tokens wins over legacy max_tokens
+    p = SamplingParams.from_payload(
+        {"max_tokens": 10, "max_completion_tokens": 7})
+    assert p.max_tokens == 7
+    p = SamplingParams.from_payload({"max_tokens": 10})
+    assert p.max_tokens == 10
+    p = SamplingParams.from_payload({}, default_max_tokens=33)
+    assert p.max_tokens == 33
